@@ -1,0 +1,22 @@
+"""MI355X-native asynchronous parameter-server training framework.
+
+A from-scratch re-design of ``FlinkML/flink-parameter-server`` for AMD
+Instinct MI355X: the reference's Flink iterative-stream loop between
+workers and parameter shards becomes round-based RCCL send/recv over xGMI
+(one process per GPU, each hosting one worker and one server shard whose
+partition lives dense in HBM3E); the server-side sparse applies and the
+worker-side model steps are hand-written HIP/CDNA4 kernels.
+
+See SURVEY.md for the reference's structure and the design mapping.
+"""
+__version__ = "0.1.0"
+
+from .api import (ParameterServer, ParameterServerClient, ServerLogic,
+                  SimpleServerLogic, WorkerLogic, add_pull_limiter)
+from .config import PSConfig
+from .table import ShardedTable, TableServerLogic
+from .parallel.runtime import PSEngine
+from .parallel.transport import (LocalTransport, LoopbackTransport,
+                                 TorchDistTransport, Transport)
+from .parallel.partitioner import (ModPartitioner, RangePartitioner,
+                                   make_partitioner)

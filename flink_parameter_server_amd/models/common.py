@@ -1,0 +1,55 @@
+"""Shared worker-side machinery for the model logics.
+
+``PullBuffer`` implements the "wait until every pulled parameter of a batch
+has been answered" reassembly the reference does per instance in the
+passive-aggressive worker (``PassiveAggressiveParameterServer`` inner
+logic) and per rating in the MF worker
+(``PSOnlineMatrixFactorization`` pending-pull queueing) —
+here batched: one buffer per training batch, filled as answer segments
+arrive (answers may arrive split across rounds under a pull limit, and the
+same id may be answered more than once when overlapping batches pull it;
+a duplicate answer must not double-count).
+"""
+from __future__ import annotations
+
+import torch
+
+
+class PullBuffer:
+    """Collects pulled rows for a fixed set of unique ids."""
+
+    def __init__(self, uids: torch.Tensor, dim: int):
+        uids, _ = torch.sort(uids)
+        self.uids = uids
+        self.values = torch.empty(uids.numel(), dim, dtype=torch.float32,
+                                  device=uids.device)
+        self._have = torch.zeros(uids.numel(), dtype=torch.bool,
+                                 device=uids.device)
+        self._n_have = 0
+
+    def slots_of(self, ids: torch.Tensor) -> torch.Tensor:
+        """Position of each id in the sorted unique-id list."""
+        return torch.searchsorted(self.uids, ids)
+
+    def offer(self, ids: torch.Tensor, values: torch.Tensor) -> None:
+        """Fill rows answered by (ids, values); ignores foreign ids and
+        rows already filled."""
+        if self.complete or not self.uids.numel():
+            return
+        pos = torch.searchsorted(self.uids, ids)
+        pos_c = torch.clamp(pos, max=self.uids.numel() - 1)
+        hit = self.uids[pos_c] == ids
+        if not bool(hit.any()):
+            return
+        rows = pos_c[hit]
+        fresh = ~self._have[rows]
+        rows = rows[fresh]
+        if not rows.numel():
+            return
+        self.values[rows] = values[hit][fresh].to(self.values.dtype)
+        self._have[rows] = True
+        self._n_have += int(rows.numel())
+
+    @property
+    def complete(self) -> bool:
+        return self._n_have >= self.uids.numel()

@@ -1,0 +1,144 @@
+"""Matrix-factorization SGD on the parameter server.
+
+Reference: ``matrix/factorization/PSOnlineMatrixFactorization.scala``
+(``psOnlineMF(ratings, numFactors, rangeMin, rangeMax, learningRate,
+negativeSampleRate, userMemory, pullLimit, workerParallelism,
+psParallelism, iterationWaitTime)``) and
+``PSOfflineMatrixFactorization.scala`` (epoch sweeps) — SURVEY.md §3.3.
+
+Model split (reference behavior, kept): user factors are WORKER-LOCAL
+state (the reference bounds them with a ``userMemory`` LRU; on MI355X with
+288 GB HBM we keep the worker's user partition dense and resident); item
+factors live on the parameter-server shards.  Per batch of ratings the
+worker pulls the item vectors, runs the fused SGD step (HIP kernel K4 on
+GPU: e = r - u.i; u += lr*(e*i - reg*u); push lr*(e*u - reg*i)), and pushes
+the item deltas; the server applies them with paramUpdate='+'
+(``SimplePSLogic(RangedRandomFactorInitializer, vectorSum)``).
+
+Negative sampling (reference: ``negativeSampleRate`` random items per
+positive, rating 0) is generated on the fly with the deterministic hash
+RNG so no host RNG state is carried.
+"""
+from __future__ import annotations
+
+from typing import Iterator, List, Optional, Tuple
+
+import torch
+
+from .. import ops
+from ..api import ParameterServerClient, WorkerLogic
+from .common import PullBuffer
+
+
+class MFBatch:
+    """One batch of ratings local to a worker: user rows are indices into
+    the worker's local user-factor table."""
+
+    __slots__ = ("user_rows", "item_ids", "ratings")
+
+    def __init__(self, user_rows: torch.Tensor, item_ids: torch.Tensor,
+                 ratings: torch.Tensor):
+        self.user_rows = user_rows
+        self.item_ids = item_ids
+        self.ratings = ratings
+
+    def __len__(self) -> int:
+        return int(self.user_rows.numel())
+
+
+class OnlineMFWorkerLogic(WorkerLogic):
+    def __init__(self, n_local_users: int, n_factors: int,
+                 learning_rate: float, reg: float = 0.0,
+                 range_min: float = -0.01, range_max: float = 0.01,
+                 negative_sample_rate: int = 0, n_items: int = 0,
+                 device: torch.device = torch.device("cpu"),
+                 pull_limit: Optional[int] = None,
+                 output_user_factors: bool = False,
+                 seed: int = 1):
+        self.n_factors = int(n_factors)
+        self.lr = float(learning_rate)
+        self.reg = float(reg)
+        self.neg_rate = int(negative_sample_rate)
+        self.n_items = int(n_items)
+        self.device = torch.device(device)
+        self.pull_limit = pull_limit
+        self.output_user_factors = output_user_factors
+        self._neg_counter = 0
+        self._seed = seed
+        # Worker-local user table (reference: per-subtask user map with
+        # deterministic ranged-random init).
+        self.user_table = torch.empty(n_local_users, n_factors,
+                                      dtype=torch.float32, device=self.device)
+        if n_local_users:
+            ops.hash_init_rows(self.user_table, 0, 1, range_min, range_max,
+                               seed=seed + 7777)
+        self._pending: List[Tuple[MFBatch, PullBuffer]] = []
+
+    def batch_size(self, batch: MFBatch) -> int:
+        return len(batch)
+
+    def on_recv(self, data: MFBatch, client: ParameterServerClient) -> None:
+        if self.neg_rate > 0:
+            data = self._with_negatives(data)
+        uids = torch.unique(data.item_ids)
+        buf = PullBuffer(uids, self.n_factors)
+        self._pending.append((data, buf))
+        client.pull(uids)
+
+    def on_pull_recv(self, ids: torch.Tensor, values: torch.Tensor,
+                     client: ParameterServerClient) -> None:
+        still: List[Tuple[MFBatch, PullBuffer]] = []
+        for batch, buf in self._pending:
+            buf.offer(ids, values)
+            if buf.complete:
+                self._train(batch, buf, client)
+            else:
+                still.append((batch, buf))
+        self._pending = still
+
+    def _train(self, batch: MFBatch, buf: PullBuffer,
+               client: ParameterServerClient) -> None:
+        slots = buf.slots_of(batch.item_ids)
+        deltas = ops.mf_sgd_step(self.user_table, batch.user_rows,
+                                 buf.values, slots, batch.ratings,
+                                 self.lr, self.reg)
+        client.push(buf.uids, deltas)
+        if self.output_user_factors:
+            client.output((batch.user_rows,
+                           self.user_table[batch.user_rows].clone()))
+
+    def _with_negatives(self, data: MFBatch) -> MFBatch:
+        """rate-N hash-RNG negative items per positive, rating 0
+        (reference: negativeSampleRate)."""
+        n = len(data)
+        k = self.neg_rate
+        keys = (torch.arange(n * k, dtype=torch.int64,
+                             device=self.device)
+                + self._neg_counter * (n * k + 1))
+        self._neg_counter += 1
+        neg_items = ops.reference.splitmix64(keys + self._seed).abs() \
+            % max(1, self.n_items)
+        neg_users = data.user_rows.repeat_interleave(k)
+        return MFBatch(
+            torch.cat([data.user_rows, neg_users]),
+            torch.cat([data.item_ids, neg_items]),
+            torch.cat([data.ratings,
+                       torch.zeros(n * k, dtype=torch.float32,
+                                   device=self.device)]))
+
+    def close(self, client: ParameterServerClient) -> None:
+        if self.output_user_factors:
+            rows = torch.arange(self.user_table.shape[0],
+                                dtype=torch.int64, device=self.device)
+            client.output((rows, self.user_table.clone()))
+
+
+def epoch_iterator(batches: List[MFBatch], epochs: int
+                   ) -> Iterator[MFBatch]:
+    """Offline MF (``PSOfflineMatrixFactorization``): the worker sweeps its
+    buffered rating partition in epochs; user factors persist across
+    epochs in the worker-local table, item factors accumulate on the PS.
+    """
+    for _ in range(epochs):
+        for b in batches:
+            yield b

@@ -1,0 +1,98 @@
+"""Word2vec skip-gram with negative sampling (SGNS) on the PS.
+
+New capability (BASELINE.json configs[2]: 1M vocab x dim 300, embedding
+table sharded across 8x MI355X) — SURVEY.md K8.
+
+Model: ONE sharded table of 2*V rows — rows [0, V) are center ("input")
+embeddings, rows [V, 2V) are context ("output") embeddings; both sharded
+by id like any other parameter, so the pull/push machinery is unchanged.
+Per (center, context) positive pair the worker adds `neg` hash-RNG
+negative context ids (unigram sampling left to the data generator),
+pulls all touched rows, runs the fused SGNS step (dots + sigmoid + axpys,
+HIP kernel K8) and pushes deltas for center and context rows.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+
+from .. import ops
+from ..api import ParameterServerClient, WorkerLogic
+from .common import PullBuffer
+
+
+class SGNSBatch:
+    """Positive skip-gram pairs (center, context) of vocab ids [0, V)."""
+
+    __slots__ = ("centers", "contexts")
+
+    def __init__(self, centers: torch.Tensor, contexts: torch.Tensor):
+        self.centers = centers
+        self.contexts = contexts
+
+    def __len__(self) -> int:
+        return int(self.centers.numel())
+
+
+class SGNSWorkerLogic(WorkerLogic):
+    def __init__(self, vocab_size: int, dim: int, learning_rate: float,
+                 negatives: int = 5,
+                 pull_limit: Optional[int] = None, seed: int = 11):
+        self.V = int(vocab_size)
+        self.dim = int(dim)
+        self.lr = float(learning_rate)
+        self.neg = int(negatives)
+        self.pull_limit = pull_limit
+        self._seed = seed
+        self._counter = 0
+        # pending: (center_ids, ctx_table_ids per pair row, labels, buffer)
+        self._pending: List[Tuple[torch.Tensor, torch.Tensor, torch.Tensor,
+                                  PullBuffer]] = []
+
+    def batch_size(self, batch: SGNSBatch) -> int:
+        return len(batch)
+
+    def on_recv(self, data: SGNSBatch,
+                client: ParameterServerClient) -> None:
+        n = len(data)
+        dev = data.centers.device
+        k = self.neg
+        # expand each positive into (1 positive + k negatives) pair rows
+        centers = data.centers.repeat_interleave(k + 1)
+        keys = (torch.arange(n * k, dtype=torch.int64, device=dev)
+                + self._counter * (n * k + 1) + self._seed)
+        self._counter += 1
+        neg_ctx = ops.reference.splitmix64(keys).abs() % self.V
+        ctx = torch.empty(n * (k + 1), dtype=torch.int64, device=dev)
+        ctx.view(n, k + 1)[:, 0] = data.contexts
+        ctx.view(n, k + 1)[:, 1:] = neg_ctx.view(n, k)
+        labels = torch.zeros(n * (k + 1), dtype=torch.float32, device=dev)
+        labels.view(n, k + 1)[:, 0] = 1.0
+        ctx_table_ids = ctx + self.V  # context rows live at [V, 2V)
+        uids = torch.unique(torch.cat([centers, ctx_table_ids]))
+        buf = PullBuffer(uids, self.dim)
+        self._pending.append((centers, ctx_table_ids, labels, buf))
+        client.pull(uids)
+
+    def on_pull_recv(self, ids: torch.Tensor, values: torch.Tensor,
+                     client: ParameterServerClient) -> None:
+        still = []
+        for centers, ctx_ids, labels, buf in self._pending:
+            buf.offer(ids, values)
+            if buf.complete:
+                self._train(centers, ctx_ids, labels, buf, client)
+            else:
+                still.append((centers, ctx_ids, labels, buf))
+        self._pending = still
+
+    def _train(self, centers: torch.Tensor, ctx_ids: torch.Tensor,
+               labels: torch.Tensor, buf: PullBuffer,
+               client: ParameterServerClient) -> None:
+        c_slots = buf.slots_of(centers)
+        o_slots = buf.slots_of(ctx_ids)
+        # One buffer holds both center and context rows; slot spaces are
+        # disjoint ids so the deltas can share one accumulator.
+        c_deltas, o_deltas = ops.sgns_step(buf.values, c_slots, buf.values,
+                                           o_slots, labels, self.lr)
+        client.push(buf.uids, c_deltas + o_deltas)

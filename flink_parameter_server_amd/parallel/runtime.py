@@ -1,0 +1,315 @@
+"""The round-based async engine — the MI355X-native replacement for the
+reference's Flink job graph.
+
+Reference topology (``FlinkParameterServer.transform``):
+worker operator --(shuffle by hash(paramId))--> server operator
+--(iteration feedback edge by workerPartitionIndex)--> worker operator,
+free-running record flow, terminated by the ``iterationWaitTime``
+quiescence timeout.
+
+MI355X design: one process per GPU; each rank hosts one
+worker role AND one server-shard role.  The free-running record flow
+becomes *rounds*: each round every rank
+
+  1. delivers last round's pull answers to ``WorkerLogic.on_pull_recv``
+  2. consumes one data batch into ``WorkerLogic.on_recv`` (unless SSP-
+     stalled or out of data)
+  3. coalesces + buckets this round's pulls/pushes by destination shard
+  4. exchanges them (grouped send/recv alltoallv — RCCL over xGMI on GPU)
+  5. runs the server role: apply pushes, gather + answer pulls
+  6. exchanges the answers back (the reference's feedback edge)
+  7. allgathers a tiny control vector (clock, done, in-flight) — this
+     replaces both the SSP clock (new capability) and the reference's
+     quiescence-timeout termination with an explicit all-drained test.
+
+Asynchrony: an answer requested in round r arrives at the start of round
+r+1, and the push it provokes is applied in round r+1 — the worker never
+blocks on the server, which is the reference's async pull/push contract;
+staleness comes from round depth.  ``pull_limit`` bounds distinct
+outstanding pull ids exactly like ``WorkerLogic.addPullLimiter`` (excess
+pulls queue locally).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, Iterable, Iterator, List, Optional, Tuple
+
+import torch
+
+from ..api import ParameterServer, ParameterServerClient, ServerLogic, WorkerLogic
+from ..config import PSConfig
+from .partitioner import Partitioner, bucket_by_shard, make_partitioner
+from .transport import Transport, LocalTransport
+
+
+@dataclass
+class Metrics:
+    rounds: int = 0
+    samples: int = 0
+    pull_ids_sent: int = 0
+    push_ids_sent: int = 0
+    ssp_stalls: int = 0
+    bytes_sent: int = 0
+
+    def as_dict(self) -> Dict[str, int]:
+        return dict(self.__dict__)
+
+
+class _Client(ParameterServerClient):
+    def __init__(self, engine: "PSEngine"):
+        self._e = engine
+
+    def pull(self, ids: torch.Tensor) -> None:
+        if ids.numel():
+            self._e._pull_buf.append(ids.reshape(-1))
+
+    def push(self, ids: torch.Tensor, deltas: torch.Tensor) -> None:
+        if ids.numel():
+            self._e._push_buf.append((ids.reshape(-1), deltas))
+
+    def output(self, out: Any) -> None:
+        self._e.worker_outputs.append(out)
+
+
+class _PS(ParameterServer):
+    def __init__(self, engine: "PSEngine"):
+        self._e = engine
+
+    def answer_pull(self, ids: torch.Tensor, values: torch.Tensor,
+                    worker: int) -> None:
+        self._e._answer_buf[worker].append((ids, values))
+
+    def output(self, out: Any) -> None:
+        self._e.server_outputs.append(out)
+
+
+def _concat(parts: List[torch.Tensor]) -> Optional[torch.Tensor]:
+    if not parts:
+        return None
+    if len(parts) == 1:
+        return parts[0]
+    return torch.cat(parts)
+
+
+class PSEngine:
+    """Drives one rank's worker + server roles through rounds."""
+
+    def __init__(self, config: PSConfig, worker_logic: WorkerLogic,
+                 server_logic: ServerLogic, transport: Optional[Transport],
+                 part: Optional[Partitioner] = None,
+                 device: Optional[torch.device] = None):
+        self.cfg = config
+        self.worker_logic = worker_logic
+        self.server_logic = server_logic
+        self.transport = transport or LocalTransport()
+        self.rank = self.transport.rank
+        self.world = self.transport.world
+        self.part = part or make_partitioner(config.partitioner,
+                                             config.num_ids, self.world)
+        self.device = torch.device(device) if device is not None else \
+            torch.device("cpu")
+        self.client = _Client(self)
+        self.ps = _PS(self)
+
+        self.worker_outputs: List[Any] = []
+        self.server_outputs: List[Any] = []
+        self.metrics = Metrics()
+
+        # round state
+        self._pull_buf: List[torch.Tensor] = []
+        self._push_buf: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        self._answer_buf: List[List[Tuple[torch.Tensor, torch.Tensor]]] = \
+            [[] for _ in range(self.world)]
+        self._pull_queue: Optional[torch.Tensor] = None  # limiter overflow
+        self._outstanding = 0
+        self._pending_answers: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        self._clock = 0
+        self._min_clock = 0
+
+        limit = worker_logic.pull_limit or config.pull_limit
+        self._pull_limit = int(limit) if limit else None
+
+    # ------------------------------------------------------------------
+    def run(self, data: Iterable[Any], max_rounds: Optional[int] = None
+            ) -> Tuple[List[Any], List[Any]]:
+        """Run to drain (or ``max_rounds``); returns (worker_outputs,
+        server_outputs)."""
+        it = iter(data)
+        exhausted = False
+        r = 0
+        while True:
+            r += 1
+            if max_rounds is not None and r > max_rounds:
+                break
+            exhausted, done = self._round(it, exhausted)
+            if done:
+                break
+        self.worker_logic.close(self.client)
+        self.server_logic.close(self.ps)
+        return self.worker_outputs, self.server_outputs
+
+    # ------------------------------------------------------------------
+    def _round(self, it: Iterator[Any], exhausted: bool
+               ) -> Tuple[bool, bool]:
+        m = self.metrics
+        m.rounds += 1
+
+        # 1. deliver last round's answers
+        if self._pending_answers:
+            ids = _concat([a for a, _ in self._pending_answers])
+            vals = _concat([v for _, v in self._pending_answers])
+            self._pending_answers = []
+            self._outstanding -= int(ids.numel())
+            self.worker_logic.on_pull_recv(ids, vals, self.client)
+
+        # 2. consume data (unless SSP-stalled / exhausted / over limit)
+        stalled = (self.cfg.staleness is not None and
+                   self._clock - self._min_clock > self.cfg.staleness)
+        backpressure = (self._pull_limit is not None and
+                        self._queued_pulls() >= self._pull_limit)
+        if stalled:
+            m.ssp_stalls += 1
+        if not exhausted and not stalled and not backpressure:
+            try:
+                batch = next(it)
+            except StopIteration:
+                exhausted = True
+            else:
+                # a streaming source may yield None = "no batch ready this
+                # round"; the round proceeds without consuming.
+                if batch is not None:
+                    self._clock += 1
+                    m.samples += _batch_size(self.worker_logic, batch)
+                    self.worker_logic.on_recv(batch, self.client)
+
+        # 3. coalesce + bucket
+        pull_ids = self._take_pulls()
+        push_ids, push_deltas = self._take_pushes()
+        send = self._bucket(pull_ids, push_ids, push_deltas)
+        m.pull_ids_sent += int(pull_ids.numel()) if pull_ids is not None else 0
+        m.push_ids_sent += int(push_ids.numel()) if push_ids is not None else 0
+
+        # 4. exchange worker->server segments
+        dim = self.cfg.dim
+        recv = self.transport.alltoallv(send, [
+            (torch.int64, ()), (torch.int64, ()),
+            (torch.float32, (dim,))])
+        r_pull, r_push_ids, r_push_deltas = recv
+
+        # 5. server role: pushes first (fresher answers), then pulls
+        for s in range(self.world):
+            ids, deltas = r_push_ids[s], r_push_deltas[s]
+            if ids is not None and ids.numel():
+                self.server_logic.on_push_recv(ids, deltas, self.ps)
+        for s in range(self.world):
+            ids = r_pull[s]
+            if ids is not None and ids.numel():
+                self.server_logic.on_pull_recv(ids, s, self.ps)
+
+        # 6. exchange answers back (the feedback edge)
+        ans_ids = [_concat([a for a, _ in self._answer_buf[w]])
+                   for w in range(self.world)]
+        ans_vals = [_concat([v for _, v in self._answer_buf[w]])
+                    for w in range(self.world)]
+        self._answer_buf = [[] for _ in range(self.world)]
+        back = self.transport.alltoallv(
+            [ans_ids, ans_vals],
+            [(torch.int64, ()), (torch.float32, (dim,))])
+        for s in range(self.world):
+            if back[0][s] is not None and back[0][s].numel():
+                self._pending_answers.append((back[0][s], back[1][s]))
+
+        # 7. control: [clock, worker_done, exhausted].  An exhausted
+        # worker's clock freezes, so it must not hold the SSP min-clock
+        # back (it would deadlock the still-running workers).
+        worker_done = int(exhausted and self._outstanding == 0 and
+                          not self._pending_answers and
+                          self._queued_pulls() == 0 and
+                          not self._pull_buf and not self._push_buf)
+        ctrl = self.transport.allgather_int(
+            [self._clock, worker_done, int(exhausted)])
+        active = ctrl[:, 2] == 0
+        self._min_clock = int(ctrl[active, 0].min()) if bool(active.any()) \
+            else int(ctrl[:, 0].max())
+        done = bool(ctrl[:, 1].all())
+        return exhausted, done
+
+    # ------------------------------------------------------------------
+    def _queued_pulls(self) -> int:
+        n = self._pull_queue.numel() if self._pull_queue is not None else 0
+        return self._outstanding + n
+
+    def _take_pulls(self) -> Optional[torch.Tensor]:
+        parts = self._pull_buf
+        self._pull_buf = []
+        if self._pull_queue is not None:
+            parts = [self._pull_queue] + parts
+            self._pull_queue = None
+        ids = _concat(parts)
+        if ids is None:
+            return None
+        ids = torch.unique(ids)
+        if self._pull_limit is not None:
+            budget = self._pull_limit - self._outstanding
+            if budget <= 0:
+                self._pull_queue = ids
+                return None
+            if ids.numel() > budget:
+                self._pull_queue = ids[budget:]
+                ids = ids[:budget]
+        self._outstanding += int(ids.numel())
+        return ids
+
+    def _take_pushes(self) -> Tuple[Optional[torch.Tensor],
+                                    Optional[torch.Tensor]]:
+        parts = self._push_buf
+        self._push_buf = []
+        if not parts:
+            return None, None
+        ids = _concat([i for i, _ in parts])
+        deltas = _concat([d for _, d in parts])
+        return ids, deltas
+
+    def _bucket(self, pull_ids, push_ids, push_deltas):
+        """K10 bucket_pack: split each channel per destination shard."""
+        W = self.world
+        pulls: List[Optional[torch.Tensor]] = [None] * W
+        pids: List[Optional[torch.Tensor]] = [None] * W
+        pdel: List[Optional[torch.Tensor]] = [None] * W
+        if pull_ids is not None and pull_ids.numel():
+            if W == 1:
+                pulls[0] = pull_ids
+            else:
+                order, counts, _ = bucket_by_shard(pull_ids, self.part)
+                sorted_ids = pull_ids[order]
+                off = 0
+                for d in range(W):
+                    n = int(counts[d])
+                    if n:
+                        pulls[d] = sorted_ids[off:off + n]
+                    off += n
+        if push_ids is not None and push_ids.numel():
+            if W == 1:
+                pids[0], pdel[0] = push_ids, push_deltas
+            else:
+                order, counts, _ = bucket_by_shard(push_ids, self.part)
+                sorted_ids = push_ids[order]
+                sorted_del = push_deltas[order]
+                off = 0
+                for d in range(W):
+                    n = int(counts[d])
+                    if n:
+                        pids[d] = sorted_ids[off:off + n]
+                        pdel[d] = sorted_del[off:off + n]
+                    off += n
+        return [pulls, pids, pdel]
+
+
+def _batch_size(logic: WorkerLogic, batch: Any) -> int:
+    fn = getattr(logic, "batch_size", None)
+    if fn is not None:
+        return int(fn(batch))
+    try:
+        return len(batch)
+    except TypeError:
+        return 0

@@ -1,0 +1,207 @@
+"""Transports: how pull/push segments move between ranks.
+
+The reference's transport is Flink's streaming network runtime (Netty TCP
+record shuffle worker->server, the iteration feedback edge server->worker —
+SURVEY.md §5.8).  Here the same two edges are variable-length all-to-all
+exchanges:
+
+* ``TorchDistTransport`` — one process per GPU over ``torch.distributed``.
+  Counts are exchanged with an all_gather of an int64 matrix, payloads with
+  ``batch_isend_irecv`` (on the "nccl" backend = RCCL this lowers to grouped
+  ncclSend/ncclRecv over xGMI — the alltoallv shape SURVEY.md §5.8 calls
+  for; a ring collective would be bound by ONE of the 7 p2p links).  The
+  same code path runs on "gloo" for CPU multi-process tests.
+* ``LoopbackTransport`` — N ranks as N threads in one process, lockstep
+  rounds through shared mailboxes: the analog of the reference's Flink
+  MiniCluster local-mode tests, so every distributed test
+  can run single-process.
+* ``LocalTransport`` — world size 1; everything short-circuits.
+
+Self-addressed segments never touch the backend in any transport
+(same-device short-circuit — the 1/world fraction of traffic the reference
+gets for free when Flink co-locates subtasks).
+"""
+from __future__ import annotations
+
+import threading
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+
+
+class Transport:
+    rank: int = 0
+    world: int = 1
+
+    def alltoallv(self, parts: Sequence[Sequence[Optional[torch.Tensor]]],
+                  specs: Sequence[Tuple[torch.dtype, Tuple[int, ...]]]
+                  ) -> List[List[Optional[torch.Tensor]]]:
+        """parts[c][d] = tensor for channel c to rank d (dim 0 = items;
+        may be None = empty).  specs[c] = (dtype, trailing_shape) of
+        channel c — every rank must pass the same specs (the receiver
+        sizes its buffers from them, never from its own outgoing data).
+        Returns recv[c][s] = tensor received on channel c from rank s.
+        """
+        raise NotImplementedError
+
+    def allgather_int(self, vec: Sequence[int]) -> torch.Tensor:
+        """Gather a small int64 host vector from every rank -> (world, len)."""
+        raise NotImplementedError
+
+    def barrier(self) -> None:
+        pass
+
+
+class LocalTransport(Transport):
+    def __init__(self) -> None:
+        self.rank, self.world = 0, 1
+
+    def alltoallv(self, parts, specs):
+        return [[p[0]] for p in parts]
+
+    def allgather_int(self, vec):
+        return torch.tensor([list(vec)], dtype=torch.int64)
+
+
+class TorchDistTransport(Transport):
+    """torch.distributed-backed transport (nccl=RCCL on ROCm, or gloo)."""
+
+    def __init__(self, device: Optional[torch.device] = None) -> None:
+        import torch.distributed as dist
+        if not dist.is_initialized():
+            raise RuntimeError("torch.distributed is not initialized")
+        self.dist = dist
+        self.rank = dist.get_rank()
+        self.world = dist.get_world_size()
+        self.device = torch.device(device) if device is not None else (
+            torch.device("cuda", torch.cuda.current_device())
+            if dist.get_backend() == "nccl" else torch.device("cpu"))
+
+    def alltoallv(self, parts, specs):
+        dist = self.dist
+        C, W, me = len(parts), self.world, self.rank
+        dev = self.device
+
+        # 1) counts: one all_gather of int64[C*W] covers every channel.
+        send_counts = torch.zeros(C * W, dtype=torch.int64)
+        for c, chan in enumerate(parts):
+            for d, t in enumerate(chan):
+                send_counts[c * W + d] = 0 if t is None else t.shape[0]
+        gdev = send_counts.to(dev)
+        gathered = [torch.empty_like(gdev) for _ in range(W)]
+        dist.all_gather(gathered, gdev)
+        counts = torch.stack([g.cpu() for g in gathered])  # (W, C*W)
+
+        # 2) payloads: grouped isend/irecv (RCCL: grouped ncclSend/Recv).
+        recv: List[List[Optional[torch.Tensor]]] = [
+            [None] * W for _ in range(C)]
+        p2p = []
+        for c, chan in enumerate(parts):
+            dtype, trailing = specs[c]
+            for s in range(W):
+                n = int(counts[s, c * W + me])
+                if s == me:
+                    t = chan[me]
+                    recv[c][me] = t if t is not None else None
+                    continue
+                if n > 0:
+                    buf = torch.empty((n, *trailing), dtype=dtype, device=dev)
+                    recv[c][s] = buf
+                    p2p.append(dist.P2POp(dist.irecv, buf, s))
+            for d, t in enumerate(chan):
+                if d == me or t is None or t.shape[0] == 0:
+                    continue
+                t = t.contiguous()
+                if t.device != dev:
+                    t = t.to(dev)
+                p2p.append(dist.P2POp(dist.isend, t, d))
+        if p2p:
+            for work in dist.batch_isend_irecv(p2p):
+                work.wait()
+        return recv
+
+    def allgather_int(self, vec):
+        t = torch.tensor(list(vec), dtype=torch.int64, device=self.device)
+        out = [torch.empty_like(t) for _ in range(self.world)]
+        self.dist.all_gather(out, t)
+        return torch.stack([o.cpu() for o in out])
+
+    def barrier(self):
+        self.dist.barrier()
+
+
+def run_loopback(world: int, fn, timeout: Optional[float] = 120.0) -> list:
+    """Run ``fn(rank, transport)`` on ``world`` lockstep threads (the
+    in-process MiniCluster).  Returns the per-rank results; re-raises the
+    first rank exception."""
+    transports = LoopbackTransport.make_group(world)
+    results: List = [None] * world
+    errors: List = [None] * world
+
+    def target(r: int) -> None:
+        try:
+            results[r] = fn(r, transports[r])
+        except BaseException as e:  # noqa: BLE001
+            errors[r] = e
+            transports[r].shared.barrier.abort()
+
+    threads = [threading.Thread(target=target, args=(r,), daemon=True)
+               for r in range(world)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout)
+    for e in errors:
+        if e is not None and not isinstance(e, threading.BrokenBarrierError):
+            raise e
+    for t in threads:
+        if t.is_alive():
+            raise TimeoutError("loopback job did not finish")
+    return results
+
+
+class _LoopbackShared:
+    def __init__(self, world: int):
+        self.world = world
+        self.barrier = threading.Barrier(world)
+        self.mailbox: List[Optional[list]] = [None] * world
+        self.intbox: List[Optional[torch.Tensor]] = [None] * world
+
+
+class LoopbackTransport(Transport):
+    """In-process multi-rank transport (threads + lockstep barriers)."""
+
+    @staticmethod
+    def make_group(world: int) -> List["LoopbackTransport"]:
+        shared = _LoopbackShared(world)
+        return [LoopbackTransport(shared, r) for r in range(world)]
+
+    def __init__(self, shared: _LoopbackShared, rank: int):
+        self.shared = shared
+        self.rank = rank
+        self.world = shared.world
+
+    def alltoallv(self, parts, specs):
+        sh = self.shared
+        sh.mailbox[self.rank] = [list(chan) for chan in parts]
+        sh.barrier.wait()
+        C = len(parts)
+        recv: List[List[Optional[torch.Tensor]]] = [
+            [None] * self.world for _ in range(C)]
+        for s in range(self.world):
+            box = sh.mailbox[s]
+            for c in range(C):
+                recv[c][s] = box[c][self.rank]
+        sh.barrier.wait()  # everyone has read before boxes are overwritten
+        return recv
+
+    def allgather_int(self, vec):
+        sh = self.shared
+        sh.intbox[self.rank] = torch.tensor(list(vec), dtype=torch.int64)
+        sh.barrier.wait()
+        out = torch.stack(list(sh.intbox))
+        sh.barrier.wait()
+        return out
+
+    def barrier(self):
+        self.shared.barrier.wait()

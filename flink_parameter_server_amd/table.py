@@ -1,0 +1,123 @@
+"""HBM-resident sharded parameter table.
+
+This is the MI355X-native replacement for the reference's
+``server/SimplePSLogic.scala`` ``mutable.HashMap[Int, P]`` (SURVEY.md
+§2.1/§2.3 K1–K3): each shard keeps its parameter partition as one dense
+f32 slab in HBM (288 GB per GPU fits every BASELINE config dense — e.g.
+1e9 features x 2 f32 = 8 GB over 8 shards), initialized eagerly with the
+deterministic per-id hash init (semantically identical to the reference's
+lazy ``getOrElseUpdate(id, paramInit(id))`` because init is a pure function
+of the id), updated by HIP scatter-apply kernels.
+"""
+from __future__ import annotations
+
+from typing import Callable, Optional, Tuple
+
+import torch
+
+from . import ops
+from .parallel.partitioner import Partitioner
+
+
+class ShardedTable:
+    """One shard (this rank's partition) of the global parameter table.
+
+    optimizer:
+      "add"     — W[id] += delta           (reference paramUpdate = vectorSum /
+                                            '+'; the worker scales by lr)
+      "adagrad" — G += g^2; W -= lr*g/sqrt(G+eps)   (new capability)
+    """
+
+    def __init__(self, part: Partitioner, shard: int, dim: int,
+                 device: torch.device,
+                 init: Tuple[float, float] = (0.0, 0.0),
+                 optimizer: str = "add", lr: float = 0.0,
+                 eps: float = 1e-8, seed: int = 0):
+        self.part = part
+        self.shard = shard
+        self.dim = int(dim)
+        self.device = torch.device(device)
+        self.optimizer = optimizer
+        self.lr = float(lr)
+        self.eps = float(eps)
+        rows = part.shard_rows(shard)
+        self.rows = rows
+        self.data = torch.empty(rows, dim, dtype=torch.float32,
+                                device=self.device)
+        lo, hi = init
+        if lo == 0.0 and hi == 0.0:
+            self.data.zero_()
+        else:
+            # Deterministic per-global-id init: row r holds global id
+            # part.global_id(shard, r).  For ModPartitioner that is
+            # first_id=shard, stride=n_shards; for RangePartitioner
+            # first_id=shard*block, stride=1.
+            first = int(self.part.global_id(
+                self.shard, torch.zeros(1, dtype=torch.int64)).item()) if rows else 0
+            if rows >= 2:
+                second = int(self.part.global_id(
+                    self.shard, torch.ones(1, dtype=torch.int64)).item())
+                stride = second - first
+            else:
+                stride = 1
+            if rows:
+                ops.hash_init_rows(self.data, first, stride, lo, hi, seed)
+        if optimizer == "adagrad":
+            self.accum = torch.zeros(rows, dim, dtype=torch.float32,
+                                     device=self.device)
+        elif optimizer == "add":
+            self.accum = None
+        else:
+            raise ValueError(f"unknown optimizer {optimizer!r}")
+
+    # -- server-side ops (K3 gather / K1 sgd apply / K2 adagrad apply) ----
+
+    def gather(self, ids: torch.Tensor) -> torch.Tensor:
+        rows = self.part.local_row(ids)
+        return ops.gather(self.data, rows)
+
+    def apply(self, ids: torch.Tensor, deltas: torch.Tensor) -> None:
+        rows = self.part.local_row(ids)
+        if self.optimizer == "add":
+            ops.apply_add(self.data, rows, deltas)
+        else:
+            ops.apply_adagrad(self.data, self.accum, rows, deltas,
+                              self.lr, self.eps)
+
+    def dump(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        """(global_ids, values) of every row of the shard — the reference's
+        ``ParameterServerLogic.close`` model dump (unordered contract;
+        ours happens to be local-row order)."""
+        rows = torch.arange(self.rows, dtype=torch.int64)
+        gids = self.part.global_id(self.shard, rows)
+        return gids, self.data
+
+    def load(self, ids: torch.Tensor, values: torch.Tensor) -> None:
+        """Warm start (``transformWithModelLoad``): overwrite rows by id."""
+        rows = self.part.local_row(ids).to(self.device)
+        self.data.index_copy_(0, rows, values.to(self.device,
+                                                 self.data.dtype))
+
+
+class TableServerLogic:
+    """ServerLogic over a ShardedTable — the production default server.
+
+    Semantics parity: ``SimplePSLogic(paramInit, paramUpdate)`` with
+    paramInit = deterministic ranged-random init and paramUpdate = '+'
+    (or the Adagrad rule).
+    """
+
+    def __init__(self, table: ShardedTable, dump_on_close: bool = True):
+        self.table = table
+        self.dump_on_close = dump_on_close
+
+    def on_pull_recv(self, ids: torch.Tensor, worker: int, ps) -> None:
+        ps.answer_pull(ids, self.table.gather(ids), worker)
+
+    def on_push_recv(self, ids: torch.Tensor, deltas: torch.Tensor,
+                     ps) -> None:
+        self.table.apply(ids, deltas)
+
+    def close(self, ps) -> None:
+        if self.dump_on_close:
+            ps.output(self.table.dump())

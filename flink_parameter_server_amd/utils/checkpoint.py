@@ -1,0 +1,94 @@
+"""Checkpoint / model dump + load.
+
+Reference contract: a model dump is the set of every
+``(paramId, value)`` pair of each shard, unordered
+(``ParameterServerLogic.close`` -> PSOut stream); a load
+(``transformWithModelLoad``) feeds such pairs back through the partitioner,
+independent of the shard count that wrote them.
+
+Concrete format (one file per shard, little-endian):
+
+    magic   8 bytes  b"FPSCKPT1"
+    dim     int64
+    dtype   int64    (0 = f32)
+    count   int64
+    ids     count * int64
+    values  count * dim * f32
+
+Unordered on read; re-partitioned on load, so a checkpoint written with N
+shards loads onto M shards.  Optimizer state (Adagrad G) dumps as a second
+file with the same layout (suffix ``.accum``).
+"""
+from __future__ import annotations
+
+import os
+import struct
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..table import ShardedTable
+
+MAGIC = b"FPSCKPT1"
+
+
+def _write_one(path: str, ids: torch.Tensor, values: torch.Tensor) -> None:
+    ids = ids.contiguous().cpu()
+    values = values.contiguous().cpu().to(torch.float32)
+    count, dim = values.shape
+    with open(path, "wb") as f:
+        f.write(MAGIC)
+        f.write(struct.pack("<qqq", dim, 0, count))
+        f.write(ids.numpy().tobytes())
+        f.write(values.numpy().tobytes())
+
+
+def _read_one(path: str) -> Tuple[torch.Tensor, torch.Tensor]:
+    import numpy as np
+    with open(path, "rb") as f:
+        if f.read(8) != MAGIC:
+            raise ValueError(f"{path}: bad checkpoint magic")
+        dim, dtype, count = struct.unpack("<qqq", f.read(24))
+        if dtype != 0:
+            raise ValueError(f"{path}: unsupported dtype {dtype}")
+        ids = np.frombuffer(f.read(count * 8), dtype=np.int64)
+        vals = np.frombuffer(f.read(count * dim * 4), dtype=np.float32)
+    return (torch.from_numpy(ids.copy()),
+            torch.from_numpy(vals.copy()).reshape(count, dim))
+
+
+def shard_path(directory: str, shard: int, accum: bool = False) -> str:
+    suffix = ".accum" if accum else ""
+    return os.path.join(directory, f"shard_{shard:05d}.fps{suffix}")
+
+
+def save_shard(directory: str, table: ShardedTable) -> None:
+    os.makedirs(directory, exist_ok=True)
+    ids, values = table.dump()
+    _write_one(shard_path(directory, table.shard), ids, values)
+    if table.accum is not None:
+        _write_one(shard_path(directory, table.shard, accum=True),
+                   ids, table.accum)
+
+
+def load_shard(directory: str, table: ShardedTable) -> None:
+    """Load this shard's rows from a checkpoint written with ANY shard
+    count: reads every file, keeps the ids this shard owns."""
+    files = sorted(f for f in os.listdir(directory) if f.endswith(".fps"))
+    if not files:
+        raise FileNotFoundError(f"no .fps files in {directory}")
+    for fname in files:
+        ids, values = _read_one(os.path.join(directory, fname))
+        mine = table.part.shard_of(ids) == table.shard
+        if bool(mine.any()):
+            table.load(ids[mine], values[mine])
+        if table.accum is not None:
+            apath = os.path.join(directory, fname + ".accum")
+            if os.path.exists(apath):
+                aids, avals = _read_one(apath)
+                amine = table.part.shard_of(aids) == table.shard
+                if bool(amine.any()):
+                    rows = table.part.local_row(aids[amine]).to(
+                        table.device)
+                    table.accum.index_copy_(
+                        0, rows, avals[amine].to(table.device))

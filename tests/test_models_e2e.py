@@ -1,0 +1,169 @@
+"""End-to-end convergence tests for the model logics on the local engine
+(single process, 1 rank).  Async training is nondeterministic, so these
+are threshold tests, not golden-numerics tests — the reference's own test
+philosophy."""
+import torch
+
+from flink_parameter_server_amd import (PSConfig, PSEngine, ShardedTable,
+                                        TableServerLogic)
+from flink_parameter_server_amd.parallel.partitioner import ModPartitioner
+from flink_parameter_server_amd.models.mf import (OnlineMFWorkerLogic,
+                                                  epoch_iterator)
+from flink_parameter_server_amd.models.pa import PAWorkerLogic
+from flink_parameter_server_amd.models.lr import LRWorkerLogic
+from flink_parameter_server_amd.models.sgns import SGNSWorkerLogic
+from flink_parameter_server_amd.utils import synth
+
+
+def _mf_loss(wl, table, batches):
+    tot, n = 0.0, 0
+    for b in batches:
+        pred = (wl.user_table[b.user_rows] * table.data[b.item_ids]).sum(1)
+        tot += float(((pred - b.ratings) ** 2).sum())
+        n += len(b)
+    return tot / n
+
+
+def test_online_mf_converges():
+    n_users, n_items, rank = 60, 50, 4
+    batches, _, _ = synth.planted_mf_ratings(n_users, n_items, rank,
+                                             3000, 128, seed=3)
+    part = ModPartitioner(n_items, 1)
+    table = ShardedTable(part, 0, rank, "cpu", init=(-0.1, 0.1))
+    wl = OnlineMFWorkerLogic(n_users, rank, learning_rate=0.05,
+                             range_min=-0.1, range_max=0.1)
+    eng = PSEngine(PSConfig(num_ids=n_items, dim=rank), wl,
+                   TableServerLogic(table), None, part)
+    l0 = _mf_loss(wl, table, batches)
+    eng.run(epoch_iterator(batches, 8))
+    l1 = _mf_loss(wl, table, batches)
+    assert l1 < 0.5 * l0
+
+
+def test_online_mf_with_negative_sampling_runs():
+    n_users, n_items, rank = 30, 25, 4
+    part = ModPartitioner(n_items, 1)
+    table = ShardedTable(part, 0, rank, "cpu", init=(-0.1, 0.1))
+    wl = OnlineMFWorkerLogic(n_users, rank, learning_rate=0.05,
+                             negative_sample_rate=2, n_items=n_items)
+    eng = PSEngine(PSConfig(num_ids=n_items, dim=rank), wl,
+                   TableServerLogic(table), None, part)
+    data = list(synth.mf_ratings(n_users, n_items, 500, 100, seed=1))
+    eng.run(data)
+    assert eng.metrics.samples == 500
+    # negatives tripled the rating rows internally; table was touched
+    assert float(table.data.abs().sum()) > 0
+
+
+def test_pa_binary_learns_planted_teacher():
+    n_feat = 200
+    train = list(synth.sparse_instances(n_feat, 10, 4000, 200,
+                                        task="binary", seed=5, teacher_seed=5))
+    part = ModPartitioner(n_feat, 1)
+    table = ShardedTable(part, 0, 1, "cpu", init=(0.0, 0.0))
+    wl = PAWorkerLogic("PA-I", C=1.0)
+    eng = PSEngine(PSConfig(num_ids=n_feat, dim=1), wl,
+                   TableServerLogic(table), None, part)
+    eng.run(train)
+    # accuracy of the final model on fresh data from the same teacher
+    test = list(synth.sparse_instances(n_feat, 10, 1000, 1000, task="binary",
+                                       seed=6, teacher_seed=5))[0]
+    w = table.data.reshape(-1)
+    seg = torch.repeat_interleave(
+        torch.arange(len(test)), test.indptr[1:] - test.indptr[:-1])
+    margins = torch.zeros(len(test))
+    margins.index_add_(0, seg, w[test.feat_ids] * test.x_vals)
+    acc = float((torch.sign(margins) == test.labels).float().mean())
+    assert acc > 0.8, acc
+
+
+def test_pa_emits_predictions_for_unlabeled():
+    n_feat = 50
+    data = list(synth.sparse_instances(n_feat, 5, 300, 100, task="binary",
+                                       unlabeled_frac=0.3, seed=7))
+    part = ModPartitioner(n_feat, 1)
+    table = ShardedTable(part, 0, 1, "cpu")
+    wl = PAWorkerLogic("PA-I")
+    eng = PSEngine(PSConfig(num_ids=n_feat, dim=1), wl,
+                   TableServerLogic(table), None, part)
+    wouts, _ = eng.run(data)
+    n_pred = sum(p.numel() for _, p in wouts)
+    assert n_pred > 0
+
+
+def test_pa_multiclass_learns():
+    n_feat, n_cls = 150, 4
+    train = list(synth.sparse_instances(n_feat, 10, 4000, 200,
+                                        task="multiclass", n_classes=n_cls,
+                                        seed=8, teacher_seed=8))
+    part = ModPartitioner(n_feat, 1)
+    table = ShardedTable(part, 0, n_cls, "cpu")
+    wl = PAWorkerLogic("PA-I", n_classes=n_cls)
+    eng = PSEngine(PSConfig(num_ids=n_feat, dim=n_cls), wl,
+                   TableServerLogic(table), None, part)
+    eng.run(train)
+    test = list(synth.sparse_instances(n_feat, 10, 1000, 1000,
+                                       task="multiclass", n_classes=n_cls,
+                                       seed=9, teacher_seed=8))[0]
+    W = table.data
+    seg = torch.repeat_interleave(
+        torch.arange(len(test)), test.indptr[1:] - test.indptr[:-1])
+    scores = torch.zeros(len(test), n_cls)
+    scores.index_add_(0, seg, W[test.feat_ids] * test.x_vals.unsqueeze(1))
+    acc = float((scores.argmax(1) == test.labels).float().mean())
+    assert acc > 0.6, acc
+
+
+def test_lr_adagrad_converges():
+    n_feat = 200
+    train = list(synth.sparse_instances(n_feat, 10, 4000, 200, task="lr",
+                                        seed=10, teacher_seed=10))
+    part = ModPartitioner(n_feat, 1)
+    table = ShardedTable(part, 0, 1, "cpu", optimizer="adagrad", lr=0.5)
+    wl = LRWorkerLogic()
+    eng = PSEngine(PSConfig(num_ids=n_feat, dim=1), wl,
+                   TableServerLogic(table), None, part)
+    eng.run(train)
+    test = list(synth.sparse_instances(n_feat, 10, 1000, 1000, task="lr",
+                                       seed=11, teacher_seed=10))[0]
+    w = table.data.reshape(-1)
+    seg = torch.repeat_interleave(
+        torch.arange(len(test)), test.indptr[1:] - test.indptr[:-1])
+    margins = torch.zeros(len(test))
+    margins.index_add_(0, seg, w[test.feat_ids] * test.x_vals)
+    acc = float(((margins >= 0).float() == test.labels).float().mean())
+    assert acc > 0.8, acc
+    assert float(table.accum.sum()) > 0  # adagrad state actually used
+
+
+def test_sgns_objective_decreases():
+    V, dim = 100, 16
+    part = ModPartitioner(2 * V, 1)
+    table = ShardedTable(part, 0, dim, "cpu", init=(-0.05, 0.05))
+    wl = SGNSWorkerLogic(V, dim, learning_rate=0.05, negatives=3)
+    eng = PSEngine(PSConfig(num_ids=2 * V, dim=dim), wl,
+                   TableServerLogic(table), None, part)
+    pairs = list(synth.sgns_pairs(V, 2000, 200, seed=12))
+    g = torch.Generator().manual_seed(99)
+    eval_negs = torch.randint(0, V, (2000 * 3,), generator=g)
+
+    def objective():
+        """Full SGNS objective: -log s(v.u_pos) - sum log s(-v.u_neg) on a
+        fixed eval negative sample."""
+        tot, n = 0.0, 0
+        for bi, b in enumerate(pairs):
+            v = table.data[b.centers]
+            u = table.data[b.contexts + V]
+            s = torch.sigmoid((v * u).sum(1)).clamp(1e-6, 1 - 1e-6)
+            nb = eval_negs[bi * 600:(bi + 1) * 600].reshape(-1, 3)[:len(b)]
+            un = table.data[nb + V]
+            sn = torch.sigmoid(-(v.unsqueeze(1) * un).sum(2)).clamp(
+                1e-6, 1 - 1e-6)
+            tot += float(-torch.log(s).sum()) + float(-torch.log(sn).sum())
+            n += len(b)
+        return tot / n
+
+    before = objective()
+    eng.run(pairs * 3)
+    after = objective()
+    assert after < before * 0.9
