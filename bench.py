@@ -1,0 +1,234 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: SGD matrix factorization rank-32 on
+MovieLens-1M-shaped synthetic data (BASELINE.json headline metric:
+samples/sec whole node + pull/push ops/sec).
+
+Single GPU:   python bench.py --steps 50 --warmup 10
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One process per GPU (reads RANK/LOCAL_RANK/WORLD_SIZE from the env); each
+rank hosts one MF worker (its user partition, worker-local) and one item
+shard (mod-partitioned).  A "step" is one engine round: in steady state
+every round consumes one batch of ratings per rank (pull item rows ->
+fused MF SGD kernel -> push deltas -> scatter-apply), so whole-node
+samples/sec = N * batch * steps / elapsed.  Warmup rounds fill the
+pull/answer pipeline; the timed region is bracketed by barrier +
+torch.cuda.synchronize on both sides and the max over ranks is reported.
+
+Other configs (BASELINE.json configs[2..4]) via --model:
+  mf32 (default) | mf128 (+ --staleness for SSP) | sgns | lr
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from flink_parameter_server_amd import (PSConfig, PSEngine, ShardedTable,  # noqa: E402
+                                        TableServerLogic)
+from flink_parameter_server_amd.parallel.partitioner import ModPartitioner  # noqa: E402
+from flink_parameter_server_amd.parallel.transport import (  # noqa: E402
+    LocalTransport, TorchDistTransport)
+from flink_parameter_server_amd.models.mf import MFBatch, OnlineMFWorkerLogic  # noqa: E402
+from flink_parameter_server_amd.models.lr import LRWorkerLogic  # noqa: E402
+from flink_parameter_server_amd.models.pa import SparseBatch  # noqa: E402
+from flink_parameter_server_amd.models.sgns import SGNSBatch, SGNSWorkerLogic  # noqa: E402
+
+
+def cycle(batches):
+    while True:
+        for b in batches:
+            yield b
+
+
+def make_mf(args, rank, world, device):
+    """MovieLens-1M shape: 6040 users x 3952 items (scaled by --scale);
+    rank-32 factors (or 128)."""
+    n_users = int(6040 * args.scale)
+    n_items = int(3952 * args.scale)
+    k = 128 if args.model == "mf128" else 32
+    users_local = (n_users + world - 1) // world
+    part = ModPartitioner(n_items, world)
+    table = ShardedTable(part, rank, k, device, init=(-0.01, 0.01))
+    wl = OnlineMFWorkerLogic(users_local, k, learning_rate=0.01,
+                             reg=0.001, range_min=-0.01, range_max=0.01,
+                             device=device, seed=rank + 1)
+    g = torch.Generator().manual_seed(1234 + rank)
+    batches = []
+    for _ in range(args.n_batches):
+        users = torch.randint(0, users_local, (args.batch,), generator=g)
+        u = torch.rand(args.batch, generator=g)
+        items = (n_items * u.pow(3.0)).long().clamp_(0, n_items - 1)
+        ratings = torch.rand(args.batch, generator=g) * 4.0 + 1.0
+        batches.append(MFBatch(users.to(device), items.to(device),
+                               ratings.to(device)))
+    cfg = PSConfig(num_ids=n_items, dim=k, staleness=args.staleness)
+    meta = {"model": f"SGD-MF rank{k} MovieLens-1M-shaped" +
+                     (f" x{args.scale:g}" if args.scale != 1 else ""),
+            "n_users": n_users, "n_items": n_items, "factors": k}
+    return cfg, part, table, wl, batches, meta
+
+
+def make_sgns(args, rank, world, device):
+    """configs[2]: 1M vocab, dim 300, table (2V rows) sharded."""
+    V, dim = 1_000_000, 300
+    part = ModPartitioner(2 * V, world)
+    table = ShardedTable(part, rank, dim, device, init=(-0.01, 0.01))
+    wl = SGNSWorkerLogic(V, dim, learning_rate=0.025, negatives=5,
+                         seed=rank + 3)
+    g = torch.Generator().manual_seed(99 + rank)
+    batches = []
+    for _ in range(args.n_batches):
+        u = torch.rand(args.batch, 2, generator=g)
+        ids = (V * u.pow(3.0)).long().clamp_(0, V - 1)
+        batches.append(SGNSBatch(ids[:, 0].to(device), ids[:, 1].to(device)))
+    cfg = PSConfig(num_ids=2 * V, dim=dim)
+    meta = {"model": "word2vec-SGNS 1M-vocab dim300 neg5", "vocab": V,
+            "dim": dim}
+    return cfg, part, table, wl, batches, meta
+
+
+def make_lr(args, rank, world, device):
+    """configs[3]: sparse LR, hashed features, async Adagrad."""
+    n_feat = args.lr_features
+    nnz = 64
+    part = ModPartitioner(n_feat, world)
+    table = ShardedTable(part, rank, 1, device, optimizer="adagrad",
+                         lr=0.05)
+    wl = LRWorkerLogic()
+    g = torch.Generator().manual_seed(7 + rank)
+    batches = []
+    for _ in range(args.n_batches):
+        B = args.batch
+        u = torch.rand(B * nnz, generator=g)
+        feats = (n_feat * u.pow(2.0)).long().clamp_(0, n_feat - 1)
+        vals = torch.randn(B * nnz, generator=g)
+        indptr = torch.arange(0, B * nnz + 1, nnz, dtype=torch.int64)
+        labels = (torch.rand(B, generator=g) < 0.5).float()
+        batches.append(SparseBatch(indptr.to(device), feats.to(device),
+                                   vals.to(device), labels.to(device)))
+    cfg = PSConfig(num_ids=n_feat, dim=1)
+    meta = {"model": f"sparse-LR {n_feat} hashed features adagrad",
+            "nnz_per_instance": nnz}
+    return cfg, part, table, wl, batches, meta
+
+
+MAKERS = {"mf32": make_mf, "mf128": make_mf, "sgns": make_sgns,
+          "lr": make_lr}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch", type=int, default=0,
+                    help="samples per step per GPU (0 = model default)")
+    ap.add_argument("--model", choices=list(MAKERS), default="mf32")
+    ap.add_argument("--scale", type=float, default=1.0,
+                    help="scale factor on the MovieLens-1M shape")
+    ap.add_argument("--staleness", type=int, default=None)
+    ap.add_argument("--lr-features", type=int, default=10**8,
+                    help="feature count for --model lr (1e9 on 8 GPUs)")
+    ap.add_argument("--n-batches", type=int, default=8,
+                    help="distinct pre-generated batches cycled")
+    ap.add_argument("--device", default=None)
+    args = ap.parse_args()
+
+    world_env = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    world = max(world_env, 1)
+    on_gpu = torch.cuda.is_available() and args.device != "cpu"
+    device = torch.device(args.device) if args.device else (
+        torch.device("cuda", local_rank) if on_gpu
+        else torch.device("cpu"))
+    if device.type == "cuda":
+        torch.cuda.set_device(device)
+
+    if args.batch == 0:
+        defaults = {"mf32": 1 << 20, "mf128": 1 << 19, "sgns": 1 << 17,
+                    "lr": 1 << 17}
+        args.batch = defaults[args.model] if device.type == "cuda" \
+            else 4096
+
+    if world > 1:
+        import torch.distributed as dist
+        backend = "nccl" if device.type == "cuda" else "gloo"
+        dist.init_process_group(backend)
+        transport = TorchDistTransport(device=device)
+    else:
+        transport = LocalTransport()
+
+    cfg, part, table, wl, batches, meta = MAKERS[args.model](
+        args, rank, world, device)
+    eng = PSEngine(cfg, wl, TableServerLogic(table, dump_on_close=False),
+                   transport, part, device=device)
+    it = cycle(batches)
+
+    def sync():
+        transport.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
+    # warmup fills the pull/answer pipeline
+    for _ in range(args.warmup):
+        eng._round(it, False)
+    sync()
+    m0 = eng.metrics.as_dict()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        eng._round(it, False)
+    sync()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    # max over ranks
+    stats = transport.allgather_int(
+        [int(elapsed * 1e9), eng.metrics.samples - m0["samples"],
+         eng.metrics.pull_ids_sent - m0["pull_ids_sent"],
+         eng.metrics.push_ids_sent - m0["push_ids_sent"]])
+    elapsed = float(stats[:, 0].max()) / 1e9
+    samples = int(stats[:, 1].sum())
+    pp_ops = int(stats[:, 2].sum() + stats[:, 3].sum())
+
+    if rank == 0:
+        k = meta.get("factors", cfg.dim)
+        result = {
+            "metric": "samples/sec (whole node) SGD MF rank-32"
+                      if args.model == "mf32" else
+                      f"samples/sec (whole node) {meta['model']}",
+            "value": samples / elapsed,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "f32",
+            "data": "synthetic",
+            "pull_push_ops_per_sec": pp_ops / elapsed,
+            "config": {**meta,
+                       "global_batch": args.batch * world,
+                       "batch_per_gpu": args.batch,
+                       "staleness": args.staleness,
+                       "parallelism": f"worker-dp{world} x ps-shard{world}",
+                       "device": str(device)},
+        }
+        print(json.dumps(result))
+
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

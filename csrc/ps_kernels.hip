@@ -1,0 +1,587 @@
+// MI355X (gfx950, CDNA4) kernels for the parameter-server framework.
+//
+// These are the MI355X-native replacements for the reference's JVM hot
+// loops:
+//   gather             <- SimplePSLogic.onPullRecv answer assembly  (K3)
+//   apply_add          <- SimplePSLogic.onPushRecv, paramUpdate='+' (K1)
+//   apply_adagrad      <- new capability (configs[3])               (K2)
+//   hash_init[_rows]   <- RangedRandomFactorInitializer             (K6)
+//   mf_sgd_step        <- SGDUpdater.delta per rating               (K4)
+//   pa_step/_mc        <- passive/aggressive/algorithm margin+tau   (K7)
+//   lr_step            <- new capability: sparse logistic reg       (K9)
+//   sgns_step          <- new capability: word2vec SGNS             (K8)
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//  * wave = 64 lanes, hard-coded; sub-wave lane groups of GP lanes
+//    (power of two <= 64) process one sample each, reducing dots with
+//    __shfl_xor across the group.
+//  * memory-bound throughout: all row traffic is vectorized float4
+//    where dim % 4 == 0 (Guideline 13); grids are capped and
+//    grid-strided (Guideline 11).
+//  * duplicate keys inside a push batch accumulate with device-scope f32
+//    atomics (correct for '+' semantics; Adagrad instead runs on a
+//    sorted batch with one segment leader per distinct row, because its
+//    update is non-linear in the aggregated gradient).
+//  * numerics: f32 master values end to end; the semantics oracle is
+//    ops/reference.py and tests compare against it elementwise.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#define WAVE 64
+#define CHECK_DEV(x) TORCH_CHECK((x).is_cuda(), #x " must be a device tensor")
+#define CHECK_CONT(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+
+static inline int grid_for(int64_t work, int block) {
+  int64_t g = (work + block - 1) / block;
+  const int64_t cap = 2048;  // 256 CUs x 8 blocks; grid-stride the rest
+  return (int)(g < cap ? (g > 0 ? g : 1) : cap);
+}
+
+// ---------------------------------------------------------------------
+// splitmix64 — bit-identical to ops/reference.py::splitmix64
+// ---------------------------------------------------------------------
+__device__ __forceinline__ uint64_t splitmix64(uint64_t z) {
+  z += 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return z ^ (z >> 31);
+}
+
+__device__ __forceinline__ float hash_uniform(uint64_t key, float lo,
+                                              float hi) {
+  float u24 = (float)(splitmix64(key) >> 40);       // [0, 2^24)
+  return lo + (hi - lo) * (u24 * (1.0f / 16777216.0f));
+}
+
+// ---------------------------------------------------------------------
+// K6: deterministic per-id init
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_hash_init_rows(
+    float* __restrict__ out, int64_t rows, int64_t dim, int64_t first_id,
+    int64_t stride, float lo, float hi, int64_t seed) {
+  int64_t total = rows * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t r = i / dim, c = i - r * dim;
+    uint64_t key = (uint64_t)((first_id + r * stride) * dim + c +
+                              seed * 1000003);
+    out[i] = hash_uniform(key, lo, hi);
+  }
+}
+
+__global__ __launch_bounds__(256) void k_hash_init_ids(
+    float* __restrict__ out, const int64_t* __restrict__ ids, int64_t n,
+    int64_t dim, float lo, float hi, int64_t seed) {
+  int64_t total = n * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t r = i / dim, c = i - r * dim;
+    uint64_t key = (uint64_t)(ids[r] * dim + c + seed * 1000003);
+    out[i] = hash_uniform(key, lo, hi);
+  }
+}
+
+// ---------------------------------------------------------------------
+// K3: gather — values[j] = table[rows[j]].  One wave-fraction per row,
+// float4 path when dim % 4 == 0 (16 B/lane coalesced).
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_gather_vec4(
+    const float4* __restrict__ table, const int64_t* __restrict__ rows,
+    float4* __restrict__ out, int64_t n, int64_t dim4) {
+  int64_t total = n * dim4;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = i / dim4, c = i - j * dim4;
+    out[i] = table[rows[j] * dim4 + c];
+  }
+}
+
+__global__ __launch_bounds__(256) void k_gather(
+    const float* __restrict__ table, const int64_t* __restrict__ rows,
+    float* __restrict__ out, int64_t n, int64_t dim) {
+  int64_t total = n * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = i / dim, c = i - j * dim;
+    out[i] = table[rows[j] * dim + c];
+  }
+}
+
+// ---------------------------------------------------------------------
+// K1: apply_add — table[rows[j]] += deltas[j]; duplicate rows accumulate
+// via device-scope f32 atomics (L2).
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_apply_add(
+    float* __restrict__ table, const int64_t* __restrict__ rows,
+    const float* __restrict__ deltas, int64_t n, int64_t dim) {
+  int64_t total = n * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = i / dim, c = i - j * dim;
+    atomicAdd(&table[rows[j] * dim + c], deltas[i]);
+  }
+}
+
+// ---------------------------------------------------------------------
+// K2: apply_adagrad on a SORTED batch (host sorts rows, gathers grads).
+// Element (j, c) acts only if j is its row's segment leader; the leader
+// walks its segment, pre-aggregating duplicate-row gradients, then does
+// ONE read-modify-write of (G, W) — the contract of
+// ops/reference.py::apply_adagrad.
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_apply_adagrad_sorted(
+    float* __restrict__ table, float* __restrict__ accum,
+    const int64_t* __restrict__ srows, const float* __restrict__ sgrads,
+    int64_t n, int64_t dim, float lr, float eps) {
+  int64_t total = n * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = i / dim, c = i - j * dim;
+    int64_t row = srows[j];
+    if (j > 0 && srows[j - 1] == row) continue;  // not the leader
+    float g = sgrads[i];
+    for (int64_t k = j + 1; k < n && srows[k] == row; ++k)
+      g += sgrads[k * dim + c];
+    float G = accum[row * dim + c] + g * g;
+    accum[row * dim + c] = G;
+    table[row * dim + c] -= lr * g * rsqrtf(G + eps);
+  }
+}
+
+// ---------------------------------------------------------------------
+// Sub-wave group reduction: sum over a lane group of GP lanes (GP a
+// power of two <= 64).
+// ---------------------------------------------------------------------
+template <int GP>
+__device__ __forceinline__ float group_sum(float v) {
+#pragma unroll
+  for (int off = GP / 2; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, GP);
+  return v;
+}
+
+// ---------------------------------------------------------------------
+// K4: fused MF SGD step.  One lane group per rating:
+//   e  = r - u.v      (v = pulled item snapshot)
+//   u += lr*(e*v - reg*u)         (atomic: duplicate users may race,
+//                                  async-parity semantics)
+//   item_deltas[slot] += lr*(e*u0 - reg*v)   (atomic accumulate; u0 is
+//                                  the value read for the dot)
+// ---------------------------------------------------------------------
+template <int GP>
+__global__ __launch_bounds__(256) void k_mf_sgd_step(
+    float* __restrict__ user_table, const int64_t* __restrict__ user_rows,
+    const float* __restrict__ item_vecs,
+    const int64_t* __restrict__ item_slots,
+    const float* __restrict__ ratings, float* __restrict__ item_deltas,
+    int64_t n, int64_t dim, float lr, float reg) {
+  constexpr int EPL = 8;           // max elements per lane
+  const int lane = threadIdx.x % GP;
+  const int64_t groups_per_block = blockDim.x / GP;
+  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
+                     threadIdx.x / GP;
+  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
+  for (int64_t g = g0; g < n; g += gstride) {
+    const int64_t urow = user_rows[g] * dim;
+    const int64_t vrow = item_slots[g] * dim;
+    float u[EPL], v[EPL];
+    float dot = 0.f;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) {
+      int c = lane + e * GP;
+      if (c < dim) {
+        u[e] = user_table[urow + c];
+        v[e] = item_vecs[vrow + c];
+        dot += u[e] * v[e];
+      }
+    }
+    float e_ = ratings[g] - group_sum<GP>(dot);
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) {
+      int c = lane + e * GP;
+      if (c < dim) {
+        atomicAdd(&user_table[urow + c], lr * (e_ * v[e] - reg * u[e]));
+        atomicAdd(&item_deltas[vrow + c], lr * (e_ * u[e] - reg * v[e]));
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// K8: fused SGNS step.  One lane group per (center, context) pair row:
+//   s = sigmoid(v.u); g = lr*(label - s)
+//   center_deltas[cslot] += g*u ; ctx_deltas[oslot] += g*v   (atomic)
+// ---------------------------------------------------------------------
+template <int GP>
+__global__ __launch_bounds__(256) void k_sgns_step(
+    const float* __restrict__ center_vecs,
+    const int64_t* __restrict__ center_slots,
+    const float* __restrict__ ctx_vecs,
+    const int64_t* __restrict__ ctx_slots,
+    const float* __restrict__ labels, float* __restrict__ center_deltas,
+    float* __restrict__ ctx_deltas, int64_t n, int64_t dim, float lr) {
+  constexpr int EPL = 8;
+  const int lane = threadIdx.x % GP;
+  const int64_t groups_per_block = blockDim.x / GP;
+  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
+                     threadIdx.x / GP;
+  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
+  for (int64_t g = g0; g < n; g += gstride) {
+    const int64_t crow = center_slots[g] * dim;
+    const int64_t orow = ctx_slots[g] * dim;
+    float v[EPL], u[EPL];
+    float dot = 0.f;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) {
+      int c = lane + e * GP;
+      if (c < dim) {
+        v[e] = center_vecs[crow + c];
+        u[e] = ctx_vecs[orow + c];
+        dot += v[e] * u[e];
+      }
+    }
+    float d = group_sum<GP>(dot);
+    float s = 1.0f / (1.0f + __expf(-d));
+    float gg = lr * (labels[g] - s);
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) {
+      int c = lane + e * GP;
+      if (c < dim) {
+        atomicAdd(&center_deltas[crow + c], gg * u[e]);
+        atomicAdd(&ctx_deltas[orow + c], gg * v[e]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// K7 (binary) / K9: PA and LR share the sparse-margin skeleton.
+// One wave per instance; lanes stride the instance's nnz.
+// mode: 0 = PA, 1 = PA-I, 2 = PA-II, 3 = logistic regression.
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_sparse_linear_step(
+    const int64_t* __restrict__ indptr,
+    const int64_t* __restrict__ feat_slots,
+    const float* __restrict__ x_vals, const float* __restrict__ labels,
+    const float* __restrict__ weights, float* __restrict__ margins,
+    float* __restrict__ deltas, int64_t B, int mode, float C) {
+  const int lane = threadIdx.x % WAVE;
+  const int64_t waves_per_block = blockDim.x / WAVE;
+  const int64_t w0 = (int64_t)blockIdx.x * waves_per_block +
+                     threadIdx.x / WAVE;
+  const int64_t wstride = (int64_t)gridDim.x * waves_per_block;
+  for (int64_t i = w0; i < B; i += wstride) {
+    const int64_t lo = indptr[i], hi = indptr[i + 1];
+    float m = 0.f, n2 = 0.f;
+    for (int64_t j = lo + lane; j < hi; j += WAVE) {
+      float x = x_vals[j];
+      m += weights[feat_slots[j]] * x;
+      n2 += x * x;
+    }
+    m = group_sum<WAVE>(m);
+    n2 = group_sum<WAVE>(n2);
+    if (lane == 0) margins[i] = m;
+    const float y = labels[i];
+    if (isnan(y)) continue;  // unlabeled: predict only
+    float coeff;             // delta_f = coeff * x_f
+    if (mode == 3) {         // logistic: push raw gradient (p - y)
+      float p = 1.0f / (1.0f + __expf(-m));
+      coeff = p - y;
+      if (coeff == 0.f) continue;
+    } else {
+      float loss = 1.0f - y * m;
+      if (loss <= 0.f) continue;
+      float sn = n2 > 1e-12f ? n2 : 1e-12f;
+      float tau;
+      if (mode == 0) tau = loss / sn;
+      else if (mode == 1) { tau = loss / sn; tau = tau > C ? C : tau; }
+      else tau = loss / (sn + 1.0f / (2.0f * C));
+      coeff = tau * y;
+    }
+    for (int64_t j = lo + lane; j < hi; j += WAVE)
+      atomicAdd(&deltas[feat_slots[j]], coeff * x_vals[j]);
+  }
+}
+
+// ---------------------------------------------------------------------
+// K7 (multiclass): one wave per instance; lane c owns class c
+// (n_classes <= 64).  scores[c] = sum_j w[slot_j][c] * x_j  — lanes read
+// consecutive class columns of each weight row (coalesced).
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_pa_multiclass_step(
+    const int64_t* __restrict__ indptr,
+    const int64_t* __restrict__ feat_slots,
+    const float* __restrict__ x_vals, const int64_t* __restrict__ labels,
+    const float* __restrict__ weights, float* __restrict__ scores,
+    float* __restrict__ deltas, int64_t B, int64_t nc, int mode, float C) {
+  const int lane = threadIdx.x % WAVE;
+  const int64_t waves_per_block = blockDim.x / WAVE;
+  const int64_t w0 = (int64_t)blockIdx.x * waves_per_block +
+                     threadIdx.x / WAVE;
+  const int64_t wstride = (int64_t)gridDim.x * waves_per_block;
+  for (int64_t i = w0; i < B; i += wstride) {
+    const int64_t lo = indptr[i], hi = indptr[i + 1];
+    float sc = 0.f, n2 = 0.f;
+    for (int64_t j = lo; j < hi; ++j) {
+      float x = x_vals[j];
+      if (lane < nc) sc += weights[feat_slots[j] * nc + lane] * x;
+      if (lane == 0) n2 += x * x;
+    }
+    n2 = __shfl(n2, 0);
+    if (lane < nc) scores[i * nc + lane] = sc;
+    const int64_t y = labels[i];
+    if (y < 0) continue;  // unlabeled
+    // max over wrong classes + the true-class score, via wave reduction
+    float sy = __shfl(sc, (int)y);
+    float wrong = (lane < nc && lane != y) ? sc : -INFINITY;
+    int ridx = lane;
+#pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+      float o = __shfl_xor(wrong, off);
+      int oi = __shfl_xor(ridx, off);
+      if (o > wrong) { wrong = o; ridx = oi; }
+    }
+    float loss = 1.0f - sy + wrong;
+    if (loss <= 0.f) continue;
+    float sn = 2.0f * n2;
+    sn = sn > 1e-12f ? sn : 1e-12f;
+    float tau;
+    if (mode == 0) tau = loss / sn;
+    else if (mode == 1) { tau = loss / sn; tau = tau > C ? C : tau; }
+    else tau = loss / (sn + 1.0f / (2.0f * C));
+    for (int64_t j = lo + lane; j < hi; j += WAVE) {
+      float t = tau * x_vals[j];
+      atomicAdd(&deltas[feat_slots[j] * nc + y], t);
+      atomicAdd(&deltas[feat_slots[j] * nc + ridx], -t);
+    }
+  }
+}
+
+// =====================================================================
+// Host wrappers
+// =====================================================================
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+torch::Tensor gather(torch::Tensor table, torch::Tensor rows) {
+  CHECK_DEV(table); CHECK_DEV(rows); CHECK_CONT(table); CHECK_CONT(rows);
+  TORCH_CHECK(table.dtype() == torch::kFloat32 &&
+              rows.dtype() == torch::kInt64);
+  int64_t n = rows.numel(), dim = table.size(1);
+  auto out = torch::empty({n, dim}, table.options());
+  if (!n) return out;
+  if (dim % 4 == 0) {
+    int64_t dim4 = dim / 4;
+    hipLaunchKernelGGL(k_gather_vec4, dim3(grid_for(n * dim4, 256)),
+                       dim3(256), 0, cur_stream(),
+                       (const float4*)table.data_ptr<float>(),
+                       rows.data_ptr<int64_t>(),
+                       (float4*)out.data_ptr<float>(), n, dim4);
+  } else {
+    hipLaunchKernelGGL(k_gather, dim3(grid_for(n * dim, 256)), dim3(256),
+                       0, cur_stream(), table.data_ptr<float>(),
+                       rows.data_ptr<int64_t>(), out.data_ptr<float>(),
+                       n, dim);
+  }
+  return out;
+}
+
+void apply_add(torch::Tensor table, torch::Tensor rows,
+               torch::Tensor deltas) {
+  CHECK_DEV(table); CHECK_DEV(rows); CHECK_DEV(deltas);
+  CHECK_CONT(table); CHECK_CONT(rows); CHECK_CONT(deltas);
+  int64_t n = rows.numel(), dim = table.size(1);
+  TORCH_CHECK(deltas.size(0) == n && deltas.size(1) == dim);
+  if (!n) return;
+  hipLaunchKernelGGL(k_apply_add, dim3(grid_for(n * dim, 256)), dim3(256),
+                     0, cur_stream(), table.data_ptr<float>(),
+                     rows.data_ptr<int64_t>(), deltas.data_ptr<float>(),
+                     n, dim);
+}
+
+void apply_adagrad_sorted(torch::Tensor table, torch::Tensor accum,
+                          torch::Tensor srows, torch::Tensor sgrads,
+                          double lr, double eps) {
+  CHECK_DEV(table); CHECK_DEV(accum); CHECK_DEV(srows); CHECK_DEV(sgrads);
+  CHECK_CONT(table); CHECK_CONT(accum); CHECK_CONT(srows); CHECK_CONT(sgrads);
+  int64_t n = srows.numel(), dim = table.size(1);
+  if (!n) return;
+  hipLaunchKernelGGL(k_apply_adagrad_sorted, dim3(grid_for(n * dim, 256)),
+                     dim3(256), 0, cur_stream(), table.data_ptr<float>(),
+                     accum.data_ptr<float>(), srows.data_ptr<int64_t>(),
+                     sgrads.data_ptr<float>(), n, dim, (float)lr,
+                     (float)eps);
+}
+
+torch::Tensor hash_init(torch::Tensor ids, int64_t dim, double lo,
+                        double hi, int64_t seed) {
+  CHECK_DEV(ids); CHECK_CONT(ids);
+  int64_t n = ids.numel();
+  auto out = torch::empty({n, dim}, torch::dtype(torch::kFloat32)
+                                        .device(ids.device()));
+  if (!n) return out;
+  hipLaunchKernelGGL(k_hash_init_ids, dim3(grid_for(n * dim, 256)),
+                     dim3(256), 0, cur_stream(), out.data_ptr<float>(),
+                     ids.data_ptr<int64_t>(), n, dim, (float)lo, (float)hi,
+                     seed);
+  return out;
+}
+
+void hash_init_rows(torch::Tensor out, int64_t first_id, int64_t stride,
+                    double lo, double hi, int64_t seed) {
+  CHECK_DEV(out); CHECK_CONT(out);
+  int64_t rows = out.size(0), dim = out.size(1);
+  if (!rows) return;
+  hipLaunchKernelGGL(k_hash_init_rows, dim3(grid_for(rows * dim, 256)),
+                     dim3(256), 0, cur_stream(), out.data_ptr<float>(),
+                     rows, dim, first_id, stride, (float)lo, (float)hi,
+                     seed);
+}
+
+torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
+                          torch::Tensor item_vecs, torch::Tensor item_slots,
+                          torch::Tensor ratings, double lr, double reg) {
+  CHECK_DEV(user_table); CHECK_CONT(user_table);
+  CHECK_DEV(item_vecs); CHECK_CONT(item_vecs);
+  CHECK_CONT(user_rows); CHECK_CONT(item_slots); CHECK_CONT(ratings);
+  int64_t n = ratings.numel(), dim = user_table.size(1);
+  TORCH_CHECK(item_vecs.size(1) == dim);
+  auto deltas = torch::zeros_like(item_vecs);
+  if (!n) return deltas;
+  auto launch = [&](auto tag) {
+    constexpr int GP = decltype(tag)::value;
+    hipLaunchKernelGGL((k_mf_sgd_step<GP>),
+                       dim3(grid_for(n * GP, 256)), dim3(256), 0,
+                       cur_stream(), user_table.data_ptr<float>(),
+                       user_rows.data_ptr<int64_t>(),
+                       item_vecs.data_ptr<float>(),
+                       item_slots.data_ptr<int64_t>(),
+                       ratings.data_ptr<float>(),
+                       deltas.data_ptr<float>(), n, dim, (float)lr,
+                       (float)reg);
+  };
+  if (dim <= 16) launch(std::integral_constant<int, 16>{});
+  else if (dim <= 256) launch(std::integral_constant<int, 32>{});
+  else { TORCH_CHECK(dim <= 512, "dim > 512 unsupported");
+         launch(std::integral_constant<int, 64>{}); }
+  return deltas;
+}
+
+std::vector<torch::Tensor> sgns_step(
+    torch::Tensor center_vecs, torch::Tensor center_slots,
+    torch::Tensor ctx_vecs, torch::Tensor ctx_slots, torch::Tensor labels,
+    double lr) {
+  CHECK_DEV(center_vecs); CHECK_CONT(center_vecs);
+  CHECK_DEV(ctx_vecs); CHECK_CONT(ctx_vecs);
+  int64_t n = labels.numel(), dim = center_vecs.size(1);
+  auto cd = torch::zeros_like(center_vecs);
+  auto od = torch::zeros_like(ctx_vecs);
+  if (!n) return {cd, od};
+  auto launch = [&](auto tag) {
+    constexpr int GP = decltype(tag)::value;
+    hipLaunchKernelGGL((k_sgns_step<GP>), dim3(grid_for(n * GP, 256)),
+                       dim3(256), 0, cur_stream(),
+                       center_vecs.data_ptr<float>(),
+                       center_slots.data_ptr<int64_t>(),
+                       ctx_vecs.data_ptr<float>(),
+                       ctx_slots.data_ptr<int64_t>(),
+                       labels.data_ptr<float>(), cd.data_ptr<float>(),
+                       od.data_ptr<float>(), n, dim, (float)lr);
+  };
+  if (dim <= 16) launch(std::integral_constant<int, 16>{});
+  else if (dim <= 256) launch(std::integral_constant<int, 32>{});
+  else { TORCH_CHECK(dim <= 512, "dim > 512 unsupported");
+         launch(std::integral_constant<int, 64>{}); }
+  return {cd, od};
+}
+
+static int pa_mode(const std::string& variant) {
+  if (variant == "PA") return 0;
+  if (variant == "PA-I") return 1;
+  if (variant == "PA-II") return 2;
+  TORCH_CHECK(false, "unknown PA variant ", variant);
+}
+
+std::vector<torch::Tensor> pa_step(torch::Tensor indptr,
+                                   torch::Tensor feat_slots,
+                                   torch::Tensor x_vals,
+                                   torch::Tensor labels,
+                                   torch::Tensor weights,
+                                   std::string variant, double C) {
+  CHECK_DEV(weights); CHECK_CONT(weights);
+  CHECK_CONT(indptr); CHECK_CONT(feat_slots); CHECK_CONT(x_vals);
+  int64_t B = indptr.numel() - 1;
+  auto margins = torch::zeros({B}, x_vals.options());
+  auto deltas = torch::zeros_like(weights);
+  if (!B) return {margins, deltas};
+  hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
+                     dim3(256), 0, cur_stream(),
+                     indptr.data_ptr<int64_t>(),
+                     feat_slots.data_ptr<int64_t>(),
+                     x_vals.data_ptr<float>(), labels.data_ptr<float>(),
+                     weights.data_ptr<float>(), margins.data_ptr<float>(),
+                     deltas.data_ptr<float>(), B, pa_mode(variant),
+                     (float)C);
+  return {margins, deltas};
+}
+
+std::vector<torch::Tensor> lr_step(torch::Tensor indptr,
+                                   torch::Tensor feat_slots,
+                                   torch::Tensor x_vals,
+                                   torch::Tensor labels,
+                                   torch::Tensor weights) {
+  CHECK_DEV(weights); CHECK_CONT(weights);
+  int64_t B = indptr.numel() - 1;
+  auto probs = torch::zeros({B}, x_vals.options());
+  auto grads = torch::zeros_like(weights);
+  if (!B) return {probs, grads};
+  hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
+                     dim3(256), 0, cur_stream(),
+                     indptr.data_ptr<int64_t>(),
+                     feat_slots.data_ptr<int64_t>(),
+                     x_vals.data_ptr<float>(), labels.data_ptr<float>(),
+                     weights.data_ptr<float>(), probs.data_ptr<float>(),
+                     grads.data_ptr<float>(), B, /*mode=*/3, 0.f);
+  // kernel writes margins; convert to probabilities in-place
+  probs.sigmoid_();
+  return {probs, grads};
+}
+
+std::vector<torch::Tensor> pa_multiclass_step(
+    torch::Tensor indptr, torch::Tensor feat_slots, torch::Tensor x_vals,
+    torch::Tensor labels, torch::Tensor weights, std::string variant,
+    double C) {
+  CHECK_DEV(weights); CHECK_CONT(weights);
+  int64_t B = indptr.numel() - 1, nc = weights.size(1);
+  TORCH_CHECK(nc <= 64, "n_classes > 64 unsupported");
+  auto scores = torch::zeros({B, nc}, x_vals.options());
+  auto deltas = torch::zeros_like(weights);
+  if (!B) return {scores, deltas};
+  hipLaunchKernelGGL(k_pa_multiclass_step, dim3(grid_for(B * WAVE, 256)),
+                     dim3(256), 0, cur_stream(),
+                     indptr.data_ptr<int64_t>(),
+                     feat_slots.data_ptr<int64_t>(),
+                     x_vals.data_ptr<float>(), labels.data_ptr<int64_t>(),
+                     weights.data_ptr<float>(), scores.data_ptr<float>(),
+                     deltas.data_ptr<float>(), B, nc, pa_mode(variant),
+                     (float)C);
+  return {scores, deltas};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gather", &gather, "K3 ps_gather");
+  m.def("apply_add", &apply_add, "K1 ps_apply_sgd (delta add)");
+  m.def("apply_adagrad_sorted", &apply_adagrad_sorted,
+        "K2 ps_apply_adagrad over a row-sorted batch");
+  m.def("hash_init", &hash_init, "K6 deterministic per-id init");
+  m.def("hash_init_rows", &hash_init_rows, "K6 slab init");
+  m.def("mf_sgd_step", &mf_sgd_step, "K4 fused MF SGD step");
+  m.def("sgns_step", &sgns_step, "K8 fused SGNS step");
+  m.def("pa_step", &pa_step, "K7 PA binary step");
+  m.def("pa_multiclass_step", &pa_multiclass_step, "K7 PA multiclass");
+  m.def("lr_step", &lr_step, "K9 sparse logistic step");
+}

@@ -43,7 +43,13 @@ def apply_adagrad(table: torch.Tensor, accum: torch.Tensor,
                   rows: torch.Tensor, grads: torch.Tensor,
                   lr: float, eps: float = 1e-8) -> None:
     if _cuda(table):
-        hip.ext().apply_adagrad(table, accum, rows, grads, lr, eps)
+        # K2 contract: duplicates pre-aggregated, one RMW per distinct
+        # row.  Sort the batch (device radix sort) and let segment
+        # leaders aggregate in-kernel.
+        srows, order = torch.sort(rows)
+        hip.ext().apply_adagrad_sorted(table, accum, srows.contiguous(),
+                                       grads.index_select(0, order)
+                                       .contiguous(), lr, eps)
         return
     reference.apply_adagrad(table, accum, rows, grads, lr, eps)
 
