@@ -23,7 +23,7 @@ setup(
             sources=["csrc/ps_kernels.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
-                "nvcc": ["-O3", "--offload-arch=gfx950"],
+                "nvcc": ["-O3", "--offload-arch=gfx950", "-ffp-contract=off"],
             },
         )
     ],
