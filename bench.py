@@ -59,7 +59,7 @@ def make_mf(args, rank, world, device):
     table = ShardedTable(part, rank, k, device, init=(-0.01, 0.01))
     wl = OnlineMFWorkerLogic(users_local, k, learning_rate=0.01,
                              reg=0.001, range_min=-0.01, range_max=0.01,
-                             device=device, seed=rank + 1)
+                             n_items=n_items, device=device, seed=rank + 1)
     g = torch.Generator().manual_seed(1234 + rank)
     batches = []
     for _ in range(args.n_batches):

@@ -163,65 +163,30 @@ __device__ __forceinline__ float group_sum(float v) {
 }
 
 // ---------------------------------------------------------------------
-// K4: fused MF SGD step.  One lane group per rating:
-//   e  = r - u.v      (v = pulled item snapshot)
-//   u += lr*(e*v - reg*u)         (atomic: duplicate users may race,
-//                                  async-parity semantics)
-//   item_deltas[slot] += lr*(e*u0 - reg*v)   (atomic accumulate; u0 is
-//                                  the value read for the dot)
+// K4/K8 shared machinery.  The zipfian workloads make naive per-sample
+// atomics serialize on hot rows (measured: the first-cut one-kernel MF
+// step ran at ~300 GB/s, 70% of step time — profiles/
+// mf32_r01_baseline_atomics.md), so the fused steps are decomposed into
+//   (1) k_pair_coeff  — per-sample dot + scalar coefficient, no writes
+//                       beyond one float per sample;
+//   (2) k_seg_accum   — segmented scatter-accumulate over KEY-SORTED
+//                       samples: each lane group walks a contiguous
+//                       chunk, accumulates same-key rows in registers,
+//                       and issues ONE atomicAdd per (segment x chunk)
+//                       — the LDS/register pre-aggregation SURVEY.md
+//                       §7.3 calls for.
+// Sorting is rocPRIM radix sort via torch::argsort in the host wrapper.
 // ---------------------------------------------------------------------
-template <int GP>
-__global__ __launch_bounds__(256) void k_mf_sgd_step(
-    float* __restrict__ user_table, const int64_t* __restrict__ user_rows,
-    const float* __restrict__ item_vecs,
-    const int64_t* __restrict__ item_slots,
-    const float* __restrict__ ratings, float* __restrict__ item_deltas,
-    int64_t n, int64_t dim, float lr, float reg) {
-  constexpr int EPL = 8;           // max elements per lane
-  const int lane = threadIdx.x % GP;
-  const int64_t groups_per_block = blockDim.x / GP;
-  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
-                     threadIdx.x / GP;
-  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
-  for (int64_t g = g0; g < n; g += gstride) {
-    const int64_t urow = user_rows[g] * dim;
-    const int64_t vrow = item_slots[g] * dim;
-    float u[EPL], v[EPL];
-    float dot = 0.f;
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) {
-      int c = lane + e * GP;
-      if (c < dim) {
-        u[e] = user_table[urow + c];
-        v[e] = item_vecs[vrow + c];
-        dot += u[e] * v[e];
-      }
-    }
-    float e_ = ratings[g] - group_sum<GP>(dot);
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) {
-      int c = lane + e * GP;
-      if (c < dim) {
-        atomicAdd(&user_table[urow + c], lr * (e_ * v[e] - reg * u[e]));
-        atomicAdd(&item_deltas[vrow + c], lr * (e_ * u[e] - reg * v[e]));
-      }
-    }
-  }
-}
 
-// ---------------------------------------------------------------------
-// K8: fused SGNS step.  One lane group per (center, context) pair row:
-//   s = sigmoid(v.u); g = lr*(label - s)
-//   center_deltas[cslot] += g*u ; ctx_deltas[oslot] += g*v   (atomic)
-// ---------------------------------------------------------------------
+// coeff[j] = mode 0: s_in[j] - A[arows[j]] . B[brows[j]]        (MF: e)
+//            mode 1: lr * (s_in[j] - sigmoid(A[arows[j]].B[brows[j]]))
+//                                                               (SGNS: g)
 template <int GP>
-__global__ __launch_bounds__(256) void k_sgns_step(
-    const float* __restrict__ center_vecs,
-    const int64_t* __restrict__ center_slots,
-    const float* __restrict__ ctx_vecs,
-    const int64_t* __restrict__ ctx_slots,
-    const float* __restrict__ labels, float* __restrict__ center_deltas,
-    float* __restrict__ ctx_deltas, int64_t n, int64_t dim, float lr) {
+__global__ __launch_bounds__(256) void k_pair_coeff(
+    const float* __restrict__ A, const int64_t* __restrict__ arows,
+    const float* __restrict__ B, const int64_t* __restrict__ brows,
+    const float* __restrict__ s_in, float* __restrict__ coeff,
+    int64_t n, int64_t dim, int mode, float lr) {
   constexpr int EPL = 8;
   const int lane = threadIdx.x % GP;
   const int64_t groups_per_block = blockDim.x / GP;
@@ -229,28 +194,78 @@ __global__ __launch_bounds__(256) void k_sgns_step(
                      threadIdx.x / GP;
   const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
   for (int64_t g = g0; g < n; g += gstride) {
-    const int64_t crow = center_slots[g] * dim;
-    const int64_t orow = ctx_slots[g] * dim;
-    float v[EPL], u[EPL];
+    const int64_t ar = arows[g] * dim, br = brows[g] * dim;
     float dot = 0.f;
 #pragma unroll
     for (int e = 0; e < EPL; ++e) {
       int c = lane + e * GP;
-      if (c < dim) {
-        v[e] = center_vecs[crow + c];
-        u[e] = ctx_vecs[orow + c];
-        dot += v[e] * u[e];
-      }
+      if (c < dim) dot += A[ar + c] * B[br + c];
     }
-    float d = group_sum<GP>(dot);
-    float s = 1.0f / (1.0f + __expf(-d));
-    float gg = lr * (labels[g] - s);
+    dot = group_sum<GP>(dot);
+    if (lane == 0) {
+      if (mode == 0) coeff[g] = s_in[g] - dot;
+      else coeff[g] = lr * (s_in[g] - 1.0f / (1.0f + __expf(-dot)));
+    }
+  }
+}
+
+// out[key_j] += sum_j scale*coeff[j]*src[slots[j]]  (+ a*cnt*self[key])
+// over samples sorted by key; one atomic flush per segment per chunk.
+template <int GP, bool SELF>
+__global__ __launch_bounds__(256) void k_seg_accum(
+    float* __restrict__ out, const int64_t* __restrict__ keys,
+    const int64_t* __restrict__ slots, const float* __restrict__ src,
+    const float* __restrict__ self, const float* __restrict__ coeff,
+    int64_t n, int64_t dim, float scale, float a, int chunk) {
+  constexpr int EPL = 8;
+  const int lane = threadIdx.x % GP;
+  const int64_t groups_per_block = blockDim.x / GP;
+  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
+                     threadIdx.x / GP;
+  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
+  const int64_t nchunks = (n + chunk - 1) / chunk;
+  for (int64_t cidx = g0; cidx < nchunks; cidx += gstride) {
+    const int64_t start = cidx * chunk;
+    const int64_t end = min(n, start + (int64_t)chunk);
+    float acc[EPL];
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) acc[e] = 0.f;
+    int cnt = 0;
+    int64_t cur = keys[start];
+    for (int64_t j = start; j < end; ++j) {
+      const int64_t k = keys[j];
+      if (k != cur) {
+        const int64_t orow = cur * dim;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e) {
+          int c = lane + e * GP;
+          if (c < dim) {
+            float t = scale * acc[e];
+            if (SELF) t += a * cnt * self[orow + c];
+            atomicAdd(&out[orow + c], t);
+          }
+          acc[e] = 0.f;
+        }
+        cnt = 0;
+        cur = k;
+      }
+      const float gj = coeff[j];
+      const int64_t srow = slots[j] * dim;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) {
+        int c = lane + e * GP;
+        if (c < dim) acc[e] += gj * src[srow + c];
+      }
+      ++cnt;
+    }
+    const int64_t orow = cur * dim;
 #pragma unroll
     for (int e = 0; e < EPL; ++e) {
       int c = lane + e * GP;
       if (c < dim) {
-        atomicAdd(&center_deltas[crow + c], gg * u[e]);
-        atomicAdd(&ctx_deltas[orow + c], gg * v[e]);
+        float t = scale * acc[e];
+        if (SELF) t += a * cnt * self[orow + c];
+        atomicAdd(&out[orow + c], t);
       }
     }
   }
@@ -442,6 +457,66 @@ void hash_init_rows(torch::Tensor out, int64_t first_id, int64_t stride,
                      seed);
 }
 
+// host-side lane-group dispatch shared by the pair/seg kernels
+template <class F>
+static void by_group(int64_t dim, F f) {
+  if (dim <= 16) f(std::integral_constant<int, 16>{});
+  else if (dim <= 256) f(std::integral_constant<int, 32>{});
+  else { TORCH_CHECK(dim <= 512, "row dim > 512 unsupported");
+         f(std::integral_constant<int, 64>{}); }
+}
+
+static constexpr int SEG_CHUNK = 64;
+
+static void seg_accum(torch::Tensor out, torch::Tensor keys,
+                      torch::Tensor slots, torch::Tensor src,
+                      const torch::Tensor* self, torch::Tensor coeff,
+                      double scale, double a) {
+  int64_t n = keys.numel(), dim = out.size(1);
+  if (!n) return;
+  int64_t nchunks = (n + SEG_CHUNK - 1) / SEG_CHUNK;
+  by_group(dim, [&](auto tag) {
+    constexpr int GP = decltype(tag)::value;
+    dim3 grid(grid_for(nchunks * GP, 256));
+    if (self) {
+      hipLaunchKernelGGL((k_seg_accum<GP, true>), grid, dim3(256), 0,
+                         cur_stream(), out.data_ptr<float>(),
+                         keys.data_ptr<int64_t>(),
+                         slots.data_ptr<int64_t>(), src.data_ptr<float>(),
+                         self->data_ptr<float>(), coeff.data_ptr<float>(),
+                         n, dim, (float)scale, (float)a, SEG_CHUNK);
+    } else {
+      hipLaunchKernelGGL((k_seg_accum<GP, false>), grid, dim3(256), 0,
+                         cur_stream(), out.data_ptr<float>(),
+                         keys.data_ptr<int64_t>(),
+                         slots.data_ptr<int64_t>(), src.data_ptr<float>(),
+                         nullptr, coeff.data_ptr<float>(), n, dim,
+                         (float)scale, (float)a, SEG_CHUNK);
+    }
+  });
+}
+
+static void pair_coeff(torch::Tensor A, torch::Tensor arows,
+                       torch::Tensor B, torch::Tensor brows,
+                       torch::Tensor s_in, torch::Tensor coeff, int mode,
+                       double lr) {
+  int64_t n = coeff.numel(), dim = A.size(1);
+  if (!n) return;
+  by_group(dim, [&](auto tag) {
+    constexpr int GP = decltype(tag)::value;
+    hipLaunchKernelGGL((k_pair_coeff<GP>),
+                       dim3(grid_for(n * GP, 256)), dim3(256), 0,
+                       cur_stream(), A.data_ptr<float>(),
+                       arows.data_ptr<int64_t>(), B.data_ptr<float>(),
+                       brows.data_ptr<int64_t>(), s_in.data_ptr<float>(),
+                       coeff.data_ptr<float>(), n, dim, mode, (float)lr);
+  });
+}
+
+// K4: e = r - u0.v computed on the item-sorted batch (exact batch-
+// snapshot semantics: every rating sees pre-update u0 and the pulled v);
+// item deltas = lr*(e*u0 - reg*v) segment-accumulated by item; user
+// updates = lr*(e*v - reg*u0) segment-accumulated by user.
 torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                           torch::Tensor item_vecs, torch::Tensor item_slots,
                           torch::Tensor ratings, double lr, double reg) {
@@ -452,50 +527,48 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   TORCH_CHECK(item_vecs.size(1) == dim);
   auto deltas = torch::zeros_like(item_vecs);
   if (!n) return deltas;
-  auto launch = [&](auto tag) {
-    constexpr int GP = decltype(tag)::value;
-    hipLaunchKernelGGL((k_mf_sgd_step<GP>),
-                       dim3(grid_for(n * GP, 256)), dim3(256), 0,
-                       cur_stream(), user_table.data_ptr<float>(),
-                       user_rows.data_ptr<int64_t>(),
-                       item_vecs.data_ptr<float>(),
-                       item_slots.data_ptr<int64_t>(),
-                       ratings.data_ptr<float>(),
-                       deltas.data_ptr<float>(), n, dim, (float)lr,
-                       (float)reg);
-  };
-  if (dim <= 16) launch(std::integral_constant<int, 16>{});
-  else if (dim <= 256) launch(std::integral_constant<int, 32>{});
-  else { TORCH_CHECK(dim <= 512, "dim > 512 unsupported");
-         launch(std::integral_constant<int, 64>{}); }
+  // pass A: sort by item, compute e, accumulate item deltas
+  auto order = torch::argsort(item_slots);
+  auto s_slots = item_slots.index_select(0, order).contiguous();
+  auto s_urows = user_rows.index_select(0, order).contiguous();
+  auto s_r = ratings.index_select(0, order).contiguous();
+  auto e = torch::empty_like(s_r);
+  pair_coeff(user_table, s_urows, item_vecs, s_slots, s_r, e, 0, 0.0);
+  seg_accum(deltas, s_slots, s_urows, user_table, &item_vecs, e, lr,
+            -lr * reg);
+  // pass B: sort by user, apply user updates
+  auto order2 = torch::argsort(s_urows);
+  auto u_rows = s_urows.index_select(0, order2).contiguous();
+  auto u_slots = s_slots.index_select(0, order2).contiguous();
+  auto u_e = e.index_select(0, order2).contiguous();
+  seg_accum(user_table, u_rows, u_slots, item_vecs, &user_table, u_e, lr,
+            -lr * reg);
   return deltas;
 }
 
+// K8: g = lr*(label - sigmoid(v.u)); center and context deltas each
+// segment-accumulated on their own sorted order.
 std::vector<torch::Tensor> sgns_step(
     torch::Tensor center_vecs, torch::Tensor center_slots,
     torch::Tensor ctx_vecs, torch::Tensor ctx_slots, torch::Tensor labels,
     double lr) {
   CHECK_DEV(center_vecs); CHECK_CONT(center_vecs);
   CHECK_DEV(ctx_vecs); CHECK_CONT(ctx_vecs);
-  int64_t n = labels.numel(), dim = center_vecs.size(1);
+  int64_t n = labels.numel();
   auto cd = torch::zeros_like(center_vecs);
   auto od = torch::zeros_like(ctx_vecs);
   if (!n) return {cd, od};
-  auto launch = [&](auto tag) {
-    constexpr int GP = decltype(tag)::value;
-    hipLaunchKernelGGL((k_sgns_step<GP>), dim3(grid_for(n * GP, 256)),
-                       dim3(256), 0, cur_stream(),
-                       center_vecs.data_ptr<float>(),
-                       center_slots.data_ptr<int64_t>(),
-                       ctx_vecs.data_ptr<float>(),
-                       ctx_slots.data_ptr<int64_t>(),
-                       labels.data_ptr<float>(), cd.data_ptr<float>(),
-                       od.data_ptr<float>(), n, dim, (float)lr);
-  };
-  if (dim <= 16) launch(std::integral_constant<int, 16>{});
-  else if (dim <= 256) launch(std::integral_constant<int, 32>{});
-  else { TORCH_CHECK(dim <= 512, "dim > 512 unsupported");
-         launch(std::integral_constant<int, 64>{}); }
+  auto g = torch::empty_like(labels);
+  pair_coeff(center_vecs, center_slots, ctx_vecs, ctx_slots, labels, g, 1,
+             lr);
+  auto orderC = torch::argsort(center_slots);
+  seg_accum(cd, center_slots.index_select(0, orderC).contiguous(),
+            ctx_slots.index_select(0, orderC).contiguous(), ctx_vecs,
+            nullptr, g.index_select(0, orderC).contiguous(), 1.0, 0.0);
+  auto orderO = torch::argsort(ctx_slots);
+  seg_accum(od, ctx_slots.index_select(0, orderO).contiguous(),
+            center_slots.index_select(0, orderO).contiguous(), center_vecs,
+            nullptr, g.index_select(0, orderO).contiguous(), 1.0, 0.0);
   return {cd, od};
 }
 

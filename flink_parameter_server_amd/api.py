@@ -46,8 +46,11 @@ class ParameterServerClient(ABC):
     """
 
     @abstractmethod
-    def pull(self, ids: torch.Tensor) -> None:
-        """Request current values for ``ids`` (async, answered later)."""
+    def pull(self, ids: torch.Tensor, unique: bool = False) -> None:
+        """Request current values for ``ids`` (async, answered later).
+        ``unique=True`` asserts the caller already de-duplicated ``ids``
+        (sorted or not), letting the engine skip its per-round coalescing
+        pass when this is the round's only pull."""
 
     @abstractmethod
     def push(self, ids: torch.Tensor, deltas: torch.Tensor) -> None:

@@ -18,8 +18,10 @@ import torch
 class PullBuffer:
     """Collects pulled rows for a fixed set of unique ids."""
 
-    def __init__(self, uids: torch.Tensor, dim: int):
-        uids, _ = torch.sort(uids)
+    def __init__(self, uids: torch.Tensor, dim: int,
+                 assume_sorted: bool = False):
+        if not assume_sorted:
+            uids, _ = torch.sort(uids)
         self.uids = uids
         self.values = torch.empty(uids.numel(), dim, dtype=torch.float32,
                                   device=uids.device)
@@ -30,6 +32,15 @@ class PullBuffer:
     def slots_of(self, ids: torch.Tensor) -> torch.Tensor:
         """Position of each id in the sorted unique-id list."""
         return torch.searchsorted(self.uids, ids)
+
+    def offer_exact(self, ids: torch.Tensor, values: torch.Tensor) -> None:
+        """Fast path: the caller guarantees ``set(ids) == set(self.uids)``
+        (e.g. the engine answered this buffer's whole pull in one round and
+        no other buffer was pending).  No host-device syncs."""
+        rows = torch.searchsorted(self.uids, ids)
+        self.values[rows] = values.to(self.values.dtype)
+        self._have.fill_(True)
+        self._n_have = int(self.uids.numel())
 
     def offer(self, ids: torch.Tensor, values: torch.Tensor) -> None:
         """Fill rows answered by (ids, values); ignores foreign ids and

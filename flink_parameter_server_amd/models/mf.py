@@ -72,7 +72,16 @@ class OnlineMFWorkerLogic(WorkerLogic):
         if n_local_users:
             ops.hash_init_rows(self.user_table, 0, 1, range_min, range_max,
                                seed=seed + 7777)
-        self._pending: List[Tuple[MFBatch, PullBuffer]] = []
+        self._pending: List[Tuple[MFBatch, PullBuffer, Optional[torch.Tensor]]] = []
+        # dense-flag coalescing: O(n) scatter + cumsum instead of a sort-
+        # based torch.unique, and rating->slot lookup for free.  Used when
+        # the item-id space is known and small enough for a dense flag
+        # array (measured: rocPRIM sort-unique was ~7% of a 1M-rating
+        # round — profiles/mf32_r01_baseline_atomics.md).
+        self._present: Optional[torch.Tensor] = None
+        if 0 < self.n_items <= (1 << 24):
+            self._present = torch.zeros(self.n_items, dtype=torch.bool,
+                                        device=self.device)
 
     def batch_size(self, batch: MFBatch) -> int:
         return len(batch)
@@ -80,25 +89,46 @@ class OnlineMFWorkerLogic(WorkerLogic):
     def on_recv(self, data: MFBatch, client: ParameterServerClient) -> None:
         if self.neg_rate > 0:
             data = self._with_negatives(data)
-        uids = torch.unique(data.item_ids)
-        buf = PullBuffer(uids, self.n_factors)
-        self._pending.append((data, buf))
-        client.pull(uids)
+        if self._present is not None:
+            self._present.zero_()
+            self._present[data.item_ids] = True
+            lookup = torch.cumsum(self._present, 0) - 1
+            uids = self._present.nonzero().reshape(-1)
+            slots = lookup[data.item_ids]
+            buf = PullBuffer(uids, self.n_factors, assume_sorted=True)
+        else:
+            uids = torch.unique(data.item_ids)
+            slots = None
+            buf = PullBuffer(uids, self.n_factors, assume_sorted=True)
+        self._pending.append((data, buf, slots))
+        client.pull(uids, unique=True)
 
     def on_pull_recv(self, ids: torch.Tensor, values: torch.Tensor,
                      client: ParameterServerClient) -> None:
-        still: List[Tuple[MFBatch, PullBuffer]] = []
-        for batch, buf in self._pending:
+        if (len(self._pending) == 1 and
+                self._pending[0][1]._n_have == 0 and
+                ids.numel() == self._pending[0][1].uids.numel()):
+            # the engine answered this buffer's whole pull at once and no
+            # other buffer could have contributed ids: sync-free path
+            batch, buf, slots = self._pending[0]
+            buf.offer_exact(ids, values)
+            self._pending = []
+            self._train(batch, buf, client, slots)
+            return
+        still: List[Tuple[MFBatch, PullBuffer, Optional[torch.Tensor]]] = []
+        for batch, buf, slots in self._pending:
             buf.offer(ids, values)
             if buf.complete:
-                self._train(batch, buf, client)
+                self._train(batch, buf, client, slots)
             else:
-                still.append((batch, buf))
+                still.append((batch, buf, slots))
         self._pending = still
 
     def _train(self, batch: MFBatch, buf: PullBuffer,
-               client: ParameterServerClient) -> None:
-        slots = buf.slots_of(batch.item_ids)
+               client: ParameterServerClient,
+               slots: Optional[torch.Tensor] = None) -> None:
+        if slots is None:
+            slots = buf.slots_of(batch.item_ids)
         deltas = ops.mf_sgd_step(self.user_table, batch.user_rows,
                                  buf.values, slots, batch.ratings,
                                  self.lr, self.reg)

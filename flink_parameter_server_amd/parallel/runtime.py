@@ -59,9 +59,9 @@ class _Client(ParameterServerClient):
     def __init__(self, engine: "PSEngine"):
         self._e = engine
 
-    def pull(self, ids: torch.Tensor) -> None:
+    def pull(self, ids: torch.Tensor, unique: bool = False) -> None:
         if ids.numel():
-            self._e._pull_buf.append(ids.reshape(-1))
+            self._e._pull_buf.append((ids.reshape(-1), unique))
 
     def push(self, ids: torch.Tensor, deltas: torch.Tensor) -> None:
         if ids.numel():
@@ -116,7 +116,7 @@ class PSEngine:
         self.metrics = Metrics()
 
         # round state
-        self._pull_buf: List[torch.Tensor] = []
+        self._pull_buf: List[Tuple[torch.Tensor, bool]] = []
         self._push_buf: List[Tuple[torch.Tensor, torch.Tensor]] = []
         self._answer_buf: List[List[Tuple[torch.Tensor, torch.Tensor]]] = \
             [[] for _ in range(self.world)]
@@ -243,12 +243,14 @@ class PSEngine:
         parts = self._pull_buf
         self._pull_buf = []
         if self._pull_queue is not None:
-            parts = [self._pull_queue] + parts
+            parts = [(self._pull_queue, False)] + parts
             self._pull_queue = None
-        ids = _concat(parts)
-        if ids is None:
+        if not parts:
             return None
-        ids = torch.unique(ids)
+        if len(parts) == 1 and parts[0][1]:
+            ids = parts[0][0]  # caller-guaranteed unique: skip coalescing
+        else:
+            ids = torch.unique(_concat([t for t, _ in parts]))
         if self._pull_limit is not None:
             budget = self._pull_limit - self._outstanding
             if budget <= 0:

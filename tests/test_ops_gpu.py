@@ -182,11 +182,11 @@ def test_mf_engine_e2e_on_gpu():
     dev = torch.device(DEV)
     n_users, n_items, rank = 200, 150, 16
     batches, _, _ = synth.planted_mf_ratings(n_users, n_items, rank,
-                                             20000, 1024, seed=3,
+                                             20000, 256, seed=3,
                                              device=dev)
     part = ModPartitioner(n_items, 1)
     table = ShardedTable(part, 0, rank, dev, init=(-0.1, 0.1))
-    wl = OnlineMFWorkerLogic(n_users, rank, learning_rate=0.05,
+    wl = OnlineMFWorkerLogic(n_users, rank, learning_rate=0.1,
                              range_min=-0.1, range_max=0.1, device=dev)
     eng = PSEngine(PSConfig(num_ids=n_items, dim=rank), wl,
                    TableServerLogic(table, dump_on_close=False), None,
@@ -202,7 +202,7 @@ def test_mf_engine_e2e_on_gpu():
         return tot / n
 
     l0 = loss()
-    eng.run(epoch_iterator(batches, 4))
+    eng.run(epoch_iterator(batches, 8))
     torch.cuda.synchronize()
     l1 = loss()
-    assert l1 < 0.5 * l0, (l0, l1)
+    assert l1 < 0.3 * l0, (l0, l1)
