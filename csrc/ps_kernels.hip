@@ -457,6 +457,110 @@ void hash_init_rows(torch::Tensor out, int64_t first_id, int64_t stride,
                      seed);
 }
 
+// ---------------------------------------------------------------------
+// K10 bucket_pack: counting-sort a batch by a small-range key (m <=
+// 16384: 2*m*4 B of dynamic LDS), fusing the payload permutation.
+// Replaces torch.argsort + 3x index_select in the MF hot path (measured:
+// rocPRIM merge sort was ~50% of step GPU time —
+// profiles/mf32_r01_baseline_atomics.md follow-up).  Order within a key
+// is unspecified (the consumers only need grouping, not stability).
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_key_hist(
+    const int64_t* __restrict__ keys, int* __restrict__ ghist, int64_t n,
+    int m) {
+  extern __shared__ int lhist[];
+  for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
+  __syncthreads();
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x)
+    atomicAdd(&lhist[(int)keys[j]], 1);
+  __syncthreads();
+  for (int i = threadIdx.x; i < m; i += blockDim.x)
+    if (lhist[i]) atomicAdd(&ghist[i], lhist[i]);
+}
+
+__global__ __launch_bounds__(256) void k_bucket_pack(
+    const int64_t* __restrict__ keys, const int64_t* __restrict__ p0,
+    const float* __restrict__ p1, int* __restrict__ gcursor,
+    int64_t* __restrict__ okeys, int64_t* __restrict__ op0,
+    float* __restrict__ op1, int64_t n, int m, int chunk) {
+  extern __shared__ int lds[];
+  int* lhist = lds;
+  int* lbase = lds + m;
+  const int64_t nchunks = (n + chunk - 1) / chunk;
+  for (int64_t c = blockIdx.x; c < nchunks; c += gridDim.x) {
+    const int64_t start = c * chunk;
+    const int64_t end = min(n, start + (int64_t)chunk);
+    for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
+    __syncthreads();
+    for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x)
+      atomicAdd(&lhist[(int)keys[j]], 1);
+    __syncthreads();
+    for (int i = threadIdx.x; i < m; i += blockDim.x) {
+      int h = lhist[i];
+      lbase[i] = h ? atomicAdd(&gcursor[i], h) : 0;
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
+    __syncthreads();
+    for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x) {
+      const int k = (int)keys[j];
+      const int lpos = atomicAdd(&lhist[k], 1);
+      const int64_t pos = (int64_t)lbase[k] + lpos;
+      okeys[pos] = k;
+      op0[pos] = p0[j];
+      op1[pos] = p1[j];
+    }
+    __syncthreads();
+  }
+}
+
+
+// counting-sort pack (K10) for small key ranges; returns
+// (sorted_keys, sorted_p0, sorted_p1).
+static constexpr int PACK_MAX_M = 16384;
+static constexpr int PACK_CHUNK = 2048;
+
+static std::vector<torch::Tensor> bucket_pack(torch::Tensor keys,
+                                              torch::Tensor p0,
+                                              torch::Tensor p1, int m) {
+  int64_t n = keys.numel();
+  auto opts_i32 = torch::dtype(torch::kInt32).device(keys.device());
+  auto hist = torch::zeros({m}, opts_i32);
+  size_t lds1 = (size_t)m * sizeof(int);
+  hipLaunchKernelGGL(k_key_hist, dim3(grid_for(n, 256)), dim3(256), lds1,
+                     cur_stream(), keys.data_ptr<int64_t>(),
+                     hist.data_ptr<int>(), n, m);
+  auto starts = torch::cumsum(hist, 0, torch::kInt32) - hist;
+  auto cursor = starts.contiguous();
+  auto okeys = torch::empty_like(keys);
+  auto op0 = torch::empty_like(p0);
+  auto op1 = torch::empty_like(p1);
+  int64_t nchunks = (n + PACK_CHUNK - 1) / PACK_CHUNK;
+  int grid = (int)std::min<int64_t>(nchunks, 2048);
+  hipLaunchKernelGGL(k_bucket_pack, dim3(grid), dim3(256),
+                     2 * (size_t)m * sizeof(int), cur_stream(),
+                     keys.data_ptr<int64_t>(), p0.data_ptr<int64_t>(),
+                     p1.data_ptr<float>(), cursor.data_ptr<int>(),
+                     okeys.data_ptr<int64_t>(), op0.data_ptr<int64_t>(),
+                     op1.data_ptr<float>(), n, m, PACK_CHUNK);
+  return {okeys, op0, op1};
+}
+
+// sort (keys, int64 payload, f32 payload) by key: counting sort when the
+// key range m is small, radix sort (torch::argsort) otherwise.
+static std::vector<torch::Tensor> sort_by_key(torch::Tensor keys,
+                                              torch::Tensor p0,
+                                              torch::Tensor p1,
+                                              int64_t m) {
+  if (m > 0 && m <= PACK_MAX_M)
+    return bucket_pack(keys, p0, p1, (int)m);
+  auto order = torch::argsort(keys);
+  return {keys.index_select(0, order).contiguous(),
+          p0.index_select(0, order).contiguous(),
+          p1.index_select(0, order).contiguous()};
+}
+
 // host-side lane-group dispatch shared by the pair/seg kernels
 template <class F>
 static void by_group(int64_t dim, F f) {
@@ -527,22 +631,20 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   TORCH_CHECK(item_vecs.size(1) == dim);
   auto deltas = torch::zeros_like(item_vecs);
   if (!n) return deltas;
-  // pass A: sort by item, compute e, accumulate item deltas
-  auto order = torch::argsort(item_slots);
-  auto s_slots = item_slots.index_select(0, order).contiguous();
-  auto s_urows = user_rows.index_select(0, order).contiguous();
-  auto s_r = ratings.index_select(0, order).contiguous();
+  // pass A: counting-sort by item (K10), compute e, accumulate item deltas
+  auto packA = sort_by_key(item_slots, user_rows, ratings,
+                           item_vecs.size(0));
+  auto& s_slots = packA[0];
+  auto& s_urows = packA[1];
+  auto& s_r = packA[2];
   auto e = torch::empty_like(s_r);
   pair_coeff(user_table, s_urows, item_vecs, s_slots, s_r, e, 0, 0.0);
   seg_accum(deltas, s_slots, s_urows, user_table, &item_vecs, e, lr,
             -lr * reg);
-  // pass B: sort by user, apply user updates
-  auto order2 = torch::argsort(s_urows);
-  auto u_rows = s_urows.index_select(0, order2).contiguous();
-  auto u_slots = s_slots.index_select(0, order2).contiguous();
-  auto u_e = e.index_select(0, order2).contiguous();
-  seg_accum(user_table, u_rows, u_slots, item_vecs, &user_table, u_e, lr,
-            -lr * reg);
+  // pass B: counting-sort by user, apply user updates
+  auto packB = sort_by_key(s_urows, s_slots, e, user_table.size(0));
+  seg_accum(user_table, packB[0], packB[1], item_vecs, &user_table,
+            packB[2], lr, -lr * reg);
   return deltas;
 }
 
@@ -561,14 +663,12 @@ std::vector<torch::Tensor> sgns_step(
   auto g = torch::empty_like(labels);
   pair_coeff(center_vecs, center_slots, ctx_vecs, ctx_slots, labels, g, 1,
              lr);
-  auto orderC = torch::argsort(center_slots);
-  seg_accum(cd, center_slots.index_select(0, orderC).contiguous(),
-            ctx_slots.index_select(0, orderC).contiguous(), ctx_vecs,
-            nullptr, g.index_select(0, orderC).contiguous(), 1.0, 0.0);
-  auto orderO = torch::argsort(ctx_slots);
-  seg_accum(od, ctx_slots.index_select(0, orderO).contiguous(),
-            center_slots.index_select(0, orderO).contiguous(), center_vecs,
-            nullptr, g.index_select(0, orderO).contiguous(), 1.0, 0.0);
+  auto packC = sort_by_key(center_slots, ctx_slots, g,
+                           center_vecs.size(0));
+  seg_accum(cd, packC[0], packC[1], ctx_vecs, nullptr, packC[2], 1.0, 0.0);
+  auto packO = sort_by_key(ctx_slots, center_slots, g, ctx_vecs.size(0));
+  seg_accum(od, packO[0], packO[1], center_vecs, nullptr, packO[2], 1.0,
+            0.0);
   return {cd, od};
 }
 
