@@ -209,6 +209,120 @@ __global__ __launch_bounds__(256) void k_pair_coeff(
   }
 }
 
+// float4-vectorized variants (dim % 4 == 0): GP lanes of float4 cover
+// the row in 16-B coalesced accesses; 3-5 shfl reductions instead of 5-6
+// and 4x fewer index loads per element.
+template <int GP>
+__global__ __launch_bounds__(256) void k_pair_coeff_v4(
+    const float4* __restrict__ A, const int64_t* __restrict__ arows,
+    const float4* __restrict__ B, const int64_t* __restrict__ brows,
+    const float* __restrict__ s_in, float* __restrict__ coeff,
+    int64_t n, int64_t dim4, int mode, float lr) {
+  constexpr int EPL = 2;
+  const int lane = threadIdx.x % GP;
+  const int64_t groups_per_block = blockDim.x / GP;
+  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
+                     threadIdx.x / GP;
+  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
+  for (int64_t g = g0; g < n; g += gstride) {
+    const int64_t ar = arows[g] * dim4, br = brows[g] * dim4;
+    float dot = 0.f;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) {
+      int c = lane + e * GP;
+      if (c < dim4) {
+        float4 a = A[ar + c], b = B[br + c];
+        dot += a.x * b.x + a.y * b.y + a.z * b.z + a.w * b.w;
+      }
+    }
+    dot = group_sum<GP>(dot);
+    if (lane == 0) {
+      if (mode == 0) coeff[g] = s_in[g] - dot;
+      else coeff[g] = lr * (s_in[g] - 1.0f / (1.0f + __expf(-dot)));
+    }
+  }
+}
+
+template <int GP, bool SELF>
+__global__ __launch_bounds__(256) void k_seg_accum_v4(
+    float4* __restrict__ out, const int64_t* __restrict__ keys,
+    const int64_t* __restrict__ slots, const float4* __restrict__ src,
+    const float4* __restrict__ self, const float* __restrict__ coeff,
+    int64_t n, int64_t dim4, float scale, float a, int chunk) {
+  constexpr int EPL = 2;
+  const int lane = threadIdx.x % GP;
+  const int64_t groups_per_block = blockDim.x / GP;
+  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
+                     threadIdx.x / GP;
+  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
+  const int64_t nchunks = (n + chunk - 1) / chunk;
+  for (int64_t cidx = g0; cidx < nchunks; cidx += gstride) {
+    const int64_t start = cidx * chunk;
+    const int64_t end = min(n, start + (int64_t)chunk);
+    float4 acc[EPL];
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) acc[e] = make_float4(0, 0, 0, 0);
+    int cnt = 0;
+    int64_t cur = keys[start];
+    for (int64_t j = start; j < end; ++j) {
+      const int64_t k = keys[j];
+      if (k != cur) {
+        const int64_t orow = cur * dim4;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e) {
+          int c = lane + e * GP;
+          if (c < dim4) {
+            float4 t = acc[e];
+            t.x *= scale; t.y *= scale; t.z *= scale; t.w *= scale;
+            if (SELF) {
+              float4 sv = self[orow + c];
+              float ac = a * cnt;
+              t.x += ac * sv.x; t.y += ac * sv.y;
+              t.z += ac * sv.z; t.w += ac * sv.w;
+            }
+            float* o = (float*)&out[orow + c];
+            atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
+            atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
+          }
+          acc[e] = make_float4(0, 0, 0, 0);
+        }
+        cnt = 0;
+        cur = k;
+      }
+      const float gj = coeff[j];
+      const int64_t srow = slots[j] * dim4;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) {
+        int c = lane + e * GP;
+        if (c < dim4) {
+          float4 sv = src[srow + c];
+          acc[e].x += gj * sv.x; acc[e].y += gj * sv.y;
+          acc[e].z += gj * sv.z; acc[e].w += gj * sv.w;
+        }
+      }
+      ++cnt;
+    }
+    const int64_t orow = cur * dim4;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) {
+      int c = lane + e * GP;
+      if (c < dim4) {
+        float4 t = acc[e];
+        t.x *= scale; t.y *= scale; t.z *= scale; t.w *= scale;
+        if (SELF) {
+          float4 sv = self[orow + c];
+          float ac = a * cnt;
+          t.x += ac * sv.x; t.y += ac * sv.y;
+          t.z += ac * sv.z; t.w += ac * sv.w;
+        }
+        float* o = (float*)&out[orow + c];
+        atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
+        atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
+      }
+    }
+  }
+}
+
 // out[key_j] += sum_j scale*coeff[j]*src[slots[j]]  (+ a*cnt*self[key])
 // over samples sorted by key; one atomic flush per segment per chunk.
 template <int GP, bool SELF>
@@ -528,8 +642,11 @@ static std::vector<torch::Tensor> bucket_pack(torch::Tensor keys,
   auto opts_i32 = torch::dtype(torch::kInt32).device(keys.device());
   auto hist = torch::zeros({m}, opts_i32);
   size_t lds1 = (size_t)m * sizeof(int);
-  hipLaunchKernelGGL(k_key_hist, dim3(grid_for(n, 256)), dim3(256), lds1,
-                     cur_stream(), keys.data_ptr<int64_t>(),
+  // few big blocks: the LDS zero + per-block global merge amortize over
+  // ~4096 keys per block instead of ~512
+  int hist_grid = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+  hipLaunchKernelGGL(k_key_hist, dim3(std::max(hist_grid, 1)), dim3(256),
+                     lds1, cur_stream(), keys.data_ptr<int64_t>(),
                      hist.data_ptr<int>(), n, m);
   auto starts = torch::cumsum(hist, 0, torch::kInt32) - hist;
   auto cursor = starts.contiguous();
@@ -572,6 +689,14 @@ static void by_group(int64_t dim, F f) {
 
 static constexpr int SEG_CHUNK = 64;
 
+template <class F>
+static void by_group4(int64_t dim4, F f) {
+  if (dim4 <= 8) f(std::integral_constant<int, 8>{});
+  else if (dim4 <= 32) f(std::integral_constant<int, 32>{});
+  else { TORCH_CHECK(dim4 <= 128, "row dim > 512 unsupported");
+         f(std::integral_constant<int, 64>{}); }
+}
+
 static void seg_accum(torch::Tensor out, torch::Tensor keys,
                       torch::Tensor slots, torch::Tensor src,
                       const torch::Tensor* self, torch::Tensor coeff,
@@ -579,6 +704,32 @@ static void seg_accum(torch::Tensor out, torch::Tensor keys,
   int64_t n = keys.numel(), dim = out.size(1);
   if (!n) return;
   int64_t nchunks = (n + SEG_CHUNK - 1) / SEG_CHUNK;
+  if (dim % 4 == 0) {
+    int64_t dim4 = dim / 4;
+    by_group4(dim4, [&](auto tag) {
+      constexpr int GP = decltype(tag)::value;
+      dim3 grid(grid_for(nchunks * GP, 256));
+      if (self) {
+        hipLaunchKernelGGL((k_seg_accum_v4<GP, true>), grid, dim3(256), 0,
+                           cur_stream(), (float4*)out.data_ptr<float>(),
+                           keys.data_ptr<int64_t>(),
+                           slots.data_ptr<int64_t>(),
+                           (const float4*)src.data_ptr<float>(),
+                           (const float4*)self->data_ptr<float>(),
+                           coeff.data_ptr<float>(), n, dim4, (float)scale,
+                           (float)a, SEG_CHUNK);
+      } else {
+        hipLaunchKernelGGL((k_seg_accum_v4<GP, false>), grid, dim3(256),
+                           0, cur_stream(), (float4*)out.data_ptr<float>(),
+                           keys.data_ptr<int64_t>(),
+                           slots.data_ptr<int64_t>(),
+                           (const float4*)src.data_ptr<float>(), nullptr,
+                           coeff.data_ptr<float>(), n, dim4, (float)scale,
+                           (float)a, SEG_CHUNK);
+      }
+    });
+    return;
+  }
   by_group(dim, [&](auto tag) {
     constexpr int GP = decltype(tag)::value;
     dim3 grid(grid_for(nchunks * GP, 256));
@@ -606,6 +757,22 @@ static void pair_coeff(torch::Tensor A, torch::Tensor arows,
                        double lr) {
   int64_t n = coeff.numel(), dim = A.size(1);
   if (!n) return;
+  if (dim % 4 == 0) {
+    int64_t dim4 = dim / 4;
+    by_group4(dim4, [&](auto tag) {
+      constexpr int GP = decltype(tag)::value;
+      hipLaunchKernelGGL((k_pair_coeff_v4<GP>),
+                         dim3(grid_for(n * GP, 256)), dim3(256), 0,
+                         cur_stream(),
+                         (const float4*)A.data_ptr<float>(),
+                         arows.data_ptr<int64_t>(),
+                         (const float4*)B.data_ptr<float>(),
+                         brows.data_ptr<int64_t>(),
+                         s_in.data_ptr<float>(), coeff.data_ptr<float>(),
+                         n, dim4, mode, (float)lr);
+    });
+    return;
+  }
   by_group(dim, [&](auto tag) {
     constexpr int GP = decltype(tag)::value;
     hipLaunchKernelGGL((k_pair_coeff<GP>),
