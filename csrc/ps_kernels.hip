@@ -212,10 +212,10 @@ __global__ __launch_bounds__(256) void k_pair_coeff(
 // float4-vectorized variants (dim % 4 == 0): GP lanes of float4 cover
 // the row in 16-B coalesced accesses; 3-5 shfl reductions instead of 5-6
 // and 4x fewer index loads per element.
-template <int GP>
+template <int GP, class TIdx>
 __global__ __launch_bounds__(256) void k_pair_coeff_v4(
-    const float4* __restrict__ A, const int64_t* __restrict__ arows,
-    const float4* __restrict__ B, const int64_t* __restrict__ brows,
+    const float4* __restrict__ A, const TIdx* __restrict__ arows,
+    const float4* __restrict__ B, const TIdx* __restrict__ brows,
     const float* __restrict__ s_in, float* __restrict__ coeff,
     int64_t n, int64_t dim4, int mode, float lr) {
   constexpr int EPL = 2;
@@ -225,7 +225,8 @@ __global__ __launch_bounds__(256) void k_pair_coeff_v4(
                      threadIdx.x / GP;
   const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
   for (int64_t g = g0; g < n; g += gstride) {
-    const int64_t ar = arows[g] * dim4, br = brows[g] * dim4;
+    const int64_t ar = (int64_t)arows[g] * dim4;
+    const int64_t br = (int64_t)brows[g] * dim4;
     float dot = 0.f;
 #pragma unroll
     for (int e = 0; e < EPL; ++e) {
@@ -243,12 +244,19 @@ __global__ __launch_bounds__(256) void k_pair_coeff_v4(
   }
 }
 
-template <int GP, bool SELF>
+// REPS > 1 spreads each flush over REPS replica copies of `out`
+// (replica = blockIdx % REPS; caller sums them afterwards): a zipfian
+// hot key otherwise serializes thousands of same-address atomic flushes.
+template <int GP, bool SELF, class TIdx>
 __global__ __launch_bounds__(256) void k_seg_accum_v4(
-    float4* __restrict__ out, const int64_t* __restrict__ keys,
-    const int64_t* __restrict__ slots, const float4* __restrict__ src,
+    float4* __restrict__ out, const TIdx* __restrict__ keys,
+    const TIdx* __restrict__ slots, const float4* __restrict__ src,
     const float4* __restrict__ self, const float* __restrict__ coeff,
-    int64_t n, int64_t dim4, float scale, float a, int chunk) {
+    int64_t n, int64_t dim4, float scale, float a, int chunk,
+    int64_t rep_stride) {
+  // rep_stride==0 -> single copy; reps fixed at 8 when used (& 7).
+  // self[] reads stay un-replicated; only the flush target spreads.
+  const int64_t rep_off = (int64_t)(blockIdx.x & 7) * rep_stride;
   constexpr int EPL = 2;
   const int lane = threadIdx.x % GP;
   const int64_t groups_per_block = blockDim.x / GP;
@@ -263,11 +271,11 @@ __global__ __launch_bounds__(256) void k_seg_accum_v4(
 #pragma unroll
     for (int e = 0; e < EPL; ++e) acc[e] = make_float4(0, 0, 0, 0);
     int cnt = 0;
-    int64_t cur = keys[start];
+    TIdx cur = keys[start];
     for (int64_t j = start; j < end; ++j) {
-      const int64_t k = keys[j];
+      const TIdx k = keys[j];
       if (k != cur) {
-        const int64_t orow = cur * dim4;
+        const int64_t orow = (int64_t)cur * dim4;
 #pragma unroll
         for (int e = 0; e < EPL; ++e) {
           int c = lane + e * GP;
@@ -280,7 +288,7 @@ __global__ __launch_bounds__(256) void k_seg_accum_v4(
               t.x += ac * sv.x; t.y += ac * sv.y;
               t.z += ac * sv.z; t.w += ac * sv.w;
             }
-            float* o = (float*)&out[orow + c];
+            float* o = (float*)&out[rep_off + orow + c];
             atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
             atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
           }
@@ -290,7 +298,7 @@ __global__ __launch_bounds__(256) void k_seg_accum_v4(
         cur = k;
       }
       const float gj = coeff[j];
-      const int64_t srow = slots[j] * dim4;
+      const int64_t srow = (int64_t)slots[j] * dim4;
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
         int c = lane + e * GP;
@@ -302,7 +310,7 @@ __global__ __launch_bounds__(256) void k_seg_accum_v4(
       }
       ++cnt;
     }
-    const int64_t orow = cur * dim4;
+    const int64_t orow = (int64_t)cur * dim4;
 #pragma unroll
     for (int e = 0; e < EPL; ++e) {
       int c = lane + e * GP;
@@ -315,7 +323,7 @@ __global__ __launch_bounds__(256) void k_seg_accum_v4(
           t.x += ac * sv.x; t.y += ac * sv.y;
           t.z += ac * sv.z; t.w += ac * sv.w;
         }
-        float* o = (float*)&out[orow + c];
+        float* o = (float*)&out[rep_off + orow + c];
         atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
         atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
       }
@@ -579,8 +587,9 @@ void hash_init_rows(torch::Tensor out, int64_t first_id, int64_t stride,
 // profiles/mf32_r01_baseline_atomics.md follow-up).  Order within a key
 // is unspecified (the consumers only need grouping, not stability).
 // ---------------------------------------------------------------------
+template <class TK>
 __global__ __launch_bounds__(256) void k_key_hist(
-    const int64_t* __restrict__ keys, int* __restrict__ ghist, int64_t n,
+    const TK* __restrict__ keys, int* __restrict__ ghist, int64_t n,
     int m) {
   extern __shared__ int lhist[];
   for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
@@ -593,10 +602,11 @@ __global__ __launch_bounds__(256) void k_key_hist(
     if (lhist[i]) atomicAdd(&ghist[i], lhist[i]);
 }
 
+template <class TK, class TP>
 __global__ __launch_bounds__(256) void k_bucket_pack(
-    const int64_t* __restrict__ keys, const int64_t* __restrict__ p0,
+    const TK* __restrict__ keys, const TP* __restrict__ p0,
     const float* __restrict__ p1, int* __restrict__ gcursor,
-    int64_t* __restrict__ okeys, int64_t* __restrict__ op0,
+    int* __restrict__ okeys, int* __restrict__ op0,
     float* __restrict__ op1, int64_t n, int m, int chunk) {
   extern __shared__ int lds[];
   int* lhist = lds;
@@ -622,7 +632,7 @@ __global__ __launch_bounds__(256) void k_bucket_pack(
       const int lpos = atomicAdd(&lhist[k], 1);
       const int64_t pos = (int64_t)lbase[k] + lpos;
       okeys[pos] = k;
-      op0[pos] = p0[j];
+      op0[pos] = (int)p0[j];
       op1[pos] = p1[j];
     }
     __syncthreads();
@@ -630,48 +640,50 @@ __global__ __launch_bounds__(256) void k_bucket_pack(
 }
 
 
-// counting-sort pack (K10) for small key ranges; returns
-// (sorted_keys, sorted_p0, sorted_p1).
+// counting-sort pack (K10) for small key ranges; returns int32
+// (sorted_keys, sorted_p0, sorted_p1) plus the histogram start offsets.
 static constexpr int PACK_MAX_M = 16384;
 static constexpr int PACK_CHUNK = 2048;
 
-static std::vector<torch::Tensor> bucket_pack(torch::Tensor keys,
-                                              torch::Tensor p0,
-                                              torch::Tensor p1, int m) {
+template <class TK, class TP>
+static std::vector<torch::Tensor> bucket_pack_t(torch::Tensor keys,
+                                                torch::Tensor p0,
+                                                torch::Tensor p1, int m) {
   int64_t n = keys.numel();
   auto opts_i32 = torch::dtype(torch::kInt32).device(keys.device());
   auto hist = torch::zeros({m}, opts_i32);
   size_t lds1 = (size_t)m * sizeof(int);
-  // few big blocks: the LDS zero + per-block global merge amortize over
-  // ~4096 keys per block instead of ~512
   int hist_grid = (int)std::min<int64_t>((n + 4095) / 4096, 512);
-  hipLaunchKernelGGL(k_key_hist, dim3(std::max(hist_grid, 1)), dim3(256),
-                     lds1, cur_stream(), keys.data_ptr<int64_t>(),
+  hipLaunchKernelGGL((k_key_hist<TK>), dim3(std::max(hist_grid, 1)),
+                     dim3(256), lds1, cur_stream(), keys.data_ptr<TK>(),
                      hist.data_ptr<int>(), n, m);
   auto starts = torch::cumsum(hist, 0, torch::kInt32) - hist;
   auto cursor = starts.contiguous();
-  auto okeys = torch::empty_like(keys);
-  auto op0 = torch::empty_like(p0);
+  auto okeys = torch::empty({n}, opts_i32);
+  auto op0 = torch::empty({n}, opts_i32);
   auto op1 = torch::empty_like(p1);
   int64_t nchunks = (n + PACK_CHUNK - 1) / PACK_CHUNK;
   int grid = (int)std::min<int64_t>(nchunks, 2048);
-  hipLaunchKernelGGL(k_bucket_pack, dim3(grid), dim3(256),
+  hipLaunchKernelGGL((k_bucket_pack<TK, TP>), dim3(grid), dim3(256),
                      2 * (size_t)m * sizeof(int), cur_stream(),
-                     keys.data_ptr<int64_t>(), p0.data_ptr<int64_t>(),
+                     keys.data_ptr<TK>(), p0.data_ptr<TP>(),
                      p1.data_ptr<float>(), cursor.data_ptr<int>(),
-                     okeys.data_ptr<int64_t>(), op0.data_ptr<int64_t>(),
+                     okeys.data_ptr<int>(), op0.data_ptr<int>(),
                      op1.data_ptr<float>(), n, m, PACK_CHUNK);
   return {okeys, op0, op1};
 }
 
-// sort (keys, int64 payload, f32 payload) by key: counting sort when the
-// key range m is small, radix sort (torch::argsort) otherwise.
+// sort (keys, index payload, f32 payload) by key: counting sort into
+// int32 when the key range m is small, int64 radix sort otherwise.
 static std::vector<torch::Tensor> sort_by_key(torch::Tensor keys,
                                               torch::Tensor p0,
                                               torch::Tensor p1,
                                               int64_t m) {
-  if (m > 0 && m <= PACK_MAX_M)
-    return bucket_pack(keys, p0, p1, (int)m);
+  if (m > 0 && m <= PACK_MAX_M) {
+    if (keys.dtype() == torch::kInt32)
+      return bucket_pack_t<int, int>(keys, p0, p1, (int)m);
+    return bucket_pack_t<int64_t, int64_t>(keys, p0, p1, (int)m);
+  }
   auto order = torch::argsort(keys);
   return {keys.index_select(0, order).contiguous(),
           p0.index_select(0, order).contiguous(),
@@ -687,8 +699,6 @@ static void by_group(int64_t dim, F f) {
          f(std::integral_constant<int, 64>{}); }
 }
 
-static constexpr int SEG_CHUNK = 64;
-
 template <class F>
 static void by_group4(int64_t dim4, F f) {
   if (dim4 <= 8) f(std::integral_constant<int, 8>{});
@@ -697,56 +707,45 @@ static void by_group4(int64_t dim4, F f) {
          f(std::integral_constant<int, 64>{}); }
 }
 
+static constexpr int SEG_CHUNK = 64;
+
+// seg_accum over key-sorted samples.  When `reps` > 1, `out` is a
+// (reps*m, dim) replica buffer the caller sums afterwards (hot-key
+// atomic spread); reps is only used on the non-SELF... both paths
+// support it via rep_stride = m*dim (SELF reads self[] un-replicated).
 static void seg_accum(torch::Tensor out, torch::Tensor keys,
                       torch::Tensor slots, torch::Tensor src,
                       const torch::Tensor* self, torch::Tensor coeff,
-                      double scale, double a) {
-  int64_t n = keys.numel(), dim = out.size(1);
+                      double scale, double a, int64_t rep_stride = 0) {
+  int64_t n = keys.numel(), dim = src.size(1);
   if (!n) return;
   int64_t nchunks = (n + SEG_CHUNK - 1) / SEG_CHUNK;
-  if (dim % 4 == 0) {
-    int64_t dim4 = dim / 4;
-    by_group4(dim4, [&](auto tag) {
-      constexpr int GP = decltype(tag)::value;
-      dim3 grid(grid_for(nchunks * GP, 256));
-      if (self) {
-        hipLaunchKernelGGL((k_seg_accum_v4<GP, true>), grid, dim3(256), 0,
-                           cur_stream(), (float4*)out.data_ptr<float>(),
-                           keys.data_ptr<int64_t>(),
-                           slots.data_ptr<int64_t>(),
-                           (const float4*)src.data_ptr<float>(),
-                           (const float4*)self->data_ptr<float>(),
-                           coeff.data_ptr<float>(), n, dim4, (float)scale,
-                           (float)a, SEG_CHUNK);
-      } else {
-        hipLaunchKernelGGL((k_seg_accum_v4<GP, false>), grid, dim3(256),
-                           0, cur_stream(), (float4*)out.data_ptr<float>(),
-                           keys.data_ptr<int64_t>(),
-                           slots.data_ptr<int64_t>(),
-                           (const float4*)src.data_ptr<float>(), nullptr,
-                           coeff.data_ptr<float>(), n, dim4, (float)scale,
-                           (float)a, SEG_CHUNK);
-      }
-    });
-    return;
-  }
-  by_group(dim, [&](auto tag) {
+  TORCH_CHECK(dim % 4 == 0, "seg_accum requires dim % 4 == 0");
+  int64_t dim4 = dim / 4;
+  bool i32 = keys.dtype() == torch::kInt32;
+  by_group4(dim4, [&](auto tag) {
     constexpr int GP = decltype(tag)::value;
     dim3 grid(grid_for(nchunks * GP, 256));
+    auto launch = [&](auto self_tag, auto idx_tag) {
+      constexpr bool SELF = decltype(self_tag)::value;
+      using TIdx = typename decltype(idx_tag)::type;
+      hipLaunchKernelGGL((k_seg_accum_v4<GP, SELF, TIdx>), grid,
+                         dim3(256), 0, cur_stream(),
+                         (float4*)out.data_ptr<float>(),
+                         keys.data_ptr<TIdx>(), slots.data_ptr<TIdx>(),
+                         (const float4*)src.data_ptr<float>(),
+                         SELF ? (const float4*)self->data_ptr<float>()
+                              : nullptr,
+                         coeff.data_ptr<float>(), n, dim4, (float)scale,
+                         (float)a, SEG_CHUNK, rep_stride / 4);
+    };
+    using TI32 = std::integral_constant<int, 0>;
     if (self) {
-      hipLaunchKernelGGL((k_seg_accum<GP, true>), grid, dim3(256), 0,
-                         cur_stream(), out.data_ptr<float>(),
-                         keys.data_ptr<int64_t>(),
-                         slots.data_ptr<int64_t>(), src.data_ptr<float>(),
-                         self->data_ptr<float>(), coeff.data_ptr<float>(),
-                         n, dim, (float)scale, (float)a, SEG_CHUNK);
+      if (i32) launch(std::true_type{}, std::common_type<int>{});
+      else launch(std::true_type{}, std::common_type<int64_t>{});
     } else {
-      hipLaunchKernelGGL((k_seg_accum<GP, false>), grid, dim3(256), 0,
-                         cur_stream(), out.data_ptr<float>(),
-                         keys.data_ptr<int64_t>(),
-                         slots.data_ptr<int64_t>(), src.data_ptr<float>(),
-                         nullptr, coeff.data_ptr<float>(), n, dim,
-                         (float)scale, (float)a, SEG_CHUNK);
+      if (i32) launch(std::false_type{}, std::common_type<int>{});
+      else launch(std::false_type{}, std::common_type<int64_t>{});
     }
   });
 }
@@ -759,20 +758,28 @@ static void pair_coeff(torch::Tensor A, torch::Tensor arows,
   if (!n) return;
   if (dim % 4 == 0) {
     int64_t dim4 = dim / 4;
+    bool i32 = arows.dtype() == torch::kInt32;
     by_group4(dim4, [&](auto tag) {
       constexpr int GP = decltype(tag)::value;
-      hipLaunchKernelGGL((k_pair_coeff_v4<GP>),
-                         dim3(grid_for(n * GP, 256)), dim3(256), 0,
-                         cur_stream(),
-                         (const float4*)A.data_ptr<float>(),
-                         arows.data_ptr<int64_t>(),
-                         (const float4*)B.data_ptr<float>(),
-                         brows.data_ptr<int64_t>(),
-                         s_in.data_ptr<float>(), coeff.data_ptr<float>(),
-                         n, dim4, mode, (float)lr);
+      auto launch = [&](auto idx_tag) {
+        using TIdx = typename decltype(idx_tag)::type;
+        hipLaunchKernelGGL((k_pair_coeff_v4<GP, TIdx>),
+                           dim3(grid_for(n * GP, 256)), dim3(256), 0,
+                           cur_stream(),
+                           (const float4*)A.data_ptr<float>(),
+                           arows.data_ptr<TIdx>(),
+                           (const float4*)B.data_ptr<float>(),
+                           brows.data_ptr<TIdx>(),
+                           s_in.data_ptr<float>(), coeff.data_ptr<float>(),
+                           n, dim4, mode, (float)lr);
+      };
+      if (i32) launch(std::common_type<int>{});
+      else launch(std::common_type<int64_t>{});
     });
     return;
   }
+  TORCH_CHECK(arows.dtype() == torch::kInt64,
+              "scalar pair_coeff path needs int64 indices");
   by_group(dim, [&](auto tag) {
     constexpr int GP = decltype(tag)::value;
     hipLaunchKernelGGL((k_pair_coeff<GP>),
@@ -782,6 +789,14 @@ static void pair_coeff(torch::Tensor A, torch::Tensor arows,
                        brows.data_ptr<int64_t>(), s_in.data_ptr<float>(),
                        coeff.data_ptr<float>(), n, dim, mode, (float)lr);
   });
+}
+
+// replica count heuristic for the hot-key delta flush
+static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
+  if (n < 8 * m) return 1;
+  int64_t r = 8;
+  if (r * m * dim * 4 > (64 << 20)) return 1;  // cap replica buffer 64 MB
+  return r;
 }
 
 // K4: e = r - u0.v computed on the item-sorted batch (exact batch-
@@ -795,20 +810,29 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   CHECK_DEV(item_vecs); CHECK_CONT(item_vecs);
   CHECK_CONT(user_rows); CHECK_CONT(item_slots); CHECK_CONT(ratings);
   int64_t n = ratings.numel(), dim = user_table.size(1);
+  int64_t m = item_vecs.size(0);
   TORCH_CHECK(item_vecs.size(1) == dim);
-  auto deltas = torch::zeros_like(item_vecs);
-  if (!n) return deltas;
-  // pass A: counting-sort by item (K10), compute e, accumulate item deltas
-  auto packA = sort_by_key(item_slots, user_rows, ratings,
-                           item_vecs.size(0));
+  if (!n) return torch::zeros_like(item_vecs);
+  // pass A: counting-sort by item (K10), compute e, accumulate deltas
+  auto packA = sort_by_key(item_slots, user_rows, ratings, m);
   auto& s_slots = packA[0];
   auto& s_urows = packA[1];
   auto& s_r = packA[2];
   auto e = torch::empty_like(s_r);
   pair_coeff(user_table, s_urows, item_vecs, s_slots, s_r, e, 0, 0.0);
-  seg_accum(deltas, s_slots, s_urows, user_table, &item_vecs, e, lr,
-            -lr * reg);
-  // pass B: counting-sort by user, apply user updates
+  int64_t reps = (dim % 4 == 0) ? delta_reps(n, m, dim) : 1;
+  torch::Tensor deltas;
+  if (reps > 1) {
+    auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
+    seg_accum(rep_buf, s_slots, s_urows, user_table, &item_vecs, e, lr,
+              -lr * reg, m * dim);
+    deltas = rep_buf.view({reps, m, dim}).sum(0);
+  } else {
+    deltas = torch::zeros_like(item_vecs);
+    seg_accum(deltas, s_slots, s_urows, user_table, &item_vecs, e, lr,
+              -lr * reg);
+  }
+  // pass B: counting-sort by user, apply user updates in place
   auto packB = sort_by_key(s_urows, s_slots, e, user_table.size(0));
   seg_accum(user_table, packB[0], packB[1], item_vecs, &user_table,
             packB[2], lr, -lr * reg);
@@ -838,6 +862,7 @@ std::vector<torch::Tensor> sgns_step(
             0.0);
   return {cd, od};
 }
+
 
 static int pa_mode(const std::string& variant) {
   if (variant == "PA") return 0;
