@@ -20,3 +20,5 @@ from .parallel.transport import (LocalTransport, LoopbackTransport,
                                  TorchDistTransport, Transport)
 from .parallel.partitioner import (ModPartitioner, RangePartitioner,
                                    make_partitioner)
+from .transform import (transform, transform_local, transform_simple,
+                        transform_with_model_load)
