@@ -102,7 +102,7 @@ def make_lr(args, rank, world, device):
     part = ModPartitioner(n_feat, world)
     table = ShardedTable(part, rank, 1, device, optimizer="adagrad",
                          lr=0.05)
-    wl = LRWorkerLogic()
+    wl = LRWorkerLogic(n_features=n_feat, device=device)
     g = torch.Generator().manual_seed(7 + rank)
     batches = []
     for _ in range(args.n_batches):
@@ -169,7 +169,8 @@ def main():
 
     cfg, part, table, wl, batches, meta = MAKERS[args.model](
         args, rank, world, device)
-    eng = PSEngine(cfg, wl, TableServerLogic(table, dump_on_close=False),
+    eng = PSEngine(cfg, wl, TableServerLogic(table, dump_on_close=False,
+                                    pushes_unique=True),
                    transport, part, device=device)
     it = cycle(batches)
 

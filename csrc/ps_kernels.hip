@@ -125,6 +125,35 @@ __global__ __launch_bounds__(256) void k_apply_add(
 }
 
 // ---------------------------------------------------------------------
+// K1 fast path: rows unique within the batch (our model logics always
+// push pre-aggregated deltas) -> plain read-add-store, no atomics.
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_apply_add_unique_vec4(
+    float4* __restrict__ table, const int64_t* __restrict__ rows,
+    const float4* __restrict__ deltas, int64_t n, int64_t dim4) {
+  int64_t total = n * dim4;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = i / dim4, c = i - j * dim4;
+    int64_t t = rows[j] * dim4 + c;
+    float4 a = table[t], d = deltas[i];
+    a.x += d.x; a.y += d.y; a.z += d.z; a.w += d.w;
+    table[t] = a;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_apply_add_unique(
+    float* __restrict__ table, const int64_t* __restrict__ rows,
+    const float* __restrict__ deltas, int64_t n, int64_t dim) {
+  int64_t total = n * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = i / dim, c = i - j * dim;
+    table[rows[j] * dim + c] += deltas[i];
+  }
+}
+
+// ---------------------------------------------------------------------
 // K2: apply_adagrad on a SORTED batch (host sorts rows, gathers grads).
 // Element (j, c) acts only if j is its row's segment leader; the leader
 // walks its segment, pre-aggregating duplicate-row gradients, then does
@@ -212,13 +241,12 @@ __global__ __launch_bounds__(256) void k_pair_coeff(
 // float4-vectorized variants (dim % 4 == 0): GP lanes of float4 cover
 // the row in 16-B coalesced accesses; 3-5 shfl reductions instead of 5-6
 // and 4x fewer index loads per element.
-template <int GP, class TIdx>
+template <int GP, int EPL, class TIdx>
 __global__ __launch_bounds__(256) void k_pair_coeff_v4(
     const float4* __restrict__ A, const TIdx* __restrict__ arows,
     const float4* __restrict__ B, const TIdx* __restrict__ brows,
     const float* __restrict__ s_in, float* __restrict__ coeff,
     int64_t n, int64_t dim4, int mode, float lr) {
-  constexpr int EPL = 2;
   const int lane = threadIdx.x % GP;
   const int64_t groups_per_block = blockDim.x / GP;
   const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
@@ -247,7 +275,7 @@ __global__ __launch_bounds__(256) void k_pair_coeff_v4(
 // REPS > 1 spreads each flush over REPS replica copies of `out`
 // (replica = blockIdx % REPS; caller sums them afterwards): a zipfian
 // hot key otherwise serializes thousands of same-address atomic flushes.
-template <int GP, bool SELF, class TIdx>
+template <int GP, int EPL, bool SELF, class TIdx>
 __global__ __launch_bounds__(256) void k_seg_accum_v4(
     float4* __restrict__ out, const TIdx* __restrict__ keys,
     const TIdx* __restrict__ slots, const float4* __restrict__ src,
@@ -257,7 +285,6 @@ __global__ __launch_bounds__(256) void k_seg_accum_v4(
   // rep_stride==0 -> single copy; reps fixed at 8 when used (& 7).
   // self[] reads stay un-replicated; only the flush target spreads.
   const int64_t rep_off = (int64_t)(blockIdx.x & 7) * rep_stride;
-  constexpr int EPL = 2;
   const int lane = threadIdx.x % GP;
   const int64_t groups_per_block = blockDim.x / GP;
   const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
@@ -398,12 +425,17 @@ __global__ __launch_bounds__(256) void k_seg_accum(
 // One wave per instance; lanes stride the instance's nnz.
 // mode: 0 = PA, 1 = PA-I, 2 = PA-II, 3 = logistic regression.
 // ---------------------------------------------------------------------
+// rep_stride > 0 spreads the delta scatter over 8 replica copies of
+// `deltas` (hot zipfian features otherwise serialize same-address
+// atomics); the host sums the replicas.
 __global__ __launch_bounds__(256) void k_sparse_linear_step(
     const int64_t* __restrict__ indptr,
     const int64_t* __restrict__ feat_slots,
     const float* __restrict__ x_vals, const float* __restrict__ labels,
     const float* __restrict__ weights, float* __restrict__ margins,
-    float* __restrict__ deltas, int64_t B, int mode, float C) {
+    float* __restrict__ deltas, int64_t B, int mode, float C,
+    int64_t rep_stride) {
+  const int64_t rep_off = (int64_t)(blockIdx.x & 7) * rep_stride;
   const int lane = threadIdx.x % WAVE;
   const int64_t waves_per_block = blockDim.x / WAVE;
   const int64_t w0 = (int64_t)blockIdx.x * waves_per_block +
@@ -438,7 +470,7 @@ __global__ __launch_bounds__(256) void k_sparse_linear_step(
       coeff = tau * y;
     }
     for (int64_t j = lo + lane; j < hi; j += WAVE)
-      atomicAdd(&deltas[feat_slots[j]], coeff * x_vals[j]);
+      atomicAdd(&deltas[rep_off + feat_slots[j]], coeff * x_vals[j]);
   }
 }
 
@@ -528,12 +560,28 @@ torch::Tensor gather(torch::Tensor table, torch::Tensor rows) {
 }
 
 void apply_add(torch::Tensor table, torch::Tensor rows,
-               torch::Tensor deltas) {
+               torch::Tensor deltas, bool unique) {
   CHECK_DEV(table); CHECK_DEV(rows); CHECK_DEV(deltas);
   CHECK_CONT(table); CHECK_CONT(rows); CHECK_CONT(deltas);
   int64_t n = rows.numel(), dim = table.size(1);
   TORCH_CHECK(deltas.size(0) == n && deltas.size(1) == dim);
   if (!n) return;
+  if (unique && dim % 4 == 0) {
+    int64_t dim4 = dim / 4;
+    hipLaunchKernelGGL(k_apply_add_unique_vec4,
+                       dim3(grid_for(n * dim4, 256)), dim3(256), 0,
+                       cur_stream(), (float4*)table.data_ptr<float>(),
+                       rows.data_ptr<int64_t>(),
+                       (const float4*)deltas.data_ptr<float>(), n, dim4);
+    return;
+  }
+  if (unique) {
+    hipLaunchKernelGGL(k_apply_add_unique, dim3(grid_for(n * dim, 256)),
+                       dim3(256), 0, cur_stream(),
+                       table.data_ptr<float>(), rows.data_ptr<int64_t>(),
+                       deltas.data_ptr<float>(), n, dim);
+    return;
+  }
   hipLaunchKernelGGL(k_apply_add, dim3(grid_for(n * dim, 256)), dim3(256),
                      0, cur_stream(), table.data_ptr<float>(),
                      rows.data_ptr<int64_t>(), deltas.data_ptr<float>(),
@@ -699,12 +747,21 @@ static void by_group(int64_t dim, F f) {
          f(std::integral_constant<int, 64>{}); }
 }
 
+// (GP, EPL) choice: EPL*GP >= dim4 with GP as small as practical —
+// a wave then holds 64/GP independent samples, which is what hides the
+// dependent slot->row load latency of the segment walk (measured: GP=64
+// on dim=300 left one sample in flight per wave, 1.4 ms per SGNS
+// seg_accum; see profiles/).
 template <class F>
 static void by_group4(int64_t dim4, F f) {
-  if (dim4 <= 8) f(std::integral_constant<int, 8>{});
-  else if (dim4 <= 32) f(std::integral_constant<int, 32>{});
+  using namespace std;
+  if (dim4 <= 8) f(integral_constant<int, 8>{}, integral_constant<int, 1>{});
+  else if (dim4 <= 16) f(integral_constant<int, 8>{}, integral_constant<int, 2>{});
+  else if (dim4 <= 32) f(integral_constant<int, 8>{}, integral_constant<int, 4>{});
+  else if (dim4 <= 48) f(integral_constant<int, 16>{}, integral_constant<int, 3>{});
+  else if (dim4 <= 80) f(integral_constant<int, 16>{}, integral_constant<int, 5>{});
   else { TORCH_CHECK(dim4 <= 128, "row dim > 512 unsupported");
-         f(std::integral_constant<int, 64>{}); }
+         f(integral_constant<int, 32>{}, integral_constant<int, 4>{}); }
 }
 
 static constexpr int SEG_CHUNK = 64;
@@ -723,13 +780,14 @@ static void seg_accum(torch::Tensor out, torch::Tensor keys,
   TORCH_CHECK(dim % 4 == 0, "seg_accum requires dim % 4 == 0");
   int64_t dim4 = dim / 4;
   bool i32 = keys.dtype() == torch::kInt32;
-  by_group4(dim4, [&](auto tag) {
+  by_group4(dim4, [&](auto tag, auto etag) {
     constexpr int GP = decltype(tag)::value;
+    constexpr int EPL = decltype(etag)::value;
     dim3 grid(grid_for(nchunks * GP, 256));
     auto launch = [&](auto self_tag, auto idx_tag) {
       constexpr bool SELF = decltype(self_tag)::value;
       using TIdx = typename decltype(idx_tag)::type;
-      hipLaunchKernelGGL((k_seg_accum_v4<GP, SELF, TIdx>), grid,
+      hipLaunchKernelGGL((k_seg_accum_v4<GP, EPL, SELF, TIdx>), grid,
                          dim3(256), 0, cur_stream(),
                          (float4*)out.data_ptr<float>(),
                          keys.data_ptr<TIdx>(), slots.data_ptr<TIdx>(),
@@ -759,11 +817,12 @@ static void pair_coeff(torch::Tensor A, torch::Tensor arows,
   if (dim % 4 == 0) {
     int64_t dim4 = dim / 4;
     bool i32 = arows.dtype() == torch::kInt32;
-    by_group4(dim4, [&](auto tag) {
+    by_group4(dim4, [&](auto tag, auto etag) {
       constexpr int GP = decltype(tag)::value;
+      constexpr int EPL = decltype(etag)::value;
       auto launch = [&](auto idx_tag) {
         using TIdx = typename decltype(idx_tag)::type;
-        hipLaunchKernelGGL((k_pair_coeff_v4<GP, TIdx>),
+        hipLaunchKernelGGL((k_pair_coeff_v4<GP, EPL, TIdx>),
                            dim3(grid_for(n * GP, 256)), dim3(256), 0,
                            cur_stream(),
                            (const float4*)A.data_ptr<float>(),
@@ -881,16 +940,20 @@ std::vector<torch::Tensor> pa_step(torch::Tensor indptr,
   CHECK_CONT(indptr); CHECK_CONT(feat_slots); CHECK_CONT(x_vals);
   int64_t B = indptr.numel() - 1;
   auto margins = torch::zeros({B}, x_vals.options());
-  auto deltas = torch::zeros_like(weights);
-  if (!B) return {margins, deltas};
+  if (!B) return {margins, torch::zeros_like(weights)};
+  int64_t m = weights.numel();
+  int64_t reps = (x_vals.numel() >= 8 * m &&
+                  8 * m * 4 <= (256ll << 20)) ? 8 : 1;
+  auto dbuf = torch::zeros({reps * m}, weights.options());
   hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
                      dim3(256), 0, cur_stream(),
                      indptr.data_ptr<int64_t>(),
                      feat_slots.data_ptr<int64_t>(),
                      x_vals.data_ptr<float>(), labels.data_ptr<float>(),
                      weights.data_ptr<float>(), margins.data_ptr<float>(),
-                     deltas.data_ptr<float>(), B, pa_mode(variant),
-                     (float)C);
+                     dbuf.data_ptr<float>(), B, pa_mode(variant),
+                     (float)C, reps > 1 ? m : 0);
+  auto deltas = reps > 1 ? dbuf.view({reps, m}).sum(0) : dbuf;
   return {margins, deltas};
 }
 
@@ -902,15 +965,20 @@ std::vector<torch::Tensor> lr_step(torch::Tensor indptr,
   CHECK_DEV(weights); CHECK_CONT(weights);
   int64_t B = indptr.numel() - 1;
   auto probs = torch::zeros({B}, x_vals.options());
-  auto grads = torch::zeros_like(weights);
-  if (!B) return {probs, grads};
+  if (!B) return {probs, torch::zeros_like(weights)};
+  int64_t m = weights.numel();
+  int64_t reps = (x_vals.numel() >= 8 * m &&
+                  8 * m * 4 <= (256ll << 20)) ? 8 : 1;
+  auto gbuf = torch::zeros({reps * m}, weights.options());
   hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
                      dim3(256), 0, cur_stream(),
                      indptr.data_ptr<int64_t>(),
                      feat_slots.data_ptr<int64_t>(),
                      x_vals.data_ptr<float>(), labels.data_ptr<float>(),
                      weights.data_ptr<float>(), probs.data_ptr<float>(),
-                     grads.data_ptr<float>(), B, /*mode=*/3, 0.f);
+                     gbuf.data_ptr<float>(), B, /*mode=*/3, 0.f,
+                     reps > 1 ? m : 0);
+  auto grads = reps > 1 ? gbuf.view({reps, m}).sum(0) : gbuf;
   // kernel writes margins; convert to probabilities in-place
   probs.sigmoid_();
   return {probs, grads};
@@ -939,7 +1007,9 @@ std::vector<torch::Tensor> pa_multiclass_step(
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather", &gather, "K3 ps_gather");
-  m.def("apply_add", &apply_add, "K1 ps_apply_sgd (delta add)");
+  m.def("apply_add", &apply_add, "K1 ps_apply_sgd (delta add)",
+        py::arg("table"), py::arg("rows"), py::arg("deltas"),
+        py::arg("unique") = false);
   m.def("apply_adagrad_sorted", &apply_adagrad_sorted,
         "K2 ps_apply_adagrad over a row-sorted batch");
   m.def("hash_init", &hash_init, "K6 deterministic per-id init");

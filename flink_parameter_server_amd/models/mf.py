@@ -27,7 +27,7 @@ import torch
 
 from .. import ops
 from ..api import ParameterServerClient, WorkerLogic
-from .common import PullBuffer
+from .common import DenseCoalescer, PullBuffer
 
 
 class MFBatch:
@@ -78,10 +78,9 @@ class OnlineMFWorkerLogic(WorkerLogic):
         # the item-id space is known and small enough for a dense flag
         # array (measured: rocPRIM sort-unique was ~7% of a 1M-rating
         # round — profiles/mf32_r01_baseline_atomics.md).
-        self._present: Optional[torch.Tensor] = None
-        if 0 < self.n_items <= (1 << 24):
-            self._present = torch.zeros(self.n_items, dtype=torch.bool,
-                                        device=self.device)
+        self._coalescer: Optional[DenseCoalescer] = None
+        if 0 < self.n_items <= DenseCoalescer.MAX_SPACE:
+            self._coalescer = DenseCoalescer(self.n_items, self.device)
 
     def batch_size(self, batch: MFBatch) -> int:
         return len(batch)
@@ -89,12 +88,8 @@ class OnlineMFWorkerLogic(WorkerLogic):
     def on_recv(self, data: MFBatch, client: ParameterServerClient) -> None:
         if self.neg_rate > 0:
             data = self._with_negatives(data)
-        if self._present is not None:
-            self._present.zero_()
-            self._present[data.item_ids] = True
-            lookup = torch.cumsum(self._present, 0) - 1
-            uids = self._present.nonzero().reshape(-1)
-            slots = lookup[data.item_ids]
+        if self._coalescer is not None:
+            uids, slots = self._coalescer(data.item_ids)
             buf = PullBuffer(uids, self.n_factors, assume_sorted=True)
         else:
             uids = torch.unique(data.item_ids)

@@ -19,7 +19,7 @@ import torch
 
 from .. import ops
 from ..api import ParameterServerClient, WorkerLogic
-from .common import PullBuffer
+from .common import DenseCoalescer, PullBuffer
 
 
 class SGNSBatch:
@@ -47,8 +47,9 @@ class SGNSWorkerLogic(WorkerLogic):
         self._seed = seed
         self._counter = 0
         # pending: (center_ids, ctx_table_ids per pair row, labels, buffer)
-        self._pending: List[Tuple[torch.Tensor, torch.Tensor, torch.Tensor,
-                                  PullBuffer]] = []
+        self._pending: List[tuple] = []
+        self._coalescer: Optional[DenseCoalescer] = None
+        self._device: Optional[torch.device] = None
 
     def batch_size(self, batch: SGNSBatch) -> int:
         return len(batch)
@@ -70,27 +71,47 @@ class SGNSWorkerLogic(WorkerLogic):
         labels = torch.zeros(n * (k + 1), dtype=torch.float32, device=dev)
         labels.view(n, k + 1)[:, 0] = 1.0
         ctx_table_ids = ctx + self.V  # context rows live at [V, 2V)
-        uids = torch.unique(torch.cat([centers, ctx_table_ids]))
-        buf = PullBuffer(uids, self.dim)
-        self._pending.append((centers, ctx_table_ids, labels, buf))
-        client.pull(uids)
+        allids = torch.cat([centers, ctx_table_ids])
+        if self._coalescer is None and 2 * self.V <= DenseCoalescer.MAX_SPACE:
+            self._coalescer = DenseCoalescer(2 * self.V, dev)
+        if self._coalescer is not None:
+            uids, slots = self._coalescer(allids)
+            c_slots, o_slots = slots[:n * (k + 1)], slots[n * (k + 1):]
+        else:
+            uids = torch.unique(allids)
+            c_slots = o_slots = None
+        buf = PullBuffer(uids, self.dim, assume_sorted=True)
+        self._pending.append((centers, ctx_table_ids, labels, buf,
+                              c_slots, o_slots))
+        client.pull(uids, unique=True)
 
     def on_pull_recv(self, ids: torch.Tensor, values: torch.Tensor,
                      client: ParameterServerClient) -> None:
+        if (len(self._pending) == 1 and
+                self._pending[0][3]._n_have == 0 and
+                ids.numel() == self._pending[0][3].uids.numel()):
+            centers, ctx_ids, labels, buf, cs, os_ = self._pending[0]
+            buf.offer_exact(ids, values)
+            self._pending = []
+            self._train(centers, ctx_ids, labels, buf, client, cs, os_)
+            return
         still = []
-        for centers, ctx_ids, labels, buf in self._pending:
+        for centers, ctx_ids, labels, buf, cs, os_ in self._pending:
             buf.offer(ids, values)
             if buf.complete:
-                self._train(centers, ctx_ids, labels, buf, client)
+                self._train(centers, ctx_ids, labels, buf, client, cs, os_)
             else:
-                still.append((centers, ctx_ids, labels, buf))
+                still.append((centers, ctx_ids, labels, buf, cs, os_))
         self._pending = still
 
     def _train(self, centers: torch.Tensor, ctx_ids: torch.Tensor,
                labels: torch.Tensor, buf: PullBuffer,
-               client: ParameterServerClient) -> None:
-        c_slots = buf.slots_of(centers)
-        o_slots = buf.slots_of(ctx_ids)
+               client: ParameterServerClient,
+               c_slots: Optional[torch.Tensor] = None,
+               o_slots: Optional[torch.Tensor] = None) -> None:
+        if c_slots is None:
+            c_slots = buf.slots_of(centers)
+            o_slots = buf.slots_of(ctx_ids)
         # One buffer holds both center and context rows; slot spaces are
         # disjoint ids so the deltas can share one accumulator.
         c_deltas, o_deltas = ops.sgns_step(buf.values, c_slots, buf.values,

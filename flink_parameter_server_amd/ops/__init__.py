@@ -32,11 +32,11 @@ def gather(table: torch.Tensor, rows: torch.Tensor) -> torch.Tensor:
 
 
 def apply_add(table: torch.Tensor, rows: torch.Tensor,
-              deltas: torch.Tensor) -> None:
+              deltas: torch.Tensor, unique: bool = False) -> None:
     if _cuda(table):
-        hip.ext().apply_add(table, rows, deltas)
+        hip.ext().apply_add(table, rows, deltas, unique)
         return
-    reference.apply_add(table, rows, deltas)
+    reference.apply_add(table, rows, deltas, unique)
 
 
 def apply_adagrad(table: torch.Tensor, accum: torch.Tensor,

@@ -81,8 +81,10 @@ def gather(table: torch.Tensor, rows: torch.Tensor) -> torch.Tensor:
 
 
 def apply_add(table: torch.Tensor, rows: torch.Tensor,
-              deltas: torch.Tensor) -> None:
-    """table[rows[j]] += deltas[j], duplicate rows accumulate (K1)."""
+              deltas: torch.Tensor, unique: bool = False) -> None:
+    """table[rows[j]] += deltas[j], duplicate rows accumulate (K1).
+    ``unique=True`` asserts rows are distinct (enables the non-atomic
+    device fast path; semantics identical)."""
     table.index_add_(0, rows, deltas.to(table.dtype))
 
 

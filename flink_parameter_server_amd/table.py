@@ -76,10 +76,11 @@ class ShardedTable:
         rows = self.part.local_row(ids)
         return ops.gather(self.data, rows)
 
-    def apply(self, ids: torch.Tensor, deltas: torch.Tensor) -> None:
+    def apply(self, ids: torch.Tensor, deltas: torch.Tensor,
+              unique: bool = False) -> None:
         rows = self.part.local_row(ids)
         if self.optimizer == "add":
-            ops.apply_add(self.data, rows, deltas)
+            ops.apply_add(self.data, rows, deltas, unique)
         else:
             ops.apply_adagrad(self.data, self.accum, rows, deltas,
                               self.lr, self.eps)
@@ -107,16 +108,22 @@ class TableServerLogic:
     (or the Adagrad rule).
     """
 
-    def __init__(self, table: ShardedTable, dump_on_close: bool = True):
+    def __init__(self, table: ShardedTable, dump_on_close: bool = True,
+                 pushes_unique: bool = False):
+        """``pushes_unique=True``: the application guarantees every push
+        batch carries distinct ids (all bundled model logics do — they
+        pre-aggregate deltas per pulled row), enabling the non-atomic
+        apply fast path."""
         self.table = table
         self.dump_on_close = dump_on_close
+        self.pushes_unique = pushes_unique
 
     def on_pull_recv(self, ids: torch.Tensor, worker: int, ps) -> None:
         ps.answer_pull(ids, self.table.gather(ids), worker)
 
     def on_push_recv(self, ids: torch.Tensor, deltas: torch.Tensor,
                      ps) -> None:
-        self.table.apply(ids, deltas)
+        self.table.apply(ids, deltas, unique=self.pushes_unique)
 
     def close(self, ps) -> None:
         if self.dump_on_close:
