@@ -194,6 +194,7 @@ class PSEngine:
         recv = self.transport.alltoallv(send, [
             (torch.int64, ()), (torch.int64, ()),
             (torch.float32, (dim,))])
+        recv = self._to_device(recv)
         r_pull, r_push_ids, r_push_deltas = recv
 
         # 5. server role: pushes first (fresher answers), then pulls
@@ -212,9 +213,9 @@ class PSEngine:
         ans_vals = [_concat([v for _, v in self._answer_buf[w]])
                     for w in range(self.world)]
         self._answer_buf = [[] for _ in range(self.world)]
-        back = self.transport.alltoallv(
+        back = self._to_device(self.transport.alltoallv(
             [ans_ids, ans_vals],
-            [(torch.int64, ()), (torch.float32, (dim,))])
+            [(torch.int64, ()), (torch.float32, (dim,))]))
         for s in range(self.world):
             if back[0][s] is not None and back[0][s].numel():
                 self._pending_answers.append((back[0][s], back[1][s]))
@@ -233,6 +234,19 @@ class PSEngine:
             else int(ctrl[:, 0].max())
         done = bool(ctrl[:, 1].all())
         return exhausted, done
+
+    # ------------------------------------------------------------------
+    def _to_device(self, recv):
+        """Move received segments onto the engine device (no-op when the
+        transport already delivered them there, e.g. RCCL; a CPU-side
+        transport like gloo hands back host tensors while compute runs on
+        the GPU)."""
+        dev = self.device
+        for chan in recv:
+            for i, t in enumerate(chan):
+                if t is not None and t.device != dev:
+                    chan[i] = t.to(dev, non_blocking=True)
+        return recv
 
     # ------------------------------------------------------------------
     def _queued_pulls(self) -> int:
