@@ -31,10 +31,11 @@ def _job(rank, world, port, outdir):
 
         dev = torch.device("cuda", 0)
         n_users, n_items, rank_k = 100, 80, 8
-        batches, _, _ = synth.planted_mf_ratings(n_users, n_items, rank_k,
-                                                 6000, 512,
-                                                 seed=31 + rank,
-                                                 device=dev)
+        # ONE planted teacher (same seed on every rank); each rank
+        # trains on its half of the ratings
+        all_batches, _, _ = synth.planted_mf_ratings(
+            n_users, n_items, rank_k, 12000, 512, seed=31, device=dev)
+        batches = all_batches[rank::world]
         part = ModPartitioner(n_items, world)
         table = ShardedTable(part, rank, rank_k, dev, init=(-0.1, 0.1))
         wl = OnlineMFWorkerLogic(n_users, rank_k, learning_rate=0.05,
@@ -55,7 +56,7 @@ def _job(rank, world, port, outdir):
                 n += len(b)
             return tot / n
 
-        eng.run(epoch_iterator(batches, 6))
+        eng.run(epoch_iterator(batches, 8))
         torch.cuda.synchronize()
 
         # reassemble the full item table across ranks via gloo
@@ -82,4 +83,4 @@ def test_two_rank_mf_on_gpu_converges():
                              weights_only=False)
             # planted model var ~ 1/rank_k = 0.125; require a real fit
             assert res["loss"] < 0.05, res
-            assert res["metrics"]["samples"] == 6000 * 6
+            assert res["metrics"]["samples"] == 6000 * 8
