@@ -22,3 +22,8 @@ from .parallel.partitioner import (ModPartitioner, RangePartitioner,
                                    make_partitioner)
 from .transform import (transform, transform_local, transform_simple,
                         transform_with_model_load)
+from .messaging import (CountFlush, FlushAlways, FlushStrategy, Pull,
+                        PullAnswer, Push, PSToWorker, SerializingTransport,
+                        TimeFlush, WorkerToPS)
+from .algorithms import (ps_offline_mf, ps_online_mf, transform_binary,
+                         transform_multiclass)

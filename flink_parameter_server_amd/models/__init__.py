@@ -1,4 +1,5 @@
-from .mf import MFBatch, OnlineMFWorkerLogic, epoch_iterator
+from .mf import (MFBatch, OnlineMFWorkerLogic, Rating,
+                 SGDUpdater, epoch_iterator)
 from .pa import PAWorkerLogic, SparseBatch
 from .lr import LRWorkerLogic
 from .sgns import SGNSBatch, SGNSWorkerLogic

@@ -21,13 +21,42 @@ RNG so no host RNG state is carried.
 """
 from __future__ import annotations
 
-from typing import Iterator, List, Optional, Tuple
+from typing import Iterator, List, NamedTuple, Optional, Tuple
 
 import torch
 
 from .. import ops
 from ..api import ParameterServerClient, WorkerLogic
 from .common import DenseCoalescer, PullBuffer
+
+
+class Rating(NamedTuple):
+    """Reference: ``matrix/factorization/utils/Rating.scala`` —
+    ``Rating(user: Int, item: Int, rating: Double)`` (ids widened to
+    int)."""
+    user: int
+    item: int
+    rating: float
+
+
+class SGDUpdater:
+    """Record-mode parity twin of the fused HIP step (K4) — reference:
+    ``matrix/factorization/utils/SGDUpdater.scala``:
+    ``delta(rating, userVec, itemVec)`` returns ``(userDelta, itemDelta)``
+    with ``e = r - u.i``, ``du = lr*(e*i - reg*u)``, ``di = lr*(e*u -
+    reg*i)`` (reg = 0 matches the reference's plain rule)."""
+
+    def __init__(self, learning_rate: float, reg: float = 0.0):
+        self.lr = float(learning_rate)
+        self.reg = float(reg)
+
+    def delta(self, rating: float, user_vec: torch.Tensor,
+              item_vec: torch.Tensor
+              ) -> Tuple[torch.Tensor, torch.Tensor]:
+        e = float(rating) - float(user_vec @ item_vec)
+        du = self.lr * (e * item_vec - self.reg * user_vec)
+        di = self.lr * (e * user_vec - self.reg * item_vec)
+        return du, di
 
 
 class MFBatch:

@@ -32,16 +32,20 @@ class SparseBatch:
 
     labels: binary — f32 in {-1,+1}, NaN = unlabeled (predict);
             multiclass — int64 class index, -1 = unlabeled.
+    ids: optional per-instance global ids carried through to prediction
+         outputs (None = outputs use batch-local row indices).
     """
 
-    __slots__ = ("indptr", "feat_ids", "x_vals", "labels")
+    __slots__ = ("indptr", "feat_ids", "x_vals", "labels", "ids")
 
     def __init__(self, indptr: torch.Tensor, feat_ids: torch.Tensor,
-                 x_vals: torch.Tensor, labels: torch.Tensor):
+                 x_vals: torch.Tensor, labels: torch.Tensor,
+                 ids: Optional[torch.Tensor] = None):
         self.indptr = indptr
         self.feat_ids = feat_ids
         self.x_vals = x_vals
         self.labels = labels
+        self.ids = ids
 
     def __len__(self) -> int:
         return int(self.indptr.numel()) - 1
@@ -96,8 +100,10 @@ class PAWorkerLogic(WorkerLogic):
                 unlabeled = torch.isnan(batch.labels)
                 if bool(unlabeled.any()):
                     pred = torch.sign(margins[unlabeled])
-                    client.output((torch.nonzero(unlabeled).reshape(-1),
-                                   pred))
+                    rows = torch.nonzero(unlabeled).reshape(-1)
+                    out_ids = batch.ids[rows] if batch.ids is not None \
+                        else rows
+                    client.output((out_ids, pred))
             client.push(buf.uids, deltas.reshape(-1, 1))
         else:
             scores, deltas = ops.pa_multiclass_step(
@@ -107,6 +113,8 @@ class PAWorkerLogic(WorkerLogic):
                 unlabeled = batch.labels < 0
                 if bool(unlabeled.any()):
                     pred = scores[unlabeled].argmax(dim=1)
-                    client.output((torch.nonzero(unlabeled).reshape(-1),
-                                   pred))
+                    rows = torch.nonzero(unlabeled).reshape(-1)
+                    out_ids = batch.ids[rows] if batch.ids is not None \
+                        else rows
+                    client.output((out_ids, pred))
             client.push(buf.uids, deltas)

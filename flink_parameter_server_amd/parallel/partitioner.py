@@ -58,6 +58,8 @@ class ModPartitioner(Partitioner):
         return rows * self.n_shards + shard
 
     def shard_rows(self, shard: int) -> int:
+        if shard >= self.n_shards:  # rank hosts no shard (ps_parallelism
+            return 0                # < world — see algorithms.py)
         return (self.num_ids - shard + self.n_shards - 1) // self.n_shards
 
 
@@ -81,6 +83,8 @@ class RangePartitioner(Partitioner):
         return rows + shard * self.block
 
     def shard_rows(self, shard: int) -> int:
+        if shard >= self.n_shards:
+            return 0
         lo = shard * self.block
         hi = min(self.num_ids, (shard + 1) * self.block)
         return max(0, hi - lo)

@@ -50,8 +50,12 @@ class Metrics:
     push_ids_sent: int = 0
     ssp_stalls: int = 0
     bytes_sent: int = 0
+    held_rounds: int = 0
+    # staleness distribution (clock - min cross-worker clock, per round) —
+    # the SSP staleness histogram BASELINE configs[4] reports.
+    staleness_hist: Dict[int, int] = field(default_factory=dict)
 
-    def as_dict(self) -> Dict[str, int]:
+    def as_dict(self) -> Dict[str, Any]:
         return dict(self.__dict__)
 
 
@@ -97,8 +101,17 @@ class PSEngine:
     def __init__(self, config: PSConfig, worker_logic: WorkerLogic,
                  server_logic: ServerLogic, transport: Optional[Transport],
                  part: Optional[Partitioner] = None,
-                 device: Optional[torch.device] = None):
+                 device: Optional[torch.device] = None,
+                 flush: Optional[Any] = None,
+                 checkpointer: Optional[Any] = None):
+        """``flush``: a messaging.FlushStrategy deciding when accumulated
+        pulls/pushes are released into a round (None = every round, the
+        reference's simple senders).  ``checkpointer``: an object with
+        ``maybe(round_idx)`` called after each round's server phase (e.g.
+        utils.checkpoint.PeriodicCheckpointer)."""
         self.cfg = config
+        self.flush = flush
+        self.checkpointer = checkpointer
         self.worker_logic = worker_logic
         self.server_logic = server_logic
         self.transport = transport or LocalTransport()
@@ -163,8 +176,10 @@ class PSEngine:
             self.worker_logic.on_pull_recv(ids, vals, self.client)
 
         # 2. consume data (unless SSP-stalled / exhausted / over limit)
+        lag = self._clock - self._min_clock
+        m.staleness_hist[lag] = m.staleness_hist.get(lag, 0) + 1
         stalled = (self.cfg.staleness is not None and
-                   self._clock - self._min_clock > self.cfg.staleness)
+                   lag > self.cfg.staleness)
         backpressure = (self._pull_limit is not None and
                         self._queued_pulls() >= self._pull_limit)
         if stalled:
@@ -182,19 +197,29 @@ class PSEngine:
                     m.samples += _batch_size(self.worker_logic, batch)
                     self.worker_logic.on_recv(batch, self.client)
 
-        # 3. coalesce + bucket
-        pull_ids = self._take_pulls()
-        push_ids, push_deltas = self._take_pushes()
+        # 3. coalesce + bucket (unless a combination-sender strategy holds
+        # this round's ops back — messaging.FlushStrategy)
+        if self.flush is not None and not self.flush.should_flush(
+                self._pending_ops(), force=exhausted or stalled):
+            m.held_rounds += 1
+            pull_ids = push_ids = push_deltas = None
+        else:
+            pull_ids = self._take_pulls()
+            push_ids, push_deltas = self._take_pushes()
         send = self._bucket(pull_ids, push_ids, push_deltas)
         m.pull_ids_sent += int(pull_ids.numel()) if pull_ids is not None else 0
         m.push_ids_sent += int(push_ids.numel()) if push_ids is not None else 0
 
         # 4. exchange worker->server segments
         dim = self.cfg.dim
+        wire = torch.bfloat16 if self.cfg.wire_dtype == "bf16" \
+            else torch.float32
+        self._wire_out(send[2], wire)
         recv = self.transport.alltoallv(send, [
             (torch.int64, ()), (torch.int64, ()),
-            (torch.float32, (dim,))])
+            (wire, (dim,))])
         recv = self._to_device(recv)
+        self._unwire(recv[2])
         r_pull, r_push_ids, r_push_deltas = recv
 
         # 5. server role: pushes first (fresher answers), then pulls
@@ -213,12 +238,17 @@ class PSEngine:
         ans_vals = [_concat([v for _, v in self._answer_buf[w]])
                     for w in range(self.world)]
         self._answer_buf = [[] for _ in range(self.world)]
+        self._wire_out(ans_vals, wire)
         back = self._to_device(self.transport.alltoallv(
             [ans_ids, ans_vals],
-            [(torch.int64, ()), (torch.float32, (dim,))]))
+            [(torch.int64, ()), (wire, (dim,))]))
+        self._unwire(back[1])
         for s in range(self.world):
             if back[0][s] is not None and back[0][s].numel():
                 self._pending_answers.append((back[0][s], back[1][s]))
+
+        if self.checkpointer is not None:
+            self.checkpointer.maybe(m.rounds)
 
         # 7. control: [clock, worker_done, exhausted].  An exhausted
         # worker's clock freezes, so it must not hold the SSP min-clock
@@ -247,6 +277,33 @@ class PSEngine:
                 if t is not None and t.device != dev:
                     chan[i] = t.to(dev, non_blocking=True)
         return recv
+
+    # ------------------------------------------------------------------
+    def _pending_ops(self) -> int:
+        """Accumulated pull+push ops a flush strategy weighs against its
+        threshold (the reference's CombinationLogic count condition)."""
+        n = sum(int(t.numel()) for t, _ in self._pull_buf)
+        n += sum(int(t.numel()) for t, _ in self._push_buf)
+        if self._pull_queue is not None:
+            n += int(self._pull_queue.numel())
+        return n
+
+    @staticmethod
+    def _wire_out(chan: List[Optional[torch.Tensor]],
+                  wire: torch.dtype) -> None:
+        """Cast a payload channel to the wire dtype in place (bf16 halves
+        xGMI bytes; tables keep f32 masters, so only the wire narrows)."""
+        if wire is torch.float32:
+            return
+        for d, t in enumerate(chan):
+            if t is not None:
+                chan[d] = t.to(wire)
+
+    @staticmethod
+    def _unwire(chan: List[Optional[torch.Tensor]]) -> None:
+        for s, t in enumerate(chan):
+            if t is not None and t.dtype is not torch.float32:
+                chan[s] = t.float()
 
     # ------------------------------------------------------------------
     def _queued_pulls(self) -> int:
@@ -296,7 +353,7 @@ class PSEngine:
             if W == 1:
                 pulls[0] = pull_ids
             else:
-                order, counts, _ = bucket_by_shard(pull_ids, self.part)
+                order, counts, _ = self._shard_counts(pull_ids)
                 sorted_ids = pull_ids[order]
                 off = 0
                 for d in range(W):
@@ -308,7 +365,7 @@ class PSEngine:
             if W == 1:
                 pids[0], pdel[0] = push_ids, push_deltas
             else:
-                order, counts, _ = bucket_by_shard(push_ids, self.part)
+                order, counts, _ = self._shard_counts(push_ids)
                 sorted_ids = push_ids[order]
                 sorted_del = push_deltas[order]
                 off = 0
@@ -319,6 +376,16 @@ class PSEngine:
                         pdel[d] = sorted_del[off:off + n]
                     off += n
         return [pulls, pids, pdel]
+
+    def _shard_counts(self, ids: torch.Tensor):
+        """bucket_by_shard with counts padded to world size (the
+        partitioner may define fewer shards than ranks when
+        ps_parallelism < worker_parallelism — algorithms.py)."""
+        order, counts, shards = bucket_by_shard(ids, self.part)
+        if counts.numel() < self.world:
+            counts = torch.cat([counts, counts.new_zeros(
+                self.world - counts.numel())])
+        return order, counts, shards
 
 
 def _batch_size(logic: WorkerLogic, batch: Any) -> int:

@@ -43,14 +43,17 @@ def transform(data: Iterable[Any], worker_logic: WorkerLogic,
               transport: Optional[Transport] = None,
               part: Optional[Partitioner] = None,
               device: Optional[torch.device] = None,
-              max_rounds: Optional[int] = None
+              max_rounds: Optional[int] = None,
+              flush: Optional[Any] = None,
+              checkpointer: Optional[Any] = None
               ) -> Tuple[List[Any], List[Any]]:
     """The base ``FlinkParameterServer.transform`` shape: run
     ``worker_logic`` against ``server_logic`` over ``data``; returns
-    (worker outputs, server outputs)."""
+    (worker outputs, server outputs).  ``flush``/``checkpointer`` as in
+    :class:`PSEngine`."""
     cfg = config or PSConfig()
     eng = PSEngine(cfg, worker_logic, server_logic, transport, part,
-                   device=device)
+                   device=device, flush=flush, checkpointer=checkpointer)
     return eng.run(data, max_rounds=max_rounds)
 
 

@@ -67,6 +67,7 @@ def _job(rank, world, port, outdir):
         table_full = full.to(dev)
         final_loss = loss()
         torch.save({"loss": final_loss,
+                    "n_local": sum(len(b) for b in batches),
                     "metrics": eng.metrics.as_dict()},
                    os.path.join(outdir, f"rank{rank}.pt"))
     finally:
@@ -83,4 +84,5 @@ def test_two_rank_mf_on_gpu_converges():
                              weights_only=False)
             # planted model var ~ 1/rank_k = 0.125; require a real fit
             assert res["loss"] < 0.05, res
-            assert res["metrics"]["samples"] == 6000 * 8
+            # every epoch consumed every local sample exactly once
+            assert res["metrics"]["samples"] == res["n_local"] * 8
