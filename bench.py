@@ -198,6 +198,8 @@ def main():
     elapsed = float(stats[:, 0].max()) / 1e9
     samples = int(stats[:, 1].sum())
     pp_ops = int(stats[:, 2].sum() + stats[:, 3].sum())
+    xgmi = transport.allgather_int([eng.metrics.bytes_sent])
+    xgmi_bytes = int(xgmi.sum())
 
     if rank == 0:
         k = meta.get("factors", cfg.dim)
@@ -217,6 +219,12 @@ def main():
             "dtype": "f32",
             "data": "synthetic",
             "pull_push_ops_per_sec": pp_ops / elapsed,
+            # whole-job bytes this node's ranks put on the wire (0 when
+            # world=1: all traffic short-circuits on-device)
+            "xgmi_bytes_total": xgmi_bytes,
+            # SSP staleness distribution {lag_rounds: count} (configs[4])
+            "staleness_hist": (eng.metrics.staleness_hist
+                               if args.staleness is not None else None),
             "config": {**meta,
                        "global_batch": args.batch * world,
                        "batch_per_gpu": args.batch,

@@ -15,19 +15,31 @@
 //  * wave = 64 lanes, hard-coded; sub-wave lane groups of GP lanes
 //    (power of two <= 64) process one sample each, reducing dots with
 //    __shfl_xor across the group.
-//  * memory-bound throughout: all row traffic is vectorized float4
-//    where dim % 4 == 0 (Guideline 13); grids are capped and
-//    grid-strided (Guideline 11).
-//  * duplicate keys inside a push batch accumulate with device-scope f32
-//    atomics (correct for '+' semantics; Adagrad instead runs on a
-//    sorted batch with one segment leader per distinct row, because its
-//    update is non-linear in the aggregated gradient).
+//  * The MF/SGNS hot path works on KEY-SORTED interleaved records
+//    rec[j] = int4{key, slot, val_bits, 0}: one 16-B load per sample
+//    replaces three 4-B strided loads (and the pack writes one 16-B
+//    store instead of three scattered 4-B stores).  Sorting is a
+//    counting sort for small key ranges (LDS histogram) or rocPRIM
+//    radix (torch::argsort) otherwise.
+//  * The per-sample coefficient (MF: e = r - u.v; SGNS: g = lr*(y -
+//    sigmoid(c.x))) is computed INSIDE the first segment walk: the
+//    keyed row is loaded once per segment into registers, so the
+//    former standalone pair kernel and its gathers disappear; the
+//    coefficient is written back into rec.val for the second pass.
+//  * duplicate keys accumulate in registers per (segment x chunk) and
+//    flush with ONE atomicAdd per column — zipfian hot keys otherwise
+//    serialize device-scope atomics (measured: first-cut one-kernel MF
+//    step ran at ~300 GB/s — profiles/mf32_r01_baseline_atomics.md).
+//    REPS replica spreading (blockIdx & 7) further splits the hottest
+//    rows across 8 buffer copies the host sums.
 //  * numerics: f32 master values end to end; the semantics oracle is
 //    ops/reference.py and tests compare against it elementwise.
 
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+
+#include <cstdlib>
 
 #define WAVE 64
 #define CHECK_DEV(x) TORCH_CHECK((x).is_cuda(), #x " must be a device tensor")
@@ -191,232 +203,291 @@ __device__ __forceinline__ float group_sum(float v) {
   return v;
 }
 
-// ---------------------------------------------------------------------
-// K4/K8 shared machinery.  The zipfian workloads make naive per-sample
-// atomics serialize on hot rows (measured: the first-cut one-kernel MF
-// step ran at ~300 GB/s, 70% of step time — profiles/
-// mf32_r01_baseline_atomics.md), so the fused steps are decomposed into
-//   (1) k_pair_coeff  — per-sample dot + scalar coefficient, no writes
-//                       beyond one float per sample;
-//   (2) k_seg_accum   — segmented scatter-accumulate over KEY-SORTED
-//                       samples: each lane group walks a contiguous
-//                       chunk, accumulates same-key rows in registers,
-//                       and issues ONE atomicAdd per (segment x chunk)
-//                       — the LDS/register pre-aggregation SURVEY.md
-//                       §7.3 calls for.
-// Sorting is rocPRIM radix sort via torch::argsort in the host wrapper.
-// ---------------------------------------------------------------------
+// =====================================================================
+// K10: sorted-record builders.
+// rec[j] = int4 { key, slot, val_bits, 0 } grouped by key ascending
+// (order within a key unspecified — consumers only need grouping).
+// =====================================================================
 
-// coeff[j] = mode 0: s_in[j] - A[arows[j]] . B[brows[j]]        (MF: e)
-//            mode 1: lr * (s_in[j] - sigmoid(A[arows[j]].B[brows[j]]))
-//                                                               (SGNS: g)
-template <int GP>
-__global__ __launch_bounds__(256) void k_pair_coeff(
-    const float* __restrict__ A, const int64_t* __restrict__ arows,
-    const float* __restrict__ B, const int64_t* __restrict__ brows,
-    const float* __restrict__ s_in, float* __restrict__ coeff,
-    int64_t n, int64_t dim, int mode, float lr) {
-  constexpr int EPL = 8;
-  const int lane = threadIdx.x % GP;
-  const int64_t groups_per_block = blockDim.x / GP;
-  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
-                     threadIdx.x / GP;
-  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
-  for (int64_t g = g0; g < n; g += gstride) {
-    const int64_t ar = arows[g] * dim, br = brows[g] * dim;
-    float dot = 0.f;
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) {
-      int c = lane + e * GP;
-      if (c < dim) dot += A[ar + c] * B[br + c];
-    }
-    dot = group_sum<GP>(dot);
-    if (lane == 0) {
-      if (mode == 0) coeff[g] = s_in[g] - dot;
-      else coeff[g] = lr * (s_in[g] - 1.0f / (1.0f + __expf(-dot)));
-    }
+// histogram of a small-range key array (counting-sort prologue)
+template <class TK>
+__global__ __launch_bounds__(256) void k_key_hist(
+    const TK* __restrict__ keys, int* __restrict__ ghist, int64_t n,
+    int m) {
+  extern __shared__ int lhist[];
+  for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
+  __syncthreads();
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x)
+    atomicAdd(&lhist[(int)keys[j]], 1);
+  __syncthreads();
+  for (int i = threadIdx.x; i < m; i += blockDim.x)
+    if (lhist[i]) atomicAdd(&ghist[i], lhist[i]);
+}
+
+// both passes' histograms in ONE read of the batch (MF: item keys for
+// pass A + user keys for pass B)
+template <class TK>
+__global__ __launch_bounds__(256) void k_dual_hist(
+    const TK* __restrict__ ka, const TK* __restrict__ kb,
+    int* __restrict__ ga, int* __restrict__ gb, int64_t n, int ma,
+    int mb) {
+  extern __shared__ int lds[];
+  int* la = lds;
+  int* lb = lds + ma;
+  for (int i = threadIdx.x; i < ma + mb; i += blockDim.x) lds[i] = 0;
+  __syncthreads();
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    atomicAdd(&la[(int)ka[j]], 1);
+    atomicAdd(&lb[(int)kb[j]], 1);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < ma; i += blockDim.x)
+    if (la[i]) atomicAdd(&ga[i], la[i]);
+  for (int i = threadIdx.x; i < mb; i += blockDim.x)
+    if (lb[i]) atomicAdd(&gb[i], lb[i]);
+}
+
+// exclusive scan of a small histogram in ONE workgroup (the rocPRIM
+// lookback scan costs 3 launches + an elementwise subtract for a
+// <=8192-element scan; this is one ~2 us launch)
+__global__ __launch_bounds__(256) void k_excl_scan_small(
+    const int* __restrict__ hist, int* __restrict__ starts, int m) {
+  __shared__ int psum[256];
+  const int T = 256, t = threadIdx.x;
+  const int per = (m + T - 1) / T;
+  const int lo = min(m, t * per), hi = min(m, lo + per);
+  int s = 0;
+  for (int i = lo; i < hi; ++i) s += hist[i];
+  psum[t] = s;
+  __syncthreads();
+  for (int off = 1; off < T; off <<= 1) {
+    int v = (t >= off) ? psum[t - off] : 0;
+    __syncthreads();
+    psum[t] += v;
+    __syncthreads();
+  }
+  int base = t ? psum[t - 1] : 0;
+  for (int i = lo; i < hi; ++i) {
+    starts[i] = base;
+    base += hist[i];
   }
 }
 
-// float4-vectorized variants (dim % 4 == 0): GP lanes of float4 cover
-// the row in 16-B coalesced accesses; 3-5 shfl reductions instead of 5-6
-// and 4x fewer index loads per element.
-template <int GP, int EPL, class TIdx>
-__global__ __launch_bounds__(256) void k_pair_coeff_v4(
-    const float4* __restrict__ A, const TIdx* __restrict__ arows,
-    const float4* __restrict__ B, const TIdx* __restrict__ brows,
-    const float* __restrict__ s_in, float* __restrict__ coeff,
-    int64_t n, int64_t dim4, int mode, float lr) {
-  const int lane = threadIdx.x % GP;
-  const int64_t groups_per_block = blockDim.x / GP;
-  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
-                     threadIdx.x / GP;
-  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
-  for (int64_t g = g0; g < n; g += gstride) {
-    const int64_t ar = (int64_t)arows[g] * dim4;
-    const int64_t br = (int64_t)brows[g] * dim4;
-    float dot = 0.f;
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) {
-      int c = lane + e * GP;
-      if (c < dim4) {
-        float4 a = A[ar + c], b = B[br + c];
-        dot += a.x * b.x + a.y * b.y + a.z * b.z + a.w * b.w;
-      }
+// counting-sort pack into interleaved recs: per chunk, an LDS histogram
+// + one global cursor reservation per touched key, then ONE int4 store
+// per sample (vs 3 scattered 4-B stores of the un-interleaved layout).
+template <class TK>
+__global__ __launch_bounds__(256) void k_pack_rec(
+    const TK* __restrict__ keys, const TK* __restrict__ slots,
+    const float* __restrict__ vals, int* __restrict__ gcursor,
+    int4* __restrict__ rec, int64_t n, int m, int chunk) {
+  extern __shared__ int lds[];
+  int* lhist = lds;
+  int* lbase = lds + m;
+  const int64_t nchunks = (n + chunk - 1) / chunk;
+  for (int64_t c = blockIdx.x; c < nchunks; c += gridDim.x) {
+    const int64_t start = c * chunk;
+    const int64_t end = min(n, start + (int64_t)chunk);
+    for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
+    __syncthreads();
+    for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x)
+      atomicAdd(&lhist[(int)keys[j]], 1);
+    __syncthreads();
+    for (int i = threadIdx.x; i < m; i += blockDim.x) {
+      int h = lhist[i];
+      lbase[i] = h ? atomicAdd(&gcursor[i], h) : 0;
     }
-    dot = group_sum<GP>(dot);
-    if (lane == 0) {
-      if (mode == 0) coeff[g] = s_in[g] - dot;
-      else coeff[g] = lr * (s_in[g] - 1.0f / (1.0f + __expf(-dot)));
+    __syncthreads();
+    for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
+    __syncthreads();
+    for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x) {
+      const int k = (int)keys[j];
+      const int lpos = atomicAdd(&lhist[k], 1);
+      rec[(int64_t)lbase[k] + lpos] =
+          make_int4(k, (int)slots[j], __float_as_int(vals[j]), 0);
     }
+    __syncthreads();
   }
 }
 
-// REPS > 1 spreads each flush over REPS replica copies of `out`
-// (replica = blockIdx % REPS; caller sums them afterwards): a zipfian
-// hot key otherwise serializes thousands of same-address atomic flushes.
-template <int GP, int EPL, bool SELF, class TIdx>
-__global__ __launch_bounds__(256) void k_seg_accum_v4(
-    float4* __restrict__ out, const TIdx* __restrict__ keys,
-    const TIdx* __restrict__ slots, const float4* __restrict__ src,
-    const float4* __restrict__ self, const float* __restrict__ coeff,
-    int64_t n, int64_t dim4, float scale, float a, int chunk,
+// re-key an existing rec array by its slot field (pass B: key<->slot
+// swap, val carried through — by now val holds the pass-A coefficient)
+__global__ __launch_bounds__(256) void k_pack_rec_swap(
+    const int4* __restrict__ src, int* __restrict__ gcursor,
+    int4* __restrict__ rec, int64_t n, int m, int chunk) {
+  extern __shared__ int lds[];
+  int* lhist = lds;
+  int* lbase = lds + m;
+  const int64_t nchunks = (n + chunk - 1) / chunk;
+  for (int64_t c = blockIdx.x; c < nchunks; c += gridDim.x) {
+    const int64_t start = c * chunk;
+    const int64_t end = min(n, start + (int64_t)chunk);
+    for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
+    __syncthreads();
+    for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x)
+      atomicAdd(&lhist[src[j].y], 1);
+    __syncthreads();
+    for (int i = threadIdx.x; i < m; i += blockDim.x) {
+      int h = lhist[i];
+      lbase[i] = h ? atomicAdd(&gcursor[i], h) : 0;
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
+    __syncthreads();
+    for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x) {
+      const int4 s = src[j];
+      const int lpos = atomicAdd(&lhist[s.y], 1);
+      rec[(int64_t)lbase[s.y] + lpos] = make_int4(s.y, s.x, s.z, 0);
+    }
+    __syncthreads();
+  }
+}
+
+// radix-sort epilogue (large key ranges): gather a rocPRIM argsort
+// permutation into interleaved recs — replaces 3 index_selects.
+template <class TK>
+__global__ __launch_bounds__(256) void k_make_rec(
+    const TK* __restrict__ keys, const TK* __restrict__ slots,
+    const float* __restrict__ vals, const int64_t* __restrict__ order,
+    int4* __restrict__ rec, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t o = order[j];
+    rec[j] = make_int4((int)keys[o], (int)slots[o],
+                       __float_as_int(vals[o]), 0);
+  }
+}
+
+__global__ __launch_bounds__(256) void k_make_rec_swap(
+    const int4* __restrict__ src, const int64_t* __restrict__ order,
+    int4* __restrict__ rec, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    const int4 s = src[order[j]];
+    rec[j] = make_int4(s.y, s.x, s.z, 0);
+  }
+}
+
+// pull the key (x) or slot (y) field out of a rec array (argsort input
+// for the large-key-range pass-B path)
+__global__ __launch_bounds__(256) void k_rec_field(
+    const int4* __restrict__ rec, int* __restrict__ out, int64_t n,
+    int field) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x)
+    out[j] = field ? rec[j].y : rec[j].x;
+}
+
+// =====================================================================
+// K4/K8: fused segmented walk over key-sorted recs.
+//
+// MODE 0 (MF pass A):   g = val - dot(keyed[key], other[slot])
+// MODE 1 (SGNS pass C): g = lrc * (val - sigmoid(dot(keyed[key],
+//                                                 other[slot])))
+// MODE 2 (second pass): g = val (coefficient precomputed by pass A/C)
+//
+// Each lane group walks one chunk of samples, caching keyed[key] in
+// registers for the current segment (so MODE 0/1 compute the per-sample
+// dot with NO extra gather), accumulating g * other[slot] in registers,
+// and flushing out[key] += scale*acc (+ a*cnt*keyed[key] when SELF —
+// the L2-regularization term) with one atomicAdd per column per
+// (segment x chunk).  MODE 0/1 write g back into rec.val for the
+// second pass.  rep_stride spreads flushes over 8 replica copies of
+// `out` (zipfian hot keys).
+// =====================================================================
+template <int GP, int EPL, int MODE, bool SELF>
+__global__ __launch_bounds__(256) void k_seg_fused(
+    float4* __restrict__ out, int4* __restrict__ rec,
+    const float4* __restrict__ keyed, const float4* __restrict__ other,
+    int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
     int64_t rep_stride) {
-  // rep_stride==0 -> single copy; reps fixed at 8 when used (& 7).
-  // self[] reads stay un-replicated; only the flush target spreads.
   const int64_t rep_off = (int64_t)(blockIdx.x & 7) * rep_stride;
   const int lane = threadIdx.x % GP;
-  const int64_t groups_per_block = blockDim.x / GP;
-  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
-                     threadIdx.x / GP;
-  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
+  const int64_t gpb = blockDim.x / GP;
+  const int64_t g0 = (int64_t)blockIdx.x * gpb + threadIdx.x / GP;
+  const int64_t gstride = (int64_t)gridDim.x * gpb;
   const int64_t nchunks = (n + chunk - 1) / chunk;
   for (int64_t cidx = g0; cidx < nchunks; cidx += gstride) {
     const int64_t start = cidx * chunk;
     const int64_t end = min(n, start + (int64_t)chunk);
-    float4 acc[EPL];
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) acc[e] = make_float4(0, 0, 0, 0);
+    float4 acc[EPL], kv[EPL];
     int cnt = 0;
-    TIdx cur = keys[start];
-    for (int64_t j = start; j < end; ++j) {
-      const TIdx k = keys[j];
-      if (k != cur) {
-        const int64_t orow = (int64_t)cur * dim4;
+    int cur = rec[start].x;
+
+    auto load_kv = [&]() {
+      if (MODE == 2 && !SELF) return;  // keyed row unused
+      const int64_t kr = (int64_t)cur * dim4;
 #pragma unroll
-        for (int e = 0; e < EPL; ++e) {
-          int c = lane + e * GP;
-          if (c < dim4) {
-            float4 t = acc[e];
-            t.x *= scale; t.y *= scale; t.z *= scale; t.w *= scale;
-            if (SELF) {
-              float4 sv = self[orow + c];
-              float ac = a * cnt;
-              t.x += ac * sv.x; t.y += ac * sv.y;
-              t.z += ac * sv.z; t.w += ac * sv.w;
-            }
-            float* o = (float*)&out[rep_off + orow + c];
-            atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
-            atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
-          }
-          acc[e] = make_float4(0, 0, 0, 0);
-        }
-        cnt = 0;
-        cur = k;
+      for (int e = 0; e < EPL; ++e) {
+        int c = lane + e * GP;
+        kv[e] = (c < dim4) ? keyed[kr + c] : make_float4(0, 0, 0, 0);
       }
-      const float gj = coeff[j];
-      const int64_t srow = (int64_t)slots[j] * dim4;
+    };
+    auto flush = [&]() {
+      const int64_t orow = rep_off + (int64_t)cur * dim4;
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
         int c = lane + e * GP;
         if (c < dim4) {
-          float4 sv = src[srow + c];
-          acc[e].x += gj * sv.x; acc[e].y += gj * sv.y;
-          acc[e].z += gj * sv.z; acc[e].w += gj * sv.w;
-        }
-      }
-      ++cnt;
-    }
-    const int64_t orow = (int64_t)cur * dim4;
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) {
-      int c = lane + e * GP;
-      if (c < dim4) {
-        float4 t = acc[e];
-        t.x *= scale; t.y *= scale; t.z *= scale; t.w *= scale;
-        if (SELF) {
-          float4 sv = self[orow + c];
-          float ac = a * cnt;
-          t.x += ac * sv.x; t.y += ac * sv.y;
-          t.z += ac * sv.z; t.w += ac * sv.w;
-        }
-        float* o = (float*)&out[rep_off + orow + c];
-        atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
-        atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
-      }
-    }
-  }
-}
-
-// out[key_j] += sum_j scale*coeff[j]*src[slots[j]]  (+ a*cnt*self[key])
-// over samples sorted by key; one atomic flush per segment per chunk.
-template <int GP, bool SELF>
-__global__ __launch_bounds__(256) void k_seg_accum(
-    float* __restrict__ out, const int64_t* __restrict__ keys,
-    const int64_t* __restrict__ slots, const float* __restrict__ src,
-    const float* __restrict__ self, const float* __restrict__ coeff,
-    int64_t n, int64_t dim, float scale, float a, int chunk) {
-  constexpr int EPL = 8;
-  const int lane = threadIdx.x % GP;
-  const int64_t groups_per_block = blockDim.x / GP;
-  const int64_t g0 = (int64_t)blockIdx.x * groups_per_block +
-                     threadIdx.x / GP;
-  const int64_t gstride = (int64_t)gridDim.x * groups_per_block;
-  const int64_t nchunks = (n + chunk - 1) / chunk;
-  for (int64_t cidx = g0; cidx < nchunks; cidx += gstride) {
-    const int64_t start = cidx * chunk;
-    const int64_t end = min(n, start + (int64_t)chunk);
-    float acc[EPL];
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) acc[e] = 0.f;
-    int cnt = 0;
-    int64_t cur = keys[start];
-    for (int64_t j = start; j < end; ++j) {
-      const int64_t k = keys[j];
-      if (k != cur) {
-        const int64_t orow = cur * dim;
-#pragma unroll
-        for (int e = 0; e < EPL; ++e) {
-          int c = lane + e * GP;
-          if (c < dim) {
-            float t = scale * acc[e];
-            if (SELF) t += a * cnt * self[orow + c];
-            atomicAdd(&out[orow + c], t);
+          float4 t = acc[e];
+          t.x *= scale; t.y *= scale; t.z *= scale; t.w *= scale;
+          if (SELF) {
+            const float ac = a * cnt;
+            t.x += ac * kv[e].x; t.y += ac * kv[e].y;
+            t.z += ac * kv[e].z; t.w += ac * kv[e].w;
           }
-          acc[e] = 0.f;
+          float* o = (float*)&out[orow + c];
+          atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
+          atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
         }
-        cnt = 0;
-        cur = k;
+        acc[e] = make_float4(0, 0, 0, 0);
       }
-      const float gj = coeff[j];
-      const int64_t srow = slots[j] * dim;
+    };
+
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) acc[e] = make_float4(0, 0, 0, 0);
+    load_kv();
+    for (int64_t j = start; j < end; ++j) {
+      const int4 rj = rec[j];
+      if (rj.x != cur) {
+        flush();
+        cur = rj.x;
+        cnt = 0;
+        load_kv();
+      }
+      const int64_t sr = (int64_t)rj.y * dim4;
+      float4 sv[EPL];
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
         int c = lane + e * GP;
-        if (c < dim) acc[e] += gj * src[srow + c];
+        sv[e] = (c < dim4) ? other[sr + c] : make_float4(0, 0, 0, 0);
+      }
+      float g;
+      if (MODE == 2) {
+        g = __int_as_float(rj.z);
+      } else {
+        float dot = 0.f;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e)
+          dot += kv[e].x * sv[e].x + kv[e].y * sv[e].y +
+                 kv[e].z * sv[e].z + kv[e].w * sv[e].w;
+        dot = group_sum<GP>(dot);
+        g = (MODE == 0)
+                ? (__int_as_float(rj.z) - dot)
+                : lrc * (__int_as_float(rj.z) -
+                         1.0f / (1.0f + __expf(-dot)));
+        if (lane == 0) rec[j].z = __float_as_int(g);
+      }
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) {
+        acc[e].x += g * sv[e].x; acc[e].y += g * sv[e].y;
+        acc[e].z += g * sv[e].z; acc[e].w += g * sv[e].w;
       }
       ++cnt;
     }
-    const int64_t orow = cur * dim;
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) {
-      int c = lane + e * GP;
-      if (c < dim) {
-        float t = scale * acc[e];
-        if (SELF) t += a * cnt * self[orow + c];
-        atomicAdd(&out[orow + c], t);
-      }
-    }
+    flush();
   }
 }
 
@@ -628,130 +699,116 @@ void hash_init_rows(torch::Tensor out, int64_t first_id, int64_t stride,
 }
 
 // ---------------------------------------------------------------------
-// K10 bucket_pack: counting-sort a batch by a small-range key (m <=
-// 16384: 2*m*4 B of dynamic LDS), fusing the payload permutation.
-// Replaces torch.argsort + 3x index_select in the MF hot path (measured:
-// rocPRIM merge sort was ~50% of step GPU time —
-// profiles/mf32_r01_baseline_atomics.md follow-up).  Order within a key
-// is unspecified (the consumers only need grouping, not stability).
+// Record-building helpers (K10).  Counting sort for key range m <=
+// PACK_MAX_M (2*m*4 B of LDS), rocPRIM radix (torch::argsort) + gather
+// otherwise.  Chunk sizes are env-tunable for on-GPU sweeps.
 // ---------------------------------------------------------------------
+// counting-sort LDS budget: 2*m*4 B (hist + base) <= 64 KB per WG
+static constexpr int PACK_MAX_M = 8192;
+
+static int env_int(const char* name, int dflt) {
+  const char* e = std::getenv(name);
+  return e ? std::atoi(e) : dflt;
+}
+
+static int pack_chunk() {
+  static int c = env_int("FPS_PACK_CHUNK", 2048);
+  return c;
+}
+
+static int seg_chunk() {
+  static int c = env_int("FPS_SEG_CHUNK", 64);
+  return c;
+}
+
+static torch::Tensor exclusive_starts(torch::Tensor hist) {
+  auto starts = torch::empty_like(hist);
+  hipLaunchKernelGGL(k_excl_scan_small, dim3(1), dim3(256), 0,
+                     cur_stream(), hist.data_ptr<int>(),
+                     starts.data_ptr<int>(), (int)hist.numel());
+  return starts;
+}
+
+// recs sorted by `keys`; hist may be precomputed (else computed here)
 template <class TK>
-__global__ __launch_bounds__(256) void k_key_hist(
-    const TK* __restrict__ keys, int* __restrict__ ghist, int64_t n,
-    int m) {
-  extern __shared__ int lhist[];
-  for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
-  __syncthreads();
-  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
-       j += (int64_t)gridDim.x * blockDim.x)
-    atomicAdd(&lhist[(int)keys[j]], 1);
-  __syncthreads();
-  for (int i = threadIdx.x; i < m; i += blockDim.x)
-    if (lhist[i]) atomicAdd(&ghist[i], lhist[i]);
-}
-
-template <class TK, class TP>
-__global__ __launch_bounds__(256) void k_bucket_pack(
-    const TK* __restrict__ keys, const TP* __restrict__ p0,
-    const float* __restrict__ p1, int* __restrict__ gcursor,
-    int* __restrict__ okeys, int* __restrict__ op0,
-    float* __restrict__ op1, int64_t n, int m, int chunk) {
-  extern __shared__ int lds[];
-  int* lhist = lds;
-  int* lbase = lds + m;
-  const int64_t nchunks = (n + chunk - 1) / chunk;
-  for (int64_t c = blockIdx.x; c < nchunks; c += gridDim.x) {
-    const int64_t start = c * chunk;
-    const int64_t end = min(n, start + (int64_t)chunk);
-    for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
-    __syncthreads();
-    for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x)
-      atomicAdd(&lhist[(int)keys[j]], 1);
-    __syncthreads();
-    for (int i = threadIdx.x; i < m; i += blockDim.x) {
-      int h = lhist[i];
-      lbase[i] = h ? atomicAdd(&gcursor[i], h) : 0;
-    }
-    __syncthreads();
-    for (int i = threadIdx.x; i < m; i += blockDim.x) lhist[i] = 0;
-    __syncthreads();
-    for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x) {
-      const int k = (int)keys[j];
-      const int lpos = atomicAdd(&lhist[k], 1);
-      const int64_t pos = (int64_t)lbase[k] + lpos;
-      okeys[pos] = k;
-      op0[pos] = (int)p0[j];
-      op1[pos] = p1[j];
-    }
-    __syncthreads();
-  }
-}
-
-
-// counting-sort pack (K10) for small key ranges; returns int32
-// (sorted_keys, sorted_p0, sorted_p1) plus the histogram start offsets.
-static constexpr int PACK_MAX_M = 16384;
-static constexpr int PACK_CHUNK = 2048;
-
-template <class TK, class TP>
-static std::vector<torch::Tensor> bucket_pack_t(torch::Tensor keys,
-                                                torch::Tensor p0,
-                                                torch::Tensor p1, int m) {
+static torch::Tensor build_rec_t(torch::Tensor keys, torch::Tensor slots,
+                                 torch::Tensor vals, int64_t m,
+                                 torch::Tensor hist) {
   int64_t n = keys.numel();
   auto opts_i32 = torch::dtype(torch::kInt32).device(keys.device());
-  auto hist = torch::zeros({m}, opts_i32);
-  size_t lds1 = (size_t)m * sizeof(int);
-  int hist_grid = (int)std::min<int64_t>((n + 4095) / 4096, 512);
-  hipLaunchKernelGGL((k_key_hist<TK>), dim3(std::max(hist_grid, 1)),
-                     dim3(256), lds1, cur_stream(), keys.data_ptr<TK>(),
-                     hist.data_ptr<int>(), n, m);
-  auto starts = torch::cumsum(hist, 0, torch::kInt32) - hist;
-  auto cursor = starts.contiguous();
-  auto okeys = torch::empty({n}, opts_i32);
-  auto op0 = torch::empty({n}, opts_i32);
-  auto op1 = torch::empty_like(p1);
-  int64_t nchunks = (n + PACK_CHUNK - 1) / PACK_CHUNK;
-  int grid = (int)std::min<int64_t>(nchunks, 2048);
-  hipLaunchKernelGGL((k_bucket_pack<TK, TP>), dim3(grid), dim3(256),
-                     2 * (size_t)m * sizeof(int), cur_stream(),
-                     keys.data_ptr<TK>(), p0.data_ptr<TP>(),
-                     p1.data_ptr<float>(), cursor.data_ptr<int>(),
-                     okeys.data_ptr<int>(), op0.data_ptr<int>(),
-                     op1.data_ptr<float>(), n, m, PACK_CHUNK);
-  return {okeys, op0, op1};
-}
-
-// sort (keys, index payload, f32 payload) by key: counting sort into
-// int32 when the key range m is small, int64 radix sort otherwise.
-static std::vector<torch::Tensor> sort_by_key(torch::Tensor keys,
-                                              torch::Tensor p0,
-                                              torch::Tensor p1,
-                                              int64_t m) {
+  auto rec = torch::empty({n, 4}, opts_i32);
   if (m > 0 && m <= PACK_MAX_M) {
-    if (keys.dtype() == torch::kInt32)
-      return bucket_pack_t<int, int>(keys, p0, p1, (int)m);
-    return bucket_pack_t<int64_t, int64_t>(keys, p0, p1, (int)m);
+    if (!hist.defined()) {
+      hist = torch::zeros({m}, opts_i32);
+      int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+      hipLaunchKernelGGL((k_key_hist<TK>), dim3(std::max(hg, 1)),
+                         dim3(256), (size_t)m * sizeof(int), cur_stream(),
+                         keys.data_ptr<TK>(), hist.data_ptr<int>(), n,
+                         (int)m);
+    }
+    auto cursor = exclusive_starts(hist).contiguous();
+    int chunk = pack_chunk();
+    int64_t nchunks = (n + chunk - 1) / chunk;
+    int grid = (int)std::min<int64_t>(nchunks, 2048);
+    hipLaunchKernelGGL((k_pack_rec<TK>), dim3(grid), dim3(256),
+                       2 * (size_t)m * sizeof(int), cur_stream(),
+                       keys.data_ptr<TK>(), slots.data_ptr<TK>(),
+                       vals.data_ptr<float>(), cursor.data_ptr<int>(),
+                       (int4*)rec.data_ptr<int>(), n, (int)m, chunk);
+  } else {
+    auto order = torch::argsort(keys);
+    hipLaunchKernelGGL((k_make_rec<TK>), dim3(grid_for(n, 256)),
+                       dim3(256), 0, cur_stream(), keys.data_ptr<TK>(),
+                       slots.data_ptr<TK>(), vals.data_ptr<float>(),
+                       order.data_ptr<int64_t>(),
+                       (int4*)rec.data_ptr<int>(), n);
   }
-  auto order = torch::argsort(keys);
-  return {keys.index_select(0, order).contiguous(),
-          p0.index_select(0, order).contiguous(),
-          p1.index_select(0, order).contiguous()};
+  return rec;
 }
 
-// host-side lane-group dispatch shared by the pair/seg kernels
-template <class F>
-static void by_group(int64_t dim, F f) {
-  if (dim <= 16) f(std::integral_constant<int, 16>{});
-  else if (dim <= 256) f(std::integral_constant<int, 32>{});
-  else { TORCH_CHECK(dim <= 512, "row dim > 512 unsupported");
-         f(std::integral_constant<int, 64>{}); }
+static torch::Tensor build_rec(torch::Tensor keys, torch::Tensor slots,
+                               torch::Tensor vals, int64_t m,
+                               torch::Tensor hist = {}) {
+  if (keys.dtype() == torch::kInt32)
+    return build_rec_t<int>(keys, slots, vals, m, hist);
+  return build_rec_t<int64_t>(keys, slots, vals, m, hist);
 }
 
-// (GP, EPL) choice: EPL*GP >= dim4 with GP as small as practical —
-// a wave then holds 64/GP independent samples, which is what hides the
-// dependent slot->row load latency of the segment walk (measured: GP=64
-// on dim=300 left one sample in flight per wave, 1.4 ms per SGNS
-// seg_accum; see profiles/).
+// recs re-keyed by the slot field of an existing rec array
+static torch::Tensor build_rec_swap(torch::Tensor src, int64_t m,
+                                    torch::Tensor hist = {}) {
+  int64_t n = src.size(0);
+  auto opts_i32 = torch::dtype(torch::kInt32).device(src.device());
+  auto rec = torch::empty({n, 4}, opts_i32);
+  if (m > 0 && m <= PACK_MAX_M) {
+    TORCH_CHECK(hist.defined(), "build_rec_swap needs a histogram");
+    auto cursor = exclusive_starts(hist).contiguous();
+    int chunk = pack_chunk();
+    int64_t nchunks = (n + chunk - 1) / chunk;
+    int grid = (int)std::min<int64_t>(nchunks, 2048);
+    hipLaunchKernelGGL(k_pack_rec_swap, dim3(grid), dim3(256),
+                       2 * (size_t)m * sizeof(int), cur_stream(),
+                       (const int4*)src.data_ptr<int>(),
+                       cursor.data_ptr<int>(), (int4*)rec.data_ptr<int>(),
+                       n, (int)m, chunk);
+  } else {
+    auto keys = torch::empty({n}, opts_i32);
+    hipLaunchKernelGGL(k_rec_field, dim3(grid_for(n, 256)), dim3(256), 0,
+                       cur_stream(), (const int4*)src.data_ptr<int>(),
+                       keys.data_ptr<int>(), n, 1);
+    auto order = torch::argsort(keys);
+    hipLaunchKernelGGL(k_make_rec_swap, dim3(grid_for(n, 256)), dim3(256),
+                       0, cur_stream(), (const int4*)src.data_ptr<int>(),
+                       order.data_ptr<int64_t>(),
+                       (int4*)rec.data_ptr<int>(), n);
+  }
+  return rec;
+}
+
+// (GP, EPL) choice: EPL*GP*4 >= dim with GP as small as practical — a
+// wave then holds 64/GP independent chunk walks, which is what hides
+// the dependent slot->row load latency (measured: GP=64 on dim=300 left
+// one walk in flight per wave — profiles/).
 template <class F>
 static void by_group4(int64_t dim4, F f) {
   using namespace std;
@@ -764,89 +821,28 @@ static void by_group4(int64_t dim4, F f) {
          f(integral_constant<int, 32>{}, integral_constant<int, 4>{}); }
 }
 
-static constexpr int SEG_CHUNK = 64;
-
-// seg_accum over key-sorted samples.  When `reps` > 1, `out` is a
-// (reps*m, dim) replica buffer the caller sums afterwards (hot-key
-// atomic spread); reps is only used on the non-SELF... both paths
-// support it via rep_stride = m*dim (SELF reads self[] un-replicated).
-static void seg_accum(torch::Tensor out, torch::Tensor keys,
-                      torch::Tensor slots, torch::Tensor src,
-                      const torch::Tensor* self, torch::Tensor coeff,
-                      double scale, double a, int64_t rep_stride = 0) {
-  int64_t n = keys.numel(), dim = src.size(1);
+// fused segment-walk launcher
+template <int MODE, bool SELF>
+static void seg_fused(torch::Tensor out, torch::Tensor rec,
+                      const torch::Tensor& keyed,
+                      const torch::Tensor& other, int64_t dim4,
+                      double scale, double a, double lrc,
+                      int64_t rep_stride = 0) {
+  int64_t n = rec.size(0);
   if (!n) return;
-  int64_t nchunks = (n + SEG_CHUNK - 1) / SEG_CHUNK;
-  TORCH_CHECK(dim % 4 == 0, "seg_accum requires dim % 4 == 0");
-  int64_t dim4 = dim / 4;
-  bool i32 = keys.dtype() == torch::kInt32;
+  int chunk = seg_chunk();
+  int64_t nchunks = (n + chunk - 1) / chunk;
   by_group4(dim4, [&](auto tag, auto etag) {
     constexpr int GP = decltype(tag)::value;
     constexpr int EPL = decltype(etag)::value;
-    dim3 grid(grid_for(nchunks * GP, 256));
-    auto launch = [&](auto self_tag, auto idx_tag) {
-      constexpr bool SELF = decltype(self_tag)::value;
-      using TIdx = typename decltype(idx_tag)::type;
-      hipLaunchKernelGGL((k_seg_accum_v4<GP, EPL, SELF, TIdx>), grid,
-                         dim3(256), 0, cur_stream(),
-                         (float4*)out.data_ptr<float>(),
-                         keys.data_ptr<TIdx>(), slots.data_ptr<TIdx>(),
-                         (const float4*)src.data_ptr<float>(),
-                         SELF ? (const float4*)self->data_ptr<float>()
-                              : nullptr,
-                         coeff.data_ptr<float>(), n, dim4, (float)scale,
-                         (float)a, SEG_CHUNK, rep_stride / 4);
-    };
-    using TI32 = std::integral_constant<int, 0>;
-    if (self) {
-      if (i32) launch(std::true_type{}, std::common_type<int>{});
-      else launch(std::true_type{}, std::common_type<int64_t>{});
-    } else {
-      if (i32) launch(std::false_type{}, std::common_type<int>{});
-      else launch(std::false_type{}, std::common_type<int64_t>{});
-    }
-  });
-}
-
-static void pair_coeff(torch::Tensor A, torch::Tensor arows,
-                       torch::Tensor B, torch::Tensor brows,
-                       torch::Tensor s_in, torch::Tensor coeff, int mode,
-                       double lr) {
-  int64_t n = coeff.numel(), dim = A.size(1);
-  if (!n) return;
-  if (dim % 4 == 0) {
-    int64_t dim4 = dim / 4;
-    bool i32 = arows.dtype() == torch::kInt32;
-    by_group4(dim4, [&](auto tag, auto etag) {
-      constexpr int GP = decltype(tag)::value;
-      constexpr int EPL = decltype(etag)::value;
-      auto launch = [&](auto idx_tag) {
-        using TIdx = typename decltype(idx_tag)::type;
-        hipLaunchKernelGGL((k_pair_coeff_v4<GP, EPL, TIdx>),
-                           dim3(grid_for(n * GP, 256)), dim3(256), 0,
-                           cur_stream(),
-                           (const float4*)A.data_ptr<float>(),
-                           arows.data_ptr<TIdx>(),
-                           (const float4*)B.data_ptr<float>(),
-                           brows.data_ptr<TIdx>(),
-                           s_in.data_ptr<float>(), coeff.data_ptr<float>(),
-                           n, dim4, mode, (float)lr);
-      };
-      if (i32) launch(std::common_type<int>{});
-      else launch(std::common_type<int64_t>{});
-    });
-    return;
-  }
-  TORCH_CHECK(arows.dtype() == torch::kInt64,
-              "scalar pair_coeff path needs int64 indices");
-  by_group(dim, [&](auto tag) {
-    constexpr int GP = decltype(tag)::value;
-    hipLaunchKernelGGL((k_pair_coeff<GP>),
-                       dim3(grid_for(n * GP, 256)), dim3(256), 0,
-                       cur_stream(), A.data_ptr<float>(),
-                       arows.data_ptr<int64_t>(), B.data_ptr<float>(),
-                       brows.data_ptr<int64_t>(), s_in.data_ptr<float>(),
-                       coeff.data_ptr<float>(), n, dim, mode, (float)lr);
+    hipLaunchKernelGGL((k_seg_fused<GP, EPL, MODE, SELF>),
+                       dim3(grid_for(nchunks * GP, 256)), dim3(256), 0,
+                       cur_stream(), (float4*)out.data_ptr<float>(),
+                       (int4*)rec.data_ptr<int>(),
+                       (const float4*)keyed.data_ptr<float>(),
+                       (const float4*)other.data_ptr<float>(), n, dim4,
+                       (float)scale, (float)a, (float)lrc, chunk,
+                       rep_stride / 4);
   });
 }
 
@@ -858,10 +854,13 @@ static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
   return r;
 }
 
-// K4: e = r - u0.v computed on the item-sorted batch (exact batch-
-// snapshot semantics: every rating sees pre-update u0 and the pulled v);
-// item deltas = lr*(e*u0 - reg*v) segment-accumulated by item; user
-// updates = lr*(e*v - reg*u0) segment-accumulated by user.
+// K4: one fused SGD pass over a batch of ratings (semantics =
+// ops/reference.py::mf_sgd_step: every rating sees the pre-update user
+// row u0 and the pulled item snapshot v):
+//   pass A (recs by item): e = r - u0.v cached-keyed dot; item deltas
+//     lr*(e*u0 - reg*v) segment-accumulated; e persisted into rec.val
+//   pass B (recs by user): user updates lr*(e*v - reg*u0) accumulated
+//     into user_table in place
 torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                           torch::Tensor item_vecs, torch::Tensor item_slots,
                           torch::Tensor ratings, double lr, double reg) {
@@ -869,59 +868,114 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   CHECK_DEV(item_vecs); CHECK_CONT(item_vecs);
   CHECK_CONT(user_rows); CHECK_CONT(item_slots); CHECK_CONT(ratings);
   int64_t n = ratings.numel(), dim = user_table.size(1);
-  int64_t m = item_vecs.size(0);
+  int64_t m = item_vecs.size(0), U = user_table.size(0);
   TORCH_CHECK(item_vecs.size(1) == dim);
+  TORCH_CHECK(dim % 4 == 0, "mf_sgd_step requires dim % 4 == 0");
   if (!n) return torch::zeros_like(item_vecs);
-  // pass A: counting-sort by item (K10), compute e, accumulate deltas
-  auto packA = sort_by_key(item_slots, user_rows, ratings, m);
-  auto& s_slots = packA[0];
-  auto& s_urows = packA[1];
-  auto& s_r = packA[2];
-  auto e = torch::empty_like(s_r);
-  pair_coeff(user_table, s_urows, item_vecs, s_slots, s_r, e, 0, 0.0);
-  int64_t reps = (dim % 4 == 0) ? delta_reps(n, m, dim) : 1;
+  int64_t dim4 = dim / 4;
+
+  // histograms: one fused read when both key ranges are LDS-sized
+  torch::Tensor histA, histB;
+  bool smallA = m <= PACK_MAX_M, smallB = U <= PACK_MAX_M;
+  if (smallA && smallB && m + U <= 2 * PACK_MAX_M &&
+      item_slots.dtype() == user_rows.dtype()) {
+    auto opts_i32 = torch::dtype(torch::kInt32).device(ratings.device());
+    histA = torch::zeros({m}, opts_i32);
+    histB = torch::zeros({U}, opts_i32);
+    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+    size_t lds = (size_t)(m + U) * sizeof(int);
+    if (item_slots.dtype() == torch::kInt32)
+      hipLaunchKernelGGL((k_dual_hist<int>), dim3(std::max(hg, 1)),
+                         dim3(256), lds, cur_stream(),
+                         item_slots.data_ptr<int>(),
+                         user_rows.data_ptr<int>(), histA.data_ptr<int>(),
+                         histB.data_ptr<int>(), n, (int)m, (int)U);
+    else
+      hipLaunchKernelGGL((k_dual_hist<int64_t>), dim3(std::max(hg, 1)),
+                         dim3(256), lds, cur_stream(),
+                         item_slots.data_ptr<int64_t>(),
+                         user_rows.data_ptr<int64_t>(),
+                         histA.data_ptr<int>(), histB.data_ptr<int>(), n,
+                         (int)m, (int)U);
+  }
+
+  // pass A: recs by item slot; fused e + item-delta accumulation
+  auto recA = build_rec(item_slots, user_rows, ratings, m, histA);
+  int64_t reps = delta_reps(n, m, dim);
   torch::Tensor deltas;
   if (reps > 1) {
     auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
-    seg_accum(rep_buf, s_slots, s_urows, user_table, &item_vecs, e, lr,
-              -lr * reg, m * dim);
+    seg_fused<0, true>(rep_buf, recA, item_vecs, user_table, dim4, lr,
+                       -lr * reg, 0.0, m * dim);
     deltas = rep_buf.view({reps, m, dim}).sum(0);
   } else {
     deltas = torch::zeros_like(item_vecs);
-    seg_accum(deltas, s_slots, s_urows, user_table, &item_vecs, e, lr,
-              -lr * reg);
+    seg_fused<0, true>(deltas, recA, item_vecs, user_table, dim4, lr,
+                       -lr * reg, 0.0);
   }
-  // pass B: counting-sort by user, apply user updates in place
-  auto packB = sort_by_key(s_urows, s_slots, e, user_table.size(0));
-  seg_accum(user_table, packB[0], packB[1], item_vecs, &user_table,
-            packB[2], lr, -lr * reg);
+
+  // pass B: recs by user row; apply user updates in place
+  if (!smallB) histB = torch::Tensor();
+  else if (!histB.defined()) {
+    auto opts_i32 = torch::dtype(torch::kInt32).device(ratings.device());
+    histB = torch::zeros({U}, opts_i32);
+    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+    if (user_rows.dtype() == torch::kInt32)
+      hipLaunchKernelGGL((k_key_hist<int>), dim3(std::max(hg, 1)),
+                         dim3(256), (size_t)U * sizeof(int), cur_stream(),
+                         user_rows.data_ptr<int>(), histB.data_ptr<int>(),
+                         n, (int)U);
+    else
+      hipLaunchKernelGGL((k_key_hist<int64_t>), dim3(std::max(hg, 1)),
+                         dim3(256), (size_t)U * sizeof(int), cur_stream(),
+                         user_rows.data_ptr<int64_t>(),
+                         histB.data_ptr<int>(), n, (int)U);
+  }
+  auto recB = build_rec_swap(recA, U, histB);
+  seg_fused<2, true>(user_table, recB, user_table, item_vecs, dim4, lr,
+                     -lr * reg, 0.0);
   return deltas;
 }
 
-// K8: g = lr*(label - sigmoid(v.u)); center and context deltas each
-// segment-accumulated on their own sorted order.
-std::vector<torch::Tensor> sgns_step(
-    torch::Tensor center_vecs, torch::Tensor center_slots,
-    torch::Tensor ctx_vecs, torch::Tensor ctx_slots, torch::Tensor labels,
-    double lr) {
-  CHECK_DEV(center_vecs); CHECK_CONT(center_vecs);
-  CHECK_DEV(ctx_vecs); CHECK_CONT(ctx_vecs);
-  int64_t n = labels.numel();
-  auto cd = torch::zeros_like(center_vecs);
-  auto od = torch::zeros_like(ctx_vecs);
-  if (!n) return {cd, od};
-  auto g = torch::empty_like(labels);
-  pair_coeff(center_vecs, center_slots, ctx_vecs, ctx_slots, labels, g, 1,
-             lr);
-  auto packC = sort_by_key(center_slots, ctx_slots, g,
-                           center_vecs.size(0));
-  seg_accum(cd, packC[0], packC[1], ctx_vecs, nullptr, packC[2], 1.0, 0.0);
-  auto packO = sort_by_key(ctx_slots, center_slots, g, ctx_vecs.size(0));
-  seg_accum(od, packO[0], packO[1], center_vecs, nullptr, packO[2], 1.0,
-            0.0);
-  return {cd, od};
+// K8: fused SGNS step over ONE shared row buffer (center and context
+// slots index disjoint rows of `vecs` — the model maps context ids to
+// [V, 2V)).  Returns the combined deltas buffer.
+//   pass C (recs by center): g = lr*(y - sigmoid(c.x)) cached-keyed
+//     dot; center deltas g*x accumulated; g persisted
+//   pass D (recs by context): context deltas g*c accumulated into the
+//     same buffer
+torch::Tensor sgns_step(torch::Tensor vecs, torch::Tensor center_slots,
+                        torch::Tensor ctx_slots, torch::Tensor labels,
+                        double lr) {
+  CHECK_DEV(vecs); CHECK_CONT(vecs);
+  CHECK_CONT(center_slots); CHECK_CONT(ctx_slots); CHECK_CONT(labels);
+  int64_t n = labels.numel(), dim = vecs.size(1), m = vecs.size(0);
+  TORCH_CHECK(dim % 4 == 0, "sgns_step requires dim % 4 == 0");
+  auto deltas = torch::zeros_like(vecs);
+  if (!n) return deltas;
+  int64_t dim4 = dim / 4;
+  auto recC = build_rec(center_slots, ctx_slots, labels, m);
+  seg_fused<1, false>(deltas, recC, vecs, vecs, dim4, 1.0, 0.0, lr);
+  torch::Tensor histD;
+  if (m <= PACK_MAX_M) {
+    auto opts_i32 = torch::dtype(torch::kInt32).device(vecs.device());
+    histD = torch::zeros({m}, opts_i32);
+    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+    if (ctx_slots.dtype() == torch::kInt32)
+      hipLaunchKernelGGL((k_key_hist<int>), dim3(std::max(hg, 1)),
+                         dim3(256), (size_t)m * sizeof(int), cur_stream(),
+                         ctx_slots.data_ptr<int>(), histD.data_ptr<int>(),
+                         n, (int)m);
+    else
+      hipLaunchKernelGGL((k_key_hist<int64_t>), dim3(std::max(hg, 1)),
+                         dim3(256), (size_t)m * sizeof(int), cur_stream(),
+                         ctx_slots.data_ptr<int64_t>(),
+                         histD.data_ptr<int>(), n, (int)m);
+  }
+  auto recD = build_rec_swap(recC, m, histD);
+  seg_fused<2, false>(deltas, recD, vecs, vecs, dim4, 1.0, 0.0, 0.0);
+  return deltas;
 }
-
 
 static int pa_mode(const std::string& variant) {
   if (variant == "PA") return 0;

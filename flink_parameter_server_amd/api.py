@@ -160,6 +160,10 @@ class SimpleServerLogic(ServerLogic):
     :class:`~flink_parameter_server_amd.table.TableServerLogic`.
     """
 
+    #: answers every pull batch in full, in received order, same round
+    #: (lets the engine elide answer ids from the feedback edge).
+    answers_in_order = True
+
     def __init__(self, param_init: Callable[[int], torch.Tensor],
                  param_update: Callable[[torch.Tensor, torch.Tensor], torch.Tensor]):
         self.param_init = param_init

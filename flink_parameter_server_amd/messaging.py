@@ -139,7 +139,7 @@ class SerializingTransport(Transport):
         self.rank = inner.rank
         self.world = inner.world
 
-    def alltoallv(self, parts, specs):
+    def alltoallv(self, parts, specs, extra=None, recv_counts=None):
         C = len(parts)
         if C == 3:  # worker -> server leg
             envelopes = [encode_worker_to_ps(self.rank, parts[0][d],
@@ -149,15 +149,21 @@ class SerializingTransport(Transport):
             for d, env in enumerate(envelopes):
                 wire[0][d], wire[1][d], wire[2][d] = \
                     decode_worker_to_ps(env)
-            return self.inner.alltoallv(wire, specs)
-        if C == 2:  # feedback edge (answers)
+            return self.inner.alltoallv(wire, specs, extra=extra,
+                                        recv_counts=recv_counts)
+        if C == 2:  # feedback edge (answers with ids)
             envelopes = [encode_ps_to_worker(d, parts[0][d], parts[1][d])
                          for d in range(self.world)]
             wire = [[None] * self.world for _ in range(2)]
             for d, env in enumerate(envelopes):
                 wire[0][d], wire[1][d] = decode_ps_to_worker(env)
-            return self.inner.alltoallv(wire, specs)
-        return self.inner.alltoallv(parts, specs)
+            return self.inner.alltoallv(wire, specs, extra=extra,
+                                        recv_counts=recv_counts)
+        # C == 1: in-order feedback edge — answer ids are intentionally
+        # elided from the wire (PullAnswer.param_ids reconstructed by
+        # the puller from its own sent segments); nothing to round-trip.
+        return self.inner.alltoallv(parts, specs, extra=extra,
+                                    recv_counts=recv_counts)
 
     def allgather_int(self, vec):
         return self.inner.allgather_int(vec)

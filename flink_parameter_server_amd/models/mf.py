@@ -102,22 +102,40 @@ class OnlineMFWorkerLogic(WorkerLogic):
             ops.hash_init_rows(self.user_table, 0, 1, range_min, range_max,
                                seed=seed + 7777)
         self._pending: List[Tuple[MFBatch, PullBuffer, Optional[torch.Tensor]]] = []
-        # dense-flag coalescing: O(n) scatter + cumsum instead of a sort-
-        # based torch.unique, and rating->slot lookup for free.  Used when
-        # the item-id space is known and small enough for a dense flag
-        # array (measured: rocPRIM sort-unique was ~7% of a 1M-rating
-        # round — profiles/mf32_r01_baseline_atomics.md).
+        # Pull policy for a known item space, fastest first:
+        # * dense pull (n_items <= DENSE_PULL_MAX): pull the WHOLE item
+        #   range every round — at MovieLens-like item counts a 1M-rating
+        #   batch touches essentially every item anyway, the answer is a
+        #   sub-MB payload, slots are the raw item ids, and the whole
+        #   coalescing pass (flag scatter + cumsum + nonzero + lookup,
+        #   ~15% of a round's kernel time) disappears.
+        # * dense-flag coalescing: O(n) scatter + cumsum instead of a
+        #   sort-based torch.unique, rating->slot lookup for free.
+        # * torch.unique fallback for unbounded id spaces.
         self._coalescer: Optional[DenseCoalescer] = None
-        if 0 < self.n_items <= DenseCoalescer.MAX_SPACE:
+        self._all_items: Optional[torch.Tensor] = None
+        self._dense_pull = 0 < self.n_items <= self.DENSE_PULL_MAX
+        if (not self._dense_pull and
+                0 < self.n_items <= DenseCoalescer.MAX_SPACE):
             self._coalescer = DenseCoalescer(self.n_items, self.device)
 
     def batch_size(self, batch: MFBatch) -> int:
         return len(batch)
 
+    #: item spaces up to this size are pulled whole each round
+    DENSE_PULL_MAX = 65536
+
     def on_recv(self, data: MFBatch, client: ParameterServerClient) -> None:
         if self.neg_rate > 0:
             data = self._with_negatives(data)
-        if self._coalescer is not None:
+        if self._dense_pull:
+            if self._all_items is None:
+                self._all_items = torch.arange(
+                    self.n_items, dtype=torch.int64,
+                    device=data.item_ids.device)
+            uids, slots = self._all_items, data.item_ids
+            buf = PullBuffer(uids, self.n_factors, assume_sorted=True)
+        elif self._coalescer is not None:
             uids, slots = self._coalescer(data.item_ids)
             buf = PullBuffer(uids, self.n_factors, assume_sorted=True)
         else:

@@ -113,7 +113,7 @@ class SGNSWorkerLogic(WorkerLogic):
             c_slots = buf.slots_of(centers)
             o_slots = buf.slots_of(ctx_ids)
         # One buffer holds both center and context rows; slot spaces are
-        # disjoint ids so the deltas can share one accumulator.
-        c_deltas, o_deltas = ops.sgns_step(buf.values, c_slots, buf.values,
-                                           o_slots, labels, self.lr)
-        client.push(buf.uids, c_deltas + o_deltas)
+        # disjoint ids so the deltas share one accumulator.
+        deltas = ops.sgns_step(buf.values, c_slots, o_slots, labels,
+                               self.lr)
+        client.push(buf.uids, deltas)

@@ -115,12 +115,12 @@ def lr_step(indptr: torch.Tensor, feat_slots: torch.Tensor,
     return reference.lr_step(indptr, feat_slots, x_vals, labels, weights)
 
 
-def sgns_step(center_vecs: torch.Tensor, center_slots: torch.Tensor,
-              ctx_vecs: torch.Tensor, ctx_slots: torch.Tensor,
-              labels: torch.Tensor, lr: float
-              ) -> Tuple[torch.Tensor, torch.Tensor]:
-    if _cuda(center_vecs):
-        return hip.ext().sgns_step(center_vecs, center_slots, ctx_vecs,
-                                   ctx_slots, labels, lr)
-    return reference.sgns_step(center_vecs, center_slots, ctx_vecs,
-                               ctx_slots, labels, lr)
+def sgns_step(vecs: torch.Tensor, center_slots: torch.Tensor,
+              ctx_slots: torch.Tensor, labels: torch.Tensor, lr: float
+              ) -> torch.Tensor:
+    """Fused SGNS step over one shared row buffer (center and context
+    slots index disjoint rows); returns the combined deltas buffer."""
+    if _cuda(vecs):
+        return hip.ext().sgns_step(vecs, center_slots, ctx_slots, labels,
+                                   lr)
+    return reference.sgns_step(vecs, center_slots, ctx_slots, labels, lr)

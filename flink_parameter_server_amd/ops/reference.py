@@ -289,19 +289,19 @@ def lr_step(indptr: torch.Tensor, feat_slots: torch.Tensor,
 # All dots read the pulled snapshot (round semantics).
 # ---------------------------------------------------------------------------
 
-def sgns_step(center_vecs: torch.Tensor, center_slots: torch.Tensor,
-              ctx_vecs: torch.Tensor, ctx_slots: torch.Tensor,
-              labels: torch.Tensor, lr: float
-              ) -> Tuple[torch.Tensor, torch.Tensor]:
+def sgns_step(vecs: torch.Tensor, center_slots: torch.Tensor,
+              ctx_slots: torch.Tensor, labels: torch.Tensor, lr: float
+              ) -> torch.Tensor:
     """center_slots/ctx_slots: (n,) pair lists (one row per (c, o) pair,
-    positives and negatives flattened).  Returns (center_deltas aligned with
-    center_vecs, ctx_deltas aligned with ctx_vecs)."""
-    v = center_vecs.index_select(0, center_slots)
-    u = ctx_vecs.index_select(0, ctx_slots)
+    positives and negatives flattened) indexing DISJOINT rows of the one
+    shared buffer ``vecs`` (the model maps context ids to [V, 2V), so
+    center and context rows never collide).  Returns the combined deltas
+    buffer aligned with ``vecs``."""
+    v = vecs.index_select(0, center_slots)
+    u = vecs.index_select(0, ctx_slots)
     s = torch.sigmoid((v * u).sum(dim=1))
     g = (lr * (labels - s)).unsqueeze(1)
-    center_deltas = torch.zeros_like(center_vecs)
-    center_deltas.index_add_(0, center_slots, g * u)
-    ctx_deltas = torch.zeros_like(ctx_vecs)
-    ctx_deltas.index_add_(0, ctx_slots, g * v)
-    return center_deltas, ctx_deltas
+    deltas = torch.zeros_like(vecs)
+    deltas.index_add_(0, center_slots, g * u)
+    deltas.index_add_(0, ctx_slots, g * v)
+    return deltas

@@ -142,6 +142,13 @@ class PSEngine:
         limit = worker_logic.pull_limit or config.pull_limit
         self._pull_limit = int(limit) if limit else None
 
+        # A server logic that answers every pull batch in full, in
+        # received order, within the same round lets the engine drop
+        # answer ids (and their counts exchange) from the feedback edge:
+        # the puller reconstructs them from its own sent segments.
+        self._answers_in_order = bool(getattr(server_logic,
+                                              "answers_in_order", False))
+
     # ------------------------------------------------------------------
     def run(self, data: Iterable[Any], max_rounds: Optional[int] = None
             ) -> Tuple[List[Any], List[Any]]:
@@ -207,17 +214,28 @@ class PSEngine:
             pull_ids = self._take_pulls()
             push_ids, push_deltas = self._take_pushes()
         send = self._bucket(pull_ids, push_ids, push_deltas)
+        sent_pulls = send[0]
         m.pull_ids_sent += int(pull_ids.numel()) if pull_ids is not None else 0
         m.push_ids_sent += int(push_ids.numel()) if push_ids is not None else 0
 
-        # 4. exchange worker->server segments
+        # 4. exchange worker->server segments.  The SSP clock + drain
+        # control rides the counts exchange (one collective, not two);
+        # its values are fully determined before the exchange, so this
+        # is the same information the reference's quiescence detection
+        # needs, one leg earlier.
+        worker_done = int(exhausted and self._outstanding == 0 and
+                          not self._pending_answers and
+                          self._queued_pulls() == 0 and
+                          not self._pull_buf and not self._push_buf and
+                          push_ids is None)
         dim = self.cfg.dim
         wire = torch.bfloat16 if self.cfg.wire_dtype == "bf16" \
             else torch.float32
         self._wire_out(send[2], wire)
-        recv = self.transport.alltoallv(send, [
-            (torch.int64, ()), (torch.int64, ()),
-            (wire, (dim,))])
+        m.bytes_sent += self._xgmi_bytes(send)
+        recv, ctrl = self.transport.alltoallv(
+            send, [(torch.int64, ()), (torch.int64, ()), (wire, (dim,))],
+            extra=[self._clock, worker_done, int(exhausted)])
         recv = self._to_device(recv)
         self._unwire(recv[2])
         r_pull, r_push_ids, r_push_deltas = recv
@@ -232,33 +250,48 @@ class PSEngine:
             if ids is not None and ids.numel():
                 self.server_logic.on_pull_recv(ids, s, self.ps)
 
-        # 6. exchange answers back (the feedback edge)
-        ans_ids = [_concat([a for a, _ in self._answer_buf[w]])
-                   for w in range(self.world)]
+        # 6. exchange answers back (the feedback edge).  When every
+        # server answers in order (our table logics do), the answer IDS
+        # never travel: the puller already knows them — its own pull
+        # segments — and their counts, so the counts exchange is skipped
+        # too and only the value payloads cross xGMI.
         ans_vals = [_concat([v for _, v in self._answer_buf[w]])
                     for w in range(self.world)]
-        self._answer_buf = [[] for _ in range(self.world)]
-        self._wire_out(ans_vals, wire)
-        back = self._to_device(self.transport.alltoallv(
-            [ans_ids, ans_vals],
-            [(torch.int64, ()), (wire, (dim,))]))
-        self._unwire(back[1])
-        for s in range(self.world):
-            if back[0][s] is not None and back[0][s].numel():
-                self._pending_answers.append((back[0][s], back[1][s]))
+        if self._answers_in_order:
+            self._answer_buf = [[] for _ in range(self.world)]
+            self._wire_out(ans_vals, wire)
+            m.bytes_sent += self._xgmi_bytes([ans_vals])
+            counts = [0 if t is None else int(t.numel())
+                      for t in sent_pulls]
+            back_vals = self._to_device(self.transport.alltoallv(
+                [ans_vals], [(wire, (dim,))], recv_counts=[counts]))[0]
+            self._unwire(back_vals)
+            for s in range(self.world):
+                if back_vals[s] is not None and back_vals[s].numel():
+                    self._pending_answers.append(
+                        (sent_pulls[s].to(back_vals[s].device),
+                         back_vals[s]))
+        else:
+            ans_ids = [_concat([a for a, _ in self._answer_buf[w]])
+                       for w in range(self.world)]
+            self._answer_buf = [[] for _ in range(self.world)]
+            self._wire_out(ans_vals, wire)
+            m.bytes_sent += self._xgmi_bytes([ans_ids, ans_vals])
+            back = self._to_device(self.transport.alltoallv(
+                [ans_ids, ans_vals],
+                [(torch.int64, ()), (wire, (dim,))]))
+            self._unwire(back[1])
+            for s in range(self.world):
+                if back[0][s] is not None and back[0][s].numel():
+                    self._pending_answers.append((back[0][s], back[1][s]))
 
         if self.checkpointer is not None:
             self.checkpointer.maybe(m.rounds)
 
-        # 7. control: [clock, worker_done, exhausted].  An exhausted
-        # worker's clock freezes, so it must not hold the SSP min-clock
-        # back (it would deadlock the still-running workers).
-        worker_done = int(exhausted and self._outstanding == 0 and
-                          not self._pending_answers and
-                          self._queued_pulls() == 0 and
-                          not self._pull_buf and not self._push_buf)
-        ctrl = self.transport.allgather_int(
-            [self._clock, worker_done, int(exhausted)])
+        # 7. act on the control matrix gathered in step 4: [clock,
+        # worker_done, exhausted].  An exhausted worker's clock freezes,
+        # so it must not hold the SSP min-clock back (it would deadlock
+        # the still-running workers).
         active = ctrl[:, 2] == 0
         self._min_clock = int(ctrl[active, 0].min()) if bool(active.any()) \
             else int(ctrl[:, 0].max())
@@ -287,6 +320,16 @@ class PSEngine:
         if self._pull_queue is not None:
             n += int(self._pull_queue.numel())
         return n
+
+    def _xgmi_bytes(self, channels) -> int:
+        """Bytes this rank puts on the wire this round (self-addressed
+        segments short-circuit in every transport and are not counted)."""
+        total = 0
+        for chan in channels:
+            for d, t in enumerate(chan):
+                if d != self.rank and t is not None:
+                    total += t.numel() * t.element_size()
+        return total
 
     @staticmethod
     def _wire_out(chan: List[Optional[torch.Tensor]],

@@ -34,13 +34,22 @@ class Transport:
     world: int = 1
 
     def alltoallv(self, parts: Sequence[Sequence[Optional[torch.Tensor]]],
-                  specs: Sequence[Tuple[torch.dtype, Tuple[int, ...]]]
-                  ) -> List[List[Optional[torch.Tensor]]]:
+                  specs: Sequence[Tuple[torch.dtype, Tuple[int, ...]]],
+                  extra: Optional[Sequence[int]] = None,
+                  recv_counts: Optional[Sequence[Sequence[int]]] = None):
         """parts[c][d] = tensor for channel c to rank d (dim 0 = items;
         may be None = empty).  specs[c] = (dtype, trailing_shape) of
         channel c — every rank must pass the same specs (the receiver
         sizes its buffers from them, never from its own outgoing data).
         Returns recv[c][s] = tensor received on channel c from rank s.
+
+        ``extra``: small host int vector folded into the counts exchange
+        (one collective instead of two — the engine's SSP clock +
+        termination control rides here); when given the return value is
+        ``(recv, extras)`` with extras int64 (world, len(extra)).
+        ``recv_counts[c][s]``: receive sizes already known to the caller
+        (e.g. in-order pull answers: the receiver sized them when it sent
+        the pulls) — the counts exchange is skipped entirely.
         """
         raise NotImplementedError
 
@@ -56,8 +65,11 @@ class LocalTransport(Transport):
     def __init__(self) -> None:
         self.rank, self.world = 0, 1
 
-    def alltoallv(self, parts, specs):
-        return [[p[0]] for p in parts]
+    def alltoallv(self, parts, specs, extra=None, recv_counts=None):
+        recv = [[p[0]] for p in parts]
+        if extra is None:
+            return recv
+        return recv, torch.tensor([list(extra)], dtype=torch.int64)
 
     def allgather_int(self, vec):
         return torch.tensor([list(vec)], dtype=torch.int64)
@@ -77,20 +89,34 @@ class TorchDistTransport(Transport):
             torch.device("cuda", torch.cuda.current_device())
             if dist.get_backend() == "nccl" else torch.device("cpu"))
 
-    def alltoallv(self, parts, specs):
+    def alltoallv(self, parts, specs, extra=None, recv_counts=None):
         dist = self.dist
         C, W, me = len(parts), self.world, self.rank
         dev = self.device
 
-        # 1) counts: one all_gather of int64[C*W] covers every channel.
-        send_counts = torch.zeros(C * W, dtype=torch.int64)
-        for c, chan in enumerate(parts):
-            for d, t in enumerate(chan):
-                send_counts[c * W + d] = 0 if t is None else t.shape[0]
-        gdev = send_counts.to(dev)
-        gathered = [torch.empty_like(gdev) for _ in range(W)]
-        dist.all_gather(gathered, gdev)
-        counts = torch.stack([g.cpu() for g in gathered])  # (W, C*W)
+        # 1) counts: one all_gather of int64[C*W (+extra)] covers every
+        # channel — skipped entirely when the caller knows its receive
+        # sizes (in-order answers) and has no control payload to fold in.
+        counts = extras = None
+        if recv_counts is None or extra is not None:
+            ne = len(extra) if extra is not None else 0
+            send_counts = torch.zeros(C * W + ne, dtype=torch.int64)
+            for c, chan in enumerate(parts):
+                for d, t in enumerate(chan):
+                    send_counts[c * W + d] = 0 if t is None else t.shape[0]
+            if ne:
+                send_counts[C * W:] = torch.tensor(list(extra),
+                                                   dtype=torch.int64)
+            gdev = send_counts.to(dev)
+            gathered = [torch.empty_like(gdev) for _ in range(W)]
+            dist.all_gather(gathered, gdev)
+            full = torch.stack([g.cpu() for g in gathered])  # (W, C*W+ne)
+            counts, extras = full[:, :C * W], full[:, C * W:]
+
+        def n_from(s, c):
+            if recv_counts is not None:
+                return int(recv_counts[c][s])
+            return int(counts[s, c * W + me])
 
         # 2) payloads: grouped isend/irecv (RCCL: grouped ncclSend/Recv).
         recv: List[List[Optional[torch.Tensor]]] = [
@@ -99,7 +125,7 @@ class TorchDistTransport(Transport):
         for c, chan in enumerate(parts):
             dtype, trailing = specs[c]
             for s in range(W):
-                n = int(counts[s, c * W + me])
+                n = n_from(s, c)
                 if s == me:
                     t = chan[me]
                     recv[c][me] = t if t is not None else None
@@ -118,7 +144,9 @@ class TorchDistTransport(Transport):
         if p2p:
             for work in dist.batch_isend_irecv(p2p):
                 work.wait()
-        return recv
+        if extra is None:
+            return recv
+        return recv, extras
 
     def allgather_int(self, vec):
         t = torch.tensor(list(vec), dtype=torch.int64, device=self.device)
@@ -181,9 +209,12 @@ class LoopbackTransport(Transport):
         self.rank = rank
         self.world = shared.world
 
-    def alltoallv(self, parts, specs):
+    def alltoallv(self, parts, specs, extra=None, recv_counts=None):
         sh = self.shared
         sh.mailbox[self.rank] = [list(chan) for chan in parts]
+        if extra is not None:
+            sh.intbox[self.rank] = torch.tensor(list(extra),
+                                                dtype=torch.int64)
         sh.barrier.wait()
         C = len(parts)
         recv: List[List[Optional[torch.Tensor]]] = [
@@ -192,8 +223,11 @@ class LoopbackTransport(Transport):
             box = sh.mailbox[s]
             for c in range(C):
                 recv[c][s] = box[c][self.rank]
+        extras = torch.stack(list(sh.intbox)) if extra is not None else None
         sh.barrier.wait()  # everyone has read before boxes are overwritten
-        return recv
+        if extra is None:
+            return recv
+        return recv, extras
 
     def allgather_int(self, vec):
         sh = self.shared

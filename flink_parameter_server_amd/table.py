@@ -108,6 +108,10 @@ class TableServerLogic:
     (or the Adagrad rule).
     """
 
+    #: answers every pull batch in full, in received order, same round —
+    #: lets the engine elide answer ids from the feedback edge.
+    answers_in_order = True
+
     def __init__(self, table: ShardedTable, dump_on_close: bool = True,
                  pushes_unique: bool = False):
         """``pushes_unique=True``: the application guarantees every push

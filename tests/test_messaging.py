@@ -149,6 +149,56 @@ def test_bf16_wire_dtype():
     assert counts[0] == 2.0 and counts[1] == 2.0
 
 
+def test_out_of_order_server_uses_id_carrying_answers():
+    """A custom ServerLogic without answers_in_order may answer in any
+    order (here: reversed, split in two calls) — the engine must fall
+    back to the id-carrying feedback edge and still route correctly."""
+    from flink_parameter_server_amd import ServerLogic
+
+    class ReversedServer(ServerLogic):
+        def __init__(self):
+            self.table = {}
+
+        def on_pull_recv(self, ids, worker, ps):
+            rev = torch.flip(ids, [0])
+            vals = torch.tensor(
+                [[self.table.get(int(i), 0.0)] for i in rev])
+            half = rev.numel() // 2
+            ps.answer_pull(rev[:half], vals[:half], worker)
+            ps.answer_pull(rev[half:], vals[half:], worker)
+
+        def on_push_recv(self, ids, deltas, ps):
+            for k, i in enumerate(ids.tolist()):
+                self.table[i] = self.table.get(i, 0.0) + float(deltas[k])
+
+        def close(self, ps):
+            for i, v in sorted(self.table.items()):
+                ps.output((i, v))
+
+    class EchoLogic(WorkerLogic):
+        """Pushes +1 per answered id (mis-routed answers show up as
+        wrong final counts)."""
+
+        def on_recv(self, data, client):
+            client.pull(data)
+
+        def on_pull_recv(self, ids, values, client):
+            client.push(ids, torch.ones(ids.numel(), 1))
+
+    def job(rank, transport):
+        data = [torch.tensor([0, 1, 2, 3]), torch.tensor([1, 2])]
+        eng = PSEngine(PSConfig(num_ids=4, dim=1), EchoLogic(),
+                       ReversedServer(), transport)
+        assert not eng._answers_in_order
+        return eng.run(iter(data))
+
+    results = run_loopback(2, job)
+    counts = {}
+    for _, souts in results:
+        counts.update({i: v for i, v in souts})
+    assert counts == {0: 2.0, 1: 4.0, 2: 4.0, 3: 2.0}
+
+
 def test_multiworker_singleshard_bucket_padding():
     """ps_parallelism < worker_parallelism: every id routes to shard 0;
     rank 1 hosts an empty shard (engine pads its bucket counts)."""

@@ -162,16 +162,15 @@ def test_lr_step_gradient_sign():
 
 def test_sgns_step_moves_pair_together():
     torch.manual_seed(4)
-    v = torch.randn(2, 8)   # center rows
-    u = torch.randn(3, 8)   # context rows
+    vecs = torch.randn(5, 8)  # rows 0-1 center, rows 2-4 context
     c_slots = torch.tensor([0, 0])
-    o_slots = torch.tensor([1, 2])
+    o_slots = torch.tensor([3, 4])
     labels = torch.tensor([1.0, 0.0])
-    dv, du = R.sgns_step(v, c_slots, u, o_slots, labels, lr=0.1)
-    s_before = torch.sigmoid(v[0] @ u[1])
-    v2, u2 = v + dv, u + du
-    s_after = torch.sigmoid(v2[0] @ u2[1])
+    d = R.sgns_step(vecs, c_slots, o_slots, labels, lr=0.1)
+    s_before = torch.sigmoid(vecs[0] @ vecs[3])
+    after = vecs + d
+    s_after = torch.sigmoid(after[0] @ after[3])
     assert s_after > s_before          # positive pair pulled together
-    neg_before = torch.sigmoid(v[0] @ u[2])
-    neg_after = torch.sigmoid(v2[0] @ u2[2])
+    neg_before = torch.sigmoid(vecs[0] @ vecs[4])
+    neg_after = torch.sigmoid(after[0] @ after[4])
     assert neg_after < neg_before      # negative pushed apart
