@@ -67,7 +67,10 @@ def make_mf(args, rank, world, device):
         u = torch.rand(args.batch, generator=g)
         items = (n_items * u.pow(3.0)).long().clamp_(0, n_items - 1)
         ratings = torch.rand(args.batch, generator=g) * 4.0 + 1.0
-        batches.append(MFBatch(users.to(device), items.to(device),
+        # int32 indices: the fused kernels take them directly, halving
+        # the metadata bytes of the histogram/pack passes
+        batches.append(MFBatch(users.to(device, torch.int32),
+                               items.to(device, torch.int32),
                                ratings.to(device)))
     cfg = PSConfig(num_ids=n_items, dim=k, staleness=args.staleness)
     meta = {"model": f"SGD-MF rank{k} MovieLens-1M-shaped" +

@@ -812,7 +812,15 @@ static torch::Tensor build_rec_swap(torch::Tensor src, int64_t m,
 template <class F>
 static void by_group4(int64_t dim4, F f) {
   using namespace std;
-  if (dim4 <= 8) f(integral_constant<int, 8>{}, integral_constant<int, 1>{});
+  // FPS_SMALL_GP=4: 4-lane groups (2 float4/lane) for dim<=32 — twice
+  // the independent chunk walks per wave at one fewer shfl per dot
+  static int small_gp = env_int("FPS_SMALL_GP", 8);
+  if (dim4 <= 8) {
+    if (small_gp == 4)
+      f(integral_constant<int, 4>{}, integral_constant<int, 2>{});
+    else
+      f(integral_constant<int, 8>{}, integral_constant<int, 1>{});
+  }
   else if (dim4 <= 16) f(integral_constant<int, 8>{}, integral_constant<int, 2>{});
   else if (dim4 <= 32) f(integral_constant<int, 8>{}, integral_constant<int, 4>{});
   else if (dim4 <= 48) f(integral_constant<int, 16>{}, integral_constant<int, 3>{});
@@ -880,8 +888,9 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   if (smallA && smallB && m + U <= 2 * PACK_MAX_M &&
       item_slots.dtype() == user_rows.dtype()) {
     auto opts_i32 = torch::dtype(torch::kInt32).device(ratings.device());
-    histA = torch::zeros({m}, opts_i32);
-    histB = torch::zeros({U}, opts_i32);
+    auto hist_both = torch::zeros({m + U}, opts_i32);  // one fill launch
+    histA = hist_both.narrow(0, 0, m);
+    histB = hist_both.narrow(0, m, U);
     int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
     size_t lds = (size_t)(m + U) * sizeof(int);
     if (item_slots.dtype() == torch::kInt32)

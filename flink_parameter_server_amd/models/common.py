@@ -37,8 +37,15 @@ class PullBuffer:
         """Fast path: the caller guarantees ``set(ids) == set(self.uids)``
         (e.g. the engine answered this buffer's whole pull in one round and
         no other buffer was pending).  No host-device syncs."""
-        rows = torch.searchsorted(self.uids, ids)
-        self.values[rows] = values.to(self.values.dtype)
+        if (ids.data_ptr() == self.uids.data_ptr()
+                and ids.numel() == self.uids.numel()
+                and values.dtype == self.values.dtype):
+            # in-order answers return the very tensor we pulled (dense
+            # pull / single-shard): adopt the payload, skip the permute
+            self.values = values
+        else:
+            rows = torch.searchsorted(self.uids, ids)
+            self.values[rows] = values.to(self.values.dtype)
         self._have.fill_(True)
         self._n_have = int(self.uids.numel())
 

@@ -81,8 +81,9 @@ def mf_sgd_step(user_table: torch.Tensor, user_rows: torch.Tensor,
     if _cuda(user_table):
         return hip.ext().mf_sgd_step(user_table, user_rows, item_vecs,
                                      item_slots, ratings, lr, reg)
-    return reference.mf_sgd_step(user_table, user_rows, item_vecs,
-                                 item_slots, ratings, lr, reg)
+    # torch index ops need int64; the HIP path takes int32 directly
+    return reference.mf_sgd_step(user_table, user_rows.long(), item_vecs,
+                                 item_slots.long(), ratings, lr, reg)
 
 
 def pa_step(indptr: torch.Tensor, feat_slots: torch.Tensor,

@@ -49,9 +49,13 @@ class ModPartitioner(Partitioner):
         self.n_shards = int(n_shards)
 
     def shard_of(self, ids: torch.Tensor) -> torch.Tensor:
+        if self.n_shards == 1:
+            return torch.zeros_like(ids)
         return ids % self.n_shards
 
     def local_row(self, ids: torch.Tensor) -> torch.Tensor:
+        if self.n_shards == 1:  # identity — skip the int64 div kernel
+            return ids
         return ids // self.n_shards
 
     def global_id(self, shard: int, rows: torch.Tensor) -> torch.Tensor:

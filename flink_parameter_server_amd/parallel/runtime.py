@@ -331,15 +331,16 @@ class PSEngine:
                     total += t.numel() * t.element_size()
         return total
 
-    @staticmethod
-    def _wire_out(chan: List[Optional[torch.Tensor]],
+    def _wire_out(self, chan: List[Optional[torch.Tensor]],
                   wire: torch.dtype) -> None:
         """Cast a payload channel to the wire dtype in place (bf16 halves
-        xGMI bytes; tables keep f32 masters, so only the wire narrows)."""
+        xGMI bytes; tables keep f32 masters, so only the wire narrows).
+        The self-addressed segment short-circuits in every transport and
+        stays f32 — full precision for the 1/world local fraction."""
         if wire is torch.float32:
             return
         for d, t in enumerate(chan):
-            if t is not None:
+            if t is not None and d != self.rank:
                 chan[d] = t.to(wire)
 
     @staticmethod

@@ -35,6 +35,7 @@ class ShardedTable:
                  eps: float = 1e-8, seed: int = 0):
         self.part = part
         self.shard = shard
+        self._full_ids_key = None
         self.dim = int(dim)
         self.device = torch.device(device)
         self.optimizer = optimizer
@@ -73,8 +74,32 @@ class ShardedTable:
     # -- server-side ops (K3 gather / K1 sgd apply / K2 adagrad apply) ----
 
     def gather(self, ids: torch.Tensor) -> torch.Tensor:
+        if self._is_full_range(ids):
+            # dense pull of the whole shard: the "gather" is the slab
+            # itself.  Answer values alias the live table; the engine's
+            # round order (worker consumes answers BEFORE this round's
+            # pushes apply) makes that observationally identical to a
+            # copy taken at answer time.
+            return self.data
         rows = self.part.local_row(ids)
         return ops.gather(self.data, rows)
+
+    def _is_full_range(self, ids: torch.Tensor) -> bool:
+        """ids == every row of this shard in local-row order?  Verified
+        with one tensor compare on first sight, then recognized by
+        buffer identity (dense-pull workers re-send one cached tensor)."""
+        if ids.numel() != self.rows or self.rows == 0:
+            return False
+        key = (ids.data_ptr(), ids.device)
+        if key == self._full_ids_key:
+            return True
+        rows = self.part.local_row(ids)
+        expect = torch.arange(self.rows, dtype=rows.dtype,
+                              device=rows.device)
+        if bool(torch.equal(rows, expect)):
+            self._full_ids_key = key
+            return True
+        return False
 
     def apply(self, ids: torch.Tensor, deltas: torch.Tensor,
               unique: bool = False) -> None:
