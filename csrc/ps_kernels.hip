@@ -499,9 +499,10 @@ __global__ __launch_bounds__(256) void k_seg_fused(
 // rep_stride > 0 spreads the delta scatter over 8 replica copies of
 // `deltas` (hot zipfian features otherwise serialize same-address
 // atomics); the host sums the replicas.
+template <class TIdx>
 __global__ __launch_bounds__(256) void k_sparse_linear_step(
     const int64_t* __restrict__ indptr,
-    const int64_t* __restrict__ feat_slots,
+    const TIdx* __restrict__ feat_slots,
     const float* __restrict__ x_vals, const float* __restrict__ labels,
     const float* __restrict__ weights, float* __restrict__ margins,
     float* __restrict__ deltas, int64_t B, int mode, float C,
@@ -517,7 +518,7 @@ __global__ __launch_bounds__(256) void k_sparse_linear_step(
     float m = 0.f, n2 = 0.f;
     for (int64_t j = lo + lane; j < hi; j += WAVE) {
       float x = x_vals[j];
-      m += weights[feat_slots[j]] * x;
+      m += weights[(int64_t)feat_slots[j]] * x;
       n2 += x * x;
     }
     m = group_sum<WAVE>(m);
@@ -541,7 +542,8 @@ __global__ __launch_bounds__(256) void k_sparse_linear_step(
       coeff = tau * y;
     }
     for (int64_t j = lo + lane; j < hi; j += WAVE)
-      atomicAdd(&deltas[rep_off + feat_slots[j]], coeff * x_vals[j]);
+      atomicAdd(&deltas[rep_off + (int64_t)feat_slots[j]],
+                coeff * x_vals[j]);
   }
 }
 
@@ -881,6 +883,13 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   TORCH_CHECK(dim % 4 == 0, "mf_sgd_step requires dim % 4 == 0");
   if (!n) return torch::zeros_like(item_vecs);
   int64_t dim4 = dim / 4;
+  if (item_slots.dtype() != user_rows.dtype()) {
+    // harmonize to int32 (both ranges are table heights < 2^31)
+    if (item_slots.dtype() != torch::kInt32)
+      item_slots = item_slots.to(torch::kInt32);
+    if (user_rows.dtype() != torch::kInt32)
+      user_rows = user_rows.to(torch::kInt32);
+  }
 
   // histograms: one fused read when both key ranges are LDS-sized
   torch::Tensor histA, histB;
@@ -1008,14 +1017,17 @@ std::vector<torch::Tensor> pa_step(torch::Tensor indptr,
   int64_t reps = (x_vals.numel() >= 8 * m &&
                   8 * m * 4 <= (256ll << 20)) ? 8 : 1;
   auto dbuf = torch::zeros({reps * m}, weights.options());
-  hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
-                     dim3(256), 0, cur_stream(),
-                     indptr.data_ptr<int64_t>(),
-                     feat_slots.data_ptr<int64_t>(),
-                     x_vals.data_ptr<float>(), labels.data_ptr<float>(),
-                     weights.data_ptr<float>(), margins.data_ptr<float>(),
-                     dbuf.data_ptr<float>(), B, pa_mode(variant),
-                     (float)C, reps > 1 ? m : 0);
+  auto launch = [&](auto* slots) {
+    hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
+                       dim3(256), 0, cur_stream(),
+                       indptr.data_ptr<int64_t>(), slots,
+                       x_vals.data_ptr<float>(), labels.data_ptr<float>(),
+                       weights.data_ptr<float>(), margins.data_ptr<float>(),
+                       dbuf.data_ptr<float>(), B, pa_mode(variant),
+                       (float)C, reps > 1 ? m : 0);
+  };
+  if (feat_slots.dtype() == torch::kInt32) launch(feat_slots.data_ptr<int>());
+  else launch(feat_slots.data_ptr<int64_t>());
   auto deltas = reps > 1 ? dbuf.view({reps, m}).sum(0) : dbuf;
   return {margins, deltas};
 }
@@ -1033,14 +1045,17 @@ std::vector<torch::Tensor> lr_step(torch::Tensor indptr,
   int64_t reps = (x_vals.numel() >= 8 * m &&
                   8 * m * 4 <= (256ll << 20)) ? 8 : 1;
   auto gbuf = torch::zeros({reps * m}, weights.options());
-  hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
-                     dim3(256), 0, cur_stream(),
-                     indptr.data_ptr<int64_t>(),
-                     feat_slots.data_ptr<int64_t>(),
-                     x_vals.data_ptr<float>(), labels.data_ptr<float>(),
-                     weights.data_ptr<float>(), probs.data_ptr<float>(),
-                     gbuf.data_ptr<float>(), B, /*mode=*/3, 0.f,
-                     reps > 1 ? m : 0);
+  auto launch = [&](auto* slots) {
+    hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
+                       dim3(256), 0, cur_stream(),
+                       indptr.data_ptr<int64_t>(), slots,
+                       x_vals.data_ptr<float>(), labels.data_ptr<float>(),
+                       weights.data_ptr<float>(), probs.data_ptr<float>(),
+                       gbuf.data_ptr<float>(), B, /*mode=*/3, 0.f,
+                       reps > 1 ? m : 0);
+  };
+  if (feat_slots.dtype() == torch::kInt32) launch(feat_slots.data_ptr<int>());
+  else launch(feat_slots.data_ptr<int64_t>());
   auto grads = reps > 1 ? gbuf.view({reps, m}).sum(0) : gbuf;
   // kernel writes margins; convert to probabilities in-place
   probs.sigmoid_();

@@ -91,10 +91,13 @@ class DenseCoalescer:
                                    device=device)
 
     def __call__(self, ids: torch.Tensor):
+        """Returns (sorted unique int64 ids, int32 slot of each input id)
+        — int32 slots feed the HIP kernels directly (half the metadata
+        bytes of the sort/pack passes)."""
         present = self.present
         present.zero_()
         present[ids] = True
         lookup = torch.cumsum(present, 0, dtype=torch.int32)
         uids = present.nonzero().reshape(-1)
-        slots = (lookup[ids].to(torch.int64) - 1)
+        slots = lookup[ids] - 1
         return uids, slots

@@ -113,7 +113,8 @@ def lr_step(indptr: torch.Tensor, feat_slots: torch.Tensor,
             weights: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     if _cuda(weights):
         return hip.ext().lr_step(indptr, feat_slots, x_vals, labels, weights)
-    return reference.lr_step(indptr, feat_slots, x_vals, labels, weights)
+    return reference.lr_step(indptr, feat_slots.long(), x_vals, labels,
+                             weights)
 
 
 def sgns_step(vecs: torch.Tensor, center_slots: torch.Tensor,
@@ -124,4 +125,5 @@ def sgns_step(vecs: torch.Tensor, center_slots: torch.Tensor,
     if _cuda(vecs):
         return hip.ext().sgns_step(vecs, center_slots, ctx_slots, labels,
                                    lr)
-    return reference.sgns_step(vecs, center_slots, ctx_slots, labels, lr)
+    return reference.sgns_step(vecs, center_slots.long(),
+                               ctx_slots.long(), labels, lr)
