@@ -41,9 +41,15 @@
 
 #include <cstdlib>
 
+#include <rocprim/rocprim.hpp>
+
 #define WAVE 64
 #define CHECK_DEV(x) TORCH_CHECK((x).is_cuda(), #x " must be a device tensor")
 #define CHECK_CONT(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
 
 static inline int grid_for(int64_t work, int block) {
   int64_t g = (work + block - 1) / block;
@@ -345,39 +351,59 @@ __global__ __launch_bounds__(256) void k_pack_rec_swap(
   }
 }
 
-// radix-sort epilogue (large key ranges): gather a rocPRIM argsort
-// permutation into interleaved recs — replaces 3 index_selects.
+// large key ranges: build UNSORTED recs, then rocPRIM radix-sorts
+// (key, rec) pairs directly over just ceil(log2(m)) bits — no argsort
+// permutation, no gather epilogue.
 template <class TK>
-__global__ __launch_bounds__(256) void k_make_rec(
+__global__ __launch_bounds__(256) void k_interleave_rec(
     const TK* __restrict__ keys, const TK* __restrict__ slots,
-    const float* __restrict__ vals, const int64_t* __restrict__ order,
+    const float* __restrict__ vals, int* __restrict__ okeys,
     int4* __restrict__ rec, int64_t n) {
   for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
        j += (int64_t)gridDim.x * blockDim.x) {
-    const int64_t o = order[j];
-    rec[j] = make_int4((int)keys[o], (int)slots[o],
-                       __float_as_int(vals[o]), 0);
+    const int k = (int)keys[j];
+    okeys[j] = k;
+    rec[j] = make_int4(k, (int)slots[j], __float_as_int(vals[j]), 0);
   }
 }
 
-__global__ __launch_bounds__(256) void k_make_rec_swap(
-    const int4* __restrict__ src, const int64_t* __restrict__ order,
+// swap key<->slot of an existing rec array; emits the new key column
+__global__ __launch_bounds__(256) void k_interleave_swap(
+    const int4* __restrict__ src, int* __restrict__ okeys,
     int4* __restrict__ rec, int64_t n) {
   for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
        j += (int64_t)gridDim.x * blockDim.x) {
-    const int4 s = src[order[j]];
+    const int4 s = src[j];
+    okeys[j] = s.y;
     rec[j] = make_int4(s.y, s.x, s.z, 0);
   }
 }
 
-// pull the key (x) or slot (y) field out of a rec array (argsort input
-// for the large-key-range pass-B path)
-__global__ __launch_bounds__(256) void k_rec_field(
-    const int4* __restrict__ rec, int* __restrict__ out, int64_t n,
-    int field) {
-  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
-       j += (int64_t)gridDim.x * blockDim.x)
-    out[j] = field ? rec[j].y : rec[j].x;
+static int bits_for(int64_t m) {
+  int b = 1;
+  while ((1ll << b) < m) ++b;
+  return b;
+}
+
+// radix-sort recs by the freshly built key column (int32, bit-bounded)
+static torch::Tensor radix_sort_recs(torch::Tensor keys,
+                                     torch::Tensor rec, int64_t n,
+                                     int64_t m) {
+  auto opts = keys.options();
+  auto keys_out = torch::empty_like(keys);
+  auto rec_out = torch::empty_like(rec);
+  size_t tmp_bytes = 0;
+  rocprim::radix_sort_pairs(
+      nullptr, tmp_bytes, keys.data_ptr<int>(), keys_out.data_ptr<int>(),
+      (const int4*)rec.data_ptr<int>(), (int4*)rec_out.data_ptr<int>(),
+      n, 0, bits_for(m), cur_stream());
+  auto tmp = torch::empty({(int64_t)tmp_bytes},
+                          opts.dtype(torch::kUInt8));
+  rocprim::radix_sort_pairs(
+      tmp.data_ptr(), tmp_bytes, keys.data_ptr<int>(),
+      keys_out.data_ptr<int>(), (const int4*)rec.data_ptr<int>(),
+      (int4*)rec_out.data_ptr<int>(), n, 0, bits_for(m), cur_stream());
+  return rec_out;
 }
 
 // =====================================================================
@@ -397,7 +423,7 @@ __global__ __launch_bounds__(256) void k_rec_field(
 // second pass.  rep_stride spreads flushes over 8 replica copies of
 // `out` (zipfian hot keys).
 // =====================================================================
-template <int GP, int EPL, int MODE, bool SELF>
+template <int GP, int EPL, int MODE, bool SELF, bool ZERO_OUT>
 __global__ __launch_bounds__(256) void k_seg_fused(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
@@ -415,6 +441,12 @@ __global__ __launch_bounds__(256) void k_seg_fused(
     float4 acc[EPL], kv[EPL];
     int cnt = 0;
     int cur = rec[start].x;
+    // a segment that both BEGINS and ENDS strictly inside this chunk is
+    // provably this walk's exclusive property (keys are globally
+    // sorted), so its flush is a plain vector store — near-unique key
+    // distributions otherwise degenerate into one 4-B-atomic storm per
+    // sample (measured: the SGNS context pass was 52% of step time).
+    bool left_owned = (start == 0);
 
     auto load_kv = [&]() {
       if (MODE == 2 && !SELF) return;  // keyed row unused
@@ -425,7 +457,8 @@ __global__ __launch_bounds__(256) void k_seg_fused(
         kv[e] = (c < dim4) ? keyed[kr + c] : make_float4(0, 0, 0, 0);
       }
     };
-    auto flush = [&]() {
+    auto flush = [&](bool right_owned) {
+      const bool excl = left_owned && right_owned;
       const int64_t orow = rep_off + (int64_t)cur * dim4;
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
@@ -438,9 +471,19 @@ __global__ __launch_bounds__(256) void k_seg_fused(
             t.x += ac * kv[e].x; t.y += ac * kv[e].y;
             t.z += ac * kv[e].z; t.w += ac * kv[e].w;
           }
-          float* o = (float*)&out[orow + c];
-          atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
-          atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
+          if (excl) {
+            if (ZERO_OUT) {
+              out[orow + c] = t;
+            } else {
+              float4 o = out[orow + c];
+              o.x += t.x; o.y += t.y; o.z += t.z; o.w += t.w;
+              out[orow + c] = o;
+            }
+          } else {
+            float* o = (float*)&out[orow + c];
+            atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
+            atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
+          }
         }
         acc[e] = make_float4(0, 0, 0, 0);
       }
@@ -452,7 +495,8 @@ __global__ __launch_bounds__(256) void k_seg_fused(
     for (int64_t j = start; j < end; ++j) {
       const int4 rj = rec[j];
       if (rj.x != cur) {
-        flush();
+        flush(true);  // key changed inside the chunk: right-bounded
+        left_owned = true;
         cur = rj.x;
         cnt = 0;
         load_kv();
@@ -487,7 +531,7 @@ __global__ __launch_bounds__(256) void k_seg_fused(
       }
       ++cnt;
     }
-    flush();
+    flush(end == n);  // right-bounded only at the array end
   }
 }
 
@@ -604,10 +648,6 @@ __global__ __launch_bounds__(256) void k_pa_multiclass_step(
 // =====================================================================
 // Host wrappers
 // =====================================================================
-
-static hipStream_t cur_stream() {
-  return at::hip::getCurrentHIPStream().stream();
-}
 
 torch::Tensor gather(torch::Tensor table, torch::Tensor rows) {
   CHECK_DEV(table); CHECK_DEV(rows); CHECK_CONT(table); CHECK_CONT(rows);
@@ -758,12 +798,13 @@ static torch::Tensor build_rec_t(torch::Tensor keys, torch::Tensor slots,
                        vals.data_ptr<float>(), cursor.data_ptr<int>(),
                        (int4*)rec.data_ptr<int>(), n, (int)m, chunk);
   } else {
-    auto order = torch::argsort(keys);
-    hipLaunchKernelGGL((k_make_rec<TK>), dim3(grid_for(n, 256)),
+    auto keys32 = torch::empty({n}, opts_i32);
+    hipLaunchKernelGGL((k_interleave_rec<TK>), dim3(grid_for(n, 256)),
                        dim3(256), 0, cur_stream(), keys.data_ptr<TK>(),
                        slots.data_ptr<TK>(), vals.data_ptr<float>(),
-                       order.data_ptr<int64_t>(),
-                       (int4*)rec.data_ptr<int>(), n);
+                       keys32.data_ptr<int>(), (int4*)rec.data_ptr<int>(),
+                       n);
+    rec = radix_sort_recs(keys32, rec, n, m);
   }
   return rec;
 }
@@ -794,15 +835,13 @@ static torch::Tensor build_rec_swap(torch::Tensor src, int64_t m,
                        cursor.data_ptr<int>(), (int4*)rec.data_ptr<int>(),
                        n, (int)m, chunk);
   } else {
-    auto keys = torch::empty({n}, opts_i32);
-    hipLaunchKernelGGL(k_rec_field, dim3(grid_for(n, 256)), dim3(256), 0,
-                       cur_stream(), (const int4*)src.data_ptr<int>(),
-                       keys.data_ptr<int>(), n, 1);
-    auto order = torch::argsort(keys);
-    hipLaunchKernelGGL(k_make_rec_swap, dim3(grid_for(n, 256)), dim3(256),
-                       0, cur_stream(), (const int4*)src.data_ptr<int>(),
-                       order.data_ptr<int64_t>(),
-                       (int4*)rec.data_ptr<int>(), n);
+    auto keys32 = torch::empty({n}, opts_i32);
+    hipLaunchKernelGGL(k_interleave_swap, dim3(grid_for(n, 256)),
+                       dim3(256), 0, cur_stream(),
+                       (const int4*)src.data_ptr<int>(),
+                       keys32.data_ptr<int>(), (int4*)rec.data_ptr<int>(),
+                       n);
+    rec = radix_sort_recs(keys32, rec, n, m);
   }
   return rec;
 }
@@ -831,8 +870,10 @@ static void by_group4(int64_t dim4, F f) {
          f(integral_constant<int, 32>{}, integral_constant<int, 4>{}); }
 }
 
-// fused segment-walk launcher
-template <int MODE, bool SELF>
+// fused segment-walk launcher.  ZERO_OUT: `out` is zero-initialized and
+// exclusively written by this launch, so exclusive-segment flushes may
+// store instead of read-add.
+template <int MODE, bool SELF, bool ZERO_OUT>
 static void seg_fused(torch::Tensor out, torch::Tensor rec,
                       const torch::Tensor& keyed,
                       const torch::Tensor& other, int64_t dim4,
@@ -845,7 +886,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
   by_group4(dim4, [&](auto tag, auto etag) {
     constexpr int GP = decltype(tag)::value;
     constexpr int EPL = decltype(etag)::value;
-    hipLaunchKernelGGL((k_seg_fused<GP, EPL, MODE, SELF>),
+    hipLaunchKernelGGL((k_seg_fused<GP, EPL, MODE, SELF, ZERO_OUT>),
                        dim3(grid_for(nchunks * GP, 256)), dim3(256), 0,
                        cur_stream(), (float4*)out.data_ptr<float>(),
                        (int4*)rec.data_ptr<int>(),
@@ -923,13 +964,13 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   torch::Tensor deltas;
   if (reps > 1) {
     auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
-    seg_fused<0, true>(rep_buf, recA, item_vecs, user_table, dim4, lr,
-                       -lr * reg, 0.0, m * dim);
+    seg_fused<0, true, true>(rep_buf, recA, item_vecs, user_table, dim4,
+                             lr, -lr * reg, 0.0, m * dim);
     deltas = rep_buf.view({reps, m, dim}).sum(0);
   } else {
     deltas = torch::zeros_like(item_vecs);
-    seg_fused<0, true>(deltas, recA, item_vecs, user_table, dim4, lr,
-                       -lr * reg, 0.0);
+    seg_fused<0, true, true>(deltas, recA, item_vecs, user_table, dim4,
+                             lr, -lr * reg, 0.0);
   }
 
   // pass B: recs by user row; apply user updates in place
@@ -950,8 +991,8 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                          histB.data_ptr<int>(), n, (int)U);
   }
   auto recB = build_rec_swap(recA, U, histB);
-  seg_fused<2, true>(user_table, recB, user_table, item_vecs, dim4, lr,
-                     -lr * reg, 0.0);
+  seg_fused<2, true, false>(user_table, recB, user_table, item_vecs,
+                            dim4, lr, -lr * reg, 0.0);
   return deltas;
 }
 
@@ -973,7 +1014,8 @@ torch::Tensor sgns_step(torch::Tensor vecs, torch::Tensor center_slots,
   if (!n) return deltas;
   int64_t dim4 = dim / 4;
   auto recC = build_rec(center_slots, ctx_slots, labels, m);
-  seg_fused<1, false>(deltas, recC, vecs, vecs, dim4, 1.0, 0.0, lr);
+  seg_fused<1, false, true>(deltas, recC, vecs, vecs, dim4, 1.0, 0.0,
+                            lr);
   torch::Tensor histD;
   if (m <= PACK_MAX_M) {
     auto opts_i32 = torch::dtype(torch::kInt32).device(vecs.device());
@@ -991,7 +1033,8 @@ torch::Tensor sgns_step(torch::Tensor vecs, torch::Tensor center_slots,
                          histD.data_ptr<int>(), n, (int)m);
   }
   auto recD = build_rec_swap(recC, m, histD);
-  seg_fused<2, false>(deltas, recD, vecs, vecs, dim4, 1.0, 0.0, 0.0);
+  seg_fused<2, false, true>(deltas, recD, vecs, vecs, dim4, 1.0, 0.0,
+                            0.0);
   return deltas;
 }
 

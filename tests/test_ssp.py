@@ -91,3 +91,11 @@ def test_exhausted_worker_releases_the_clock():
     (fast_c, _, _), (slow_c, _, _) = _run(staleness=1, fast_n=40,
                                           slow_n=3, slow_every=2)
     assert fast_c == 40 and slow_c == 3
+
+
+def test_staleness_histogram_recorded():
+    (fast_c, _, fast_m), _ = _run(staleness=2)
+    hist = fast_m["staleness_hist"]
+    assert sum(hist.values()) == fast_m["rounds"]
+    assert max(hist) <= 2 + 1  # bounded lag distribution
+    assert min(hist) >= 0
