@@ -27,3 +27,5 @@ from .messaging import (CountFlush, FlushAlways, FlushStrategy, Pull,
                         TimeFlush, WorkerToPS)
 from .algorithms import (ps_offline_mf, ps_online_mf, transform_binary,
                          transform_multiclass)
+from .utils.checkpoint import (PeriodicCheckpointer, latest_checkpoint,
+                               load_shard, save_shard)

@@ -12,6 +12,8 @@ a duplicate answer must not double-count).
 """
 from __future__ import annotations
 
+from typing import Optional
+
 import torch
 
 
@@ -25,9 +27,18 @@ class PullBuffer:
         self.uids = uids
         self.values = torch.empty(uids.numel(), dim, dtype=torch.float32,
                                   device=uids.device)
-        self._have = torch.zeros(uids.numel(), dtype=torch.bool,
-                                 device=uids.device)
+        # lazily allocated: the whole-answer fast path never needs the
+        # per-row flags (two kernel launches per round saved)
+        self._have_t: Optional[torch.Tensor] = None
         self._n_have = 0
+
+    @property
+    def _have(self) -> torch.Tensor:
+        if self._have_t is None:
+            self._have_t = torch.zeros(self.uids.numel(),
+                                       dtype=torch.bool,
+                                       device=self.uids.device)
+        return self._have_t
 
     def slots_of(self, ids: torch.Tensor) -> torch.Tensor:
         """Position of each id in the sorted unique-id list."""
@@ -46,7 +57,8 @@ class PullBuffer:
         else:
             rows = torch.searchsorted(self.uids, ids)
             self.values[rows] = values.to(self.values.dtype)
-        self._have.fill_(True)
+        if self._have_t is not None:
+            self._have_t.fill_(True)
         self._n_have = int(self.uids.numel())
 
     def offer(self, ids: torch.Tensor, values: torch.Tensor) -> None:

@@ -41,11 +41,20 @@ def apply_add(table: torch.Tensor, rows: torch.Tensor,
 
 def apply_adagrad(table: torch.Tensor, accum: torch.Tensor,
                   rows: torch.Tensor, grads: torch.Tensor,
-                  lr: float, eps: float = 1e-8) -> None:
+                  lr: float, eps: float = 1e-8,
+                  sorted_unique: bool = False) -> None:
+    """``sorted_unique``: the caller guarantees ``rows`` ascending with no
+    duplicates (our model logics push per sorted unique pulled id), so
+    the pre-sort is skipped."""
     if _cuda(table):
         # K2 contract: duplicates pre-aggregated, one RMW per distinct
         # row.  Sort the batch (device radix sort) and let segment
         # leaders aggregate in-kernel.
+        if sorted_unique:
+            hip.ext().apply_adagrad_sorted(table, accum,
+                                           rows.contiguous(),
+                                           grads.contiguous(), lr, eps)
+            return
         srows, order = torch.sort(rows)
         hip.ext().apply_adagrad_sorted(table, accum, srows.contiguous(),
                                        grads.index_select(0, order)

@@ -107,8 +107,11 @@ class ShardedTable:
         if self.optimizer == "add":
             ops.apply_add(self.data, rows, deltas, unique)
         else:
+            # `unique` batches arrive as sorted unique pulled ids, and
+            # both partitioners are monotone per shard, so the local
+            # rows are already sorted — the adagrad pre-sort is skipped.
             ops.apply_adagrad(self.data, self.accum, rows, deltas,
-                              self.lr, self.eps)
+                              self.lr, self.eps, sorted_unique=unique)
 
     def dump(self) -> Tuple[torch.Tensor, torch.Tensor]:
         """(global_ids, values) of every row of the shard — the reference's
@@ -140,9 +143,12 @@ class TableServerLogic:
     def __init__(self, table: ShardedTable, dump_on_close: bool = True,
                  pushes_unique: bool = False):
         """``pushes_unique=True``: the application guarantees every push
-        batch carries distinct ids (all bundled model logics do — they
-        pre-aggregate deltas per pulled row), enabling the non-atomic
-        apply fast path."""
+        batch carries SORTED distinct ids (all bundled model logics do —
+        they push pre-aggregated deltas per sorted unique pulled id, one
+        push per round), enabling the non-atomic add fast path and the
+        sort-free adagrad path.  Leave False when pushes may repeat or
+        interleave ids (e.g. under a count/time flush strategy that
+        coalesces several logical pushes into one round)."""
         self.table = table
         self.dump_on_close = dump_on_close
         self.pushes_unique = pushes_unique
