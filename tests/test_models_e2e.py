@@ -167,3 +167,42 @@ def test_sgns_objective_decreases():
     eng.run(pairs * 3)
     after = objective()
     assert after < before * 0.9
+
+
+def test_online_mf_dense_pull_under_pull_limit():
+    """Dense pull (whole item range each round) with a pull limit smaller
+    than the item count: answers split across rounds, PullBuffer
+    reassembles, training still converges."""
+    n_users, n_items, rank = 40, 30, 4
+    batches, _, _ = synth.planted_mf_ratings(n_users, n_items, rank,
+                                             2000, 128, seed=7)
+    part = ModPartitioner(n_items, 1)
+    table = ShardedTable(part, 0, rank, "cpu", init=(-0.1, 0.1))
+    wl = OnlineMFWorkerLogic(n_users, rank, learning_rate=0.05,
+                             range_min=-0.1, range_max=0.1,
+                             n_items=n_items, pull_limit=7)
+    assert wl._dense_pull
+    eng = PSEngine(PSConfig(num_ids=n_items, dim=rank), wl,
+                   TableServerLogic(table), None, part)
+    l0 = _mf_loss(wl, table, batches)
+    eng.run(epoch_iterator(batches, 6))
+    l1 = _mf_loss(wl, table, batches)
+    # every sample processed despite split answers; learning happened
+    # (staleness under a tight limit slows convergence, not correctness)
+    assert eng.metrics.samples == 2000 * 6
+    assert eng.metrics.pull_ids_sent > 6 * (2000 // 128) * 7
+    assert l1 < 0.9 * l0
+
+
+def test_full_range_gather_aliases_slab():
+    part = ModPartitioner(16, 1)
+    table = ShardedTable(part, 0, 2, "cpu", init=(-0.1, 0.1))
+    full = torch.arange(16, dtype=torch.int64)
+    out = table.gather(full)
+    assert out.data_ptr() == table.data.data_ptr()  # alias, no copy
+    out2 = table.gather(full)  # cached-key fast path
+    assert out2.data_ptr() == table.data.data_ptr()
+    part_ids = torch.tensor([3, 5])
+    sub = table.gather(part_ids)
+    assert sub.data_ptr() != table.data.data_ptr()
+    assert torch.equal(sub, table.data[[3, 5]])
