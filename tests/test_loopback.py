@@ -146,3 +146,58 @@ def test_range_partitioner_across_ranks():
         merged.update(d)
     assert len(merged) == n_ids
     assert all(v == world for v in merged.values())
+
+
+def test_randomized_world4_stress():
+    """4 ranks, 25 randomized rounds of interleaved pulls/pushes.
+    '+' server semantics are order-independent, so the final table must
+    equal the global per-id push totals, and every pull must have been
+    answered exactly once (in-order feedback edge)."""
+    world, n_ids, dim, rounds = 4, 57, 3, 25
+
+    class RandomTraffic(WorkerLogic):
+        def __init__(self, rank):
+            self.g = torch.Generator().manual_seed(100 + rank)
+            self.pushed = torch.zeros(n_ids, dim)
+            self.pending = {}
+            self.answered = 0
+
+        def on_recv(self, data, client):
+            k = int(torch.randint(1, 12, (1,), generator=self.g))
+            ids = torch.unique(torch.randint(0, n_ids, (k,),
+                                             generator=self.g))
+            deltas = torch.randn(ids.numel(), dim, generator=self.g)
+            client.push(ids, deltas)
+            self.pushed[ids] += deltas
+            pull = torch.unique(torch.randint(0, n_ids, (k,),
+                                              generator=self.g))
+            client.pull(pull, unique=True)
+            for i in pull.tolist():
+                self.pending[i] = self.pending.get(i, 0) + 1
+
+        def on_pull_recv(self, ids, values, client):
+            self.answered += int(ids.numel())
+            for i in ids.tolist():
+                assert self.pending.get(i, 0) > 0, "unrequested answer"
+                self.pending[i] -= 1
+
+    def job(rank, transport):
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, dim, "cpu")
+        logic = RandomTraffic(rank)
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=dim), logic,
+                       TableServerLogic(table), transport, part)
+        eng.run([None or r for r in range(rounds)])
+        assert all(v == 0 for v in logic.pending.values()), \
+            "unanswered pulls at drain"
+        gids, vals = table.dump()
+        return logic.pushed, dict(
+            zip(gids.tolist(), [v.clone() for v in vals]))
+
+    results = run_loopback(world, job)
+    expected = sum(r[0] for r in results)
+    merged = {}
+    for _, d in results:
+        merged.update(d)
+    got = torch.stack([merged[i] for i in range(n_ids)])
+    assert torch.allclose(got, expected, atol=1e-5)
