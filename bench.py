@@ -143,6 +143,9 @@ def main():
     ap.add_argument("--n-batches", type=int, default=8,
                     help="distinct pre-generated batches cycled")
     ap.add_argument("--device", default=None)
+    ap.add_argument("--feedback", choices=["fresh", "merged"], default=None,
+                    help="feedback-edge schedule (default: fresh at 1 "
+                         "GPU, merged for multi-GPU)")
     args = ap.parse_args()
 
     world_env = int(os.environ.get("WORLD_SIZE", "1"))
@@ -172,6 +175,11 @@ def main():
 
     cfg, part, table, wl, batches, meta = MAKERS[args.model](
         args, rank, world, device)
+    # multi-GPU: answers ride the next round's single exchange — one
+    # all_gather + one grouped send/recv per round (collective latency
+    # is the scaling tax; +1 round of async staleness is free throughput)
+    fb = args.feedback or ("merged" if world > 1 else "fresh")
+    cfg.feedback = fb
     eng = PSEngine(cfg, wl, TableServerLogic(table, dump_on_close=False,
                                     pushes_unique=True),
                    transport, part, device=device)
@@ -233,6 +241,7 @@ def main():
                        "batch_per_gpu": args.batch,
                        "staleness": args.staleness,
                        "parallelism": f"worker-dp{world} x ps-shard{world}",
+                       "feedback": cfg.feedback,
                        "device": str(device)},
         }
         print(json.dumps(result))

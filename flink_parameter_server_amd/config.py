@@ -48,6 +48,16 @@ class PSConfig:
     # Tables always keep f32 master values.
     wire_dtype: str = "f32"
 
+    # Feedback-edge scheduling:
+    #   "fresh"  (default) — answers return on a second grouped
+    #       send/recv in the SAME round (1 all_gather + 2 p2p phases per
+    #       round; freshest values, 1-round answer latency).
+    #   "merged" — answers ride the NEXT round's single exchange
+    #       (1 all_gather + 1 p2p phase per round; +1 round of answer
+    #       latency / staleness — the throughput option for large
+    #       multi-GPU jobs where per-round collective latency dominates).
+    feedback: str = "fresh"
+
     # Partitioning of parameter ids over shards: "mod" (hash partitioner —
     # the reference's hash(paramId) % psParallelism; Scala Int hashCode is
     # the identity) or "range" (contiguous blocks, the PA path option).
@@ -65,3 +75,5 @@ class PSConfig:
             raise ValueError(f"unknown partitioner {self.partitioner!r}")
         if self.wire_dtype not in ("f32", "bf16"):
             raise ValueError(f"unknown wire_dtype {self.wire_dtype!r}")
+        if self.feedback not in ("fresh", "merged"):
+            raise ValueError(f"unknown feedback {self.feedback!r}")

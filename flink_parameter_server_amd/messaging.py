@@ -141,28 +141,22 @@ class SerializingTransport(Transport):
 
     def alltoallv(self, parts, specs, extra=None, recv_counts=None):
         C = len(parts)
-        if C == 3:  # worker -> server leg
-            envelopes = [encode_worker_to_ps(self.rank, parts[0][d],
-                                             parts[1][d], parts[2][d])
-                         for d in range(self.world)]
-            wire = [[None] * self.world for _ in range(3)]
-            for d, env in enumerate(envelopes):
+        wire = [list(chan) for chan in parts]
+        if C >= 3:  # worker -> server channels (always the first 3)
+            for d in range(self.world):
+                env = encode_worker_to_ps(self.rank, parts[0][d],
+                                          parts[1][d], parts[2][d])
                 wire[0][d], wire[1][d], wire[2][d] = \
                     decode_worker_to_ps(env)
-            return self.inner.alltoallv(wire, specs, extra=extra,
-                                        recv_counts=recv_counts)
-        if C == 2:  # feedback edge (answers with ids)
-            envelopes = [encode_ps_to_worker(d, parts[0][d], parts[1][d])
-                         for d in range(self.world)]
-            wire = [[None] * self.world for _ in range(2)]
-            for d, env in enumerate(envelopes):
-                wire[0][d], wire[1][d] = decode_ps_to_worker(env)
-            return self.inner.alltoallv(wire, specs, extra=extra,
-                                        recv_counts=recv_counts)
-        # C == 1: in-order feedback edge — answer ids are intentionally
+        if C == 2 or C == 5:  # id-carrying feedback channels (last 2)
+            for d in range(self.world):
+                env = encode_ps_to_worker(d, parts[C - 2][d],
+                                          parts[C - 1][d])
+                wire[C - 2][d], wire[C - 1][d] = decode_ps_to_worker(env)
+        # C == 1 / C == 4: in-order feedback — answer ids intentionally
         # elided from the wire (PullAnswer.param_ids reconstructed by
-        # the puller from its own sent segments); nothing to round-trip.
-        return self.inner.alltoallv(parts, specs, extra=extra,
+        # the puller from its own sent segments); values pass as-is.
+        return self.inner.alltoallv(wire, specs, extra=extra,
                                     recv_counts=recv_counts)
 
     def allgather_int(self, vec):

@@ -135,7 +135,13 @@ class PSEngine:
             [[] for _ in range(self.world)]
         self._pull_queue: Optional[torch.Tensor] = None  # limiter overflow
         self._outstanding = 0
+        # fresh mode: answers received at round end, delivered next round
         self._pending_answers: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        # merged mode: retained sent pull segments (answer ids of the
+        # next round's in-order feedback)
+        self._prev_sent_pulls: List[Optional[torch.Tensor]] = \
+            [None] * self.world
+        self._stalled = False
         self._clock = 0
         self._min_clock = 0
 
@@ -171,6 +177,19 @@ class PSEngine:
     # ------------------------------------------------------------------
     def _round(self, it: Iterator[Any], exhausted: bool
                ) -> Tuple[bool, bool]:
+        if self.cfg.feedback == "merged":
+            return self._round_merged(it, exhausted)
+        return self._round_fresh(it, exhausted)
+
+    # ------------------------------------------------------------------
+    def _round_fresh(self, it: Iterator[Any], exhausted: bool
+                     ) -> Tuple[bool, bool]:
+        """feedback="fresh": answers return on a second grouped
+        send/recv within the SAME round (one counts+control all_gather +
+        two p2p phases) — a pull requested by a callback in round r-1
+        ships in r and is delivered in r+1, and pushes ship in the round
+        after the callback that queued them.  Freshest values; the
+        default."""
         m = self.metrics
         m.rounds += 1
 
@@ -295,6 +314,141 @@ class PSEngine:
         active = ctrl[:, 2] == 0
         self._min_clock = int(ctrl[active, 0].min()) if bool(active.any()) \
             else int(ctrl[:, 0].max())
+        done = bool(ctrl[:, 1].all())
+        return exhausted, done
+
+    # ------------------------------------------------------------------
+    def _round_merged(self, it: Iterator[Any], exhausted: bool
+                      ) -> Tuple[bool, bool]:
+        """feedback="merged": ONE merged exchange carries this rank's
+        bucketed pulls/pushes AND the answers its shard produced LAST
+        round — the reference's feedback edge rides the next round's
+        shuffle, so each round costs exactly one counts+control
+        all_gather and one grouped send/recv phase (RCCL: grouped
+        ncclSend/ncclRecv over xGMI).
+
+        Timeline of a pull: requested by a callback in round r-1,
+        shipped in r, served in r, shipped back in r+1, delivered to
+        ``on_pull_recv`` in r+1 — the worker never blocks on the server
+        (the reference's async contract); a push queued by a callback in
+        round r-1 ships and applies in round r.  Staleness comes from
+        this pipeline depth; the SSP knob bounds it on top.
+        """
+        m = self.metrics
+        m.rounds += 1
+
+        # 1. bucket the ops queued by last round's callbacks (unless a
+        # combination-sender strategy holds them — messaging.FlushStrategy)
+        if self.flush is not None and not self.flush.should_flush(
+                self._pending_ops(), force=exhausted or self._stalled):
+            m.held_rounds += 1
+            pull_ids = push_ids = push_deltas = None
+        else:
+            pull_ids = self._take_pulls()
+            push_ids, push_deltas = self._take_pushes()
+        send = self._bucket(pull_ids, push_ids, push_deltas)
+        sent_pulls = send[0]
+        m.pull_ids_sent += int(pull_ids.numel()) if pull_ids is not None else 0
+        m.push_ids_sent += int(push_ids.numel()) if push_ids is not None else 0
+
+        # 2. ONE merged exchange.  Channels: pull ids / push ids / push
+        # deltas / [answer ids when the server may answer out of order] /
+        # answer values; the SSP clock + drain control rides the counts
+        # all_gather.  In-order servers (all bundled ones) never put
+        # answer ids on the wire — the puller retained its sent segments.
+        worker_done = int(exhausted and self._outstanding == 0 and
+                          self._queued_pulls() == 0 and
+                          not self._pull_buf and not self._push_buf and
+                          push_ids is None and
+                          not any(self._answer_buf))
+        dim = self.cfg.dim
+        wire = torch.bfloat16 if self.cfg.wire_dtype == "bf16" \
+            else torch.float32
+        self._wire_out(send[2], wire)
+        ans_vals = [_concat([v for _, v in self._answer_buf[w]])
+                    for w in range(self.world)]
+        self._wire_out(ans_vals, wire)
+        chans = [send[0], send[1], send[2]]
+        specs = [(torch.int64, ()), (torch.int64, ()), (wire, (dim,))]
+        if not self._answers_in_order:
+            chans.append([_concat([a for a, _ in self._answer_buf[w]])
+                          for w in range(self.world)])
+            specs.append((torch.int64, ()))
+        chans.append(ans_vals)
+        specs.append((wire, (dim,)))
+        self._answer_buf = [[] for _ in range(self.world)]
+        m.bytes_sent += self._xgmi_bytes(chans)
+        recv, ctrl = self.transport.alltoallv(
+            chans, specs,
+            extra=[self._clock, worker_done, int(exhausted)])
+        recv = self._to_device(recv)
+        self._unwire(recv[2])
+        self._unwire(recv[-1])
+        r_pull, r_push_ids, r_push_deltas = recv[0], recv[1], recv[2]
+        back_vals = recv[-1]
+        back_ids = self._prev_sent_pulls if self._answers_in_order \
+            else recv[3]
+
+        # 3. deliver the answers — they answer the pulls sent LAST round
+        pend_ids: List[torch.Tensor] = []
+        pend_vals: List[torch.Tensor] = []
+        for s in range(self.world):
+            v = back_vals[s]
+            if v is not None and v.numel():
+                i = back_ids[s]
+                pend_ids.append(i if i.device == v.device else
+                                i.to(v.device))
+                pend_vals.append(v)
+        if pend_ids:
+            ids = _concat(pend_ids)
+            self._outstanding -= int(ids.numel())
+            self.worker_logic.on_pull_recv(ids, _concat(pend_vals),
+                                           self.client)
+        self._prev_sent_pulls = sent_pulls
+
+        # 4. control from THIS exchange (fresh min clock), then consume
+        # data unless SSP-stalled / exhausted / over the pull limit.  An
+        # exhausted worker's clock freezes, so it must not hold the SSP
+        # min-clock back (it would deadlock the still-running workers).
+        active = ctrl[:, 2] == 0
+        self._min_clock = int(ctrl[active, 0].min()) if bool(active.any()) \
+            else int(ctrl[:, 0].max())
+        lag = self._clock - self._min_clock
+        m.staleness_hist[lag] = m.staleness_hist.get(lag, 0) + 1
+        stalled = (self.cfg.staleness is not None and
+                   lag > self.cfg.staleness)
+        self._stalled = stalled
+        backpressure = (self._pull_limit is not None and
+                        self._queued_pulls() >= self._pull_limit)
+        if stalled:
+            m.ssp_stalls += 1
+        if not exhausted and not stalled and not backpressure:
+            try:
+                batch = next(it)
+            except StopIteration:
+                exhausted = True
+            else:
+                # a streaming source may yield None = "no batch ready
+                # this round"; the round proceeds without consuming.
+                if batch is not None:
+                    self._clock += 1
+                    m.samples += _batch_size(self.worker_logic, batch)
+                    self.worker_logic.on_recv(batch, self.client)
+
+        # 5. server phase: pushes first (fresher answers), then pulls;
+        # produced answers are stashed for the NEXT round's exchange
+        for s in range(self.world):
+            ids, deltas = r_push_ids[s], r_push_deltas[s]
+            if ids is not None and ids.numel():
+                self.server_logic.on_push_recv(ids, deltas, self.ps)
+        for s in range(self.world):
+            ids = r_pull[s]
+            if ids is not None and ids.numel():
+                self.server_logic.on_pull_recv(ids, s, self.ps)
+
+        if self.checkpointer is not None:
+            self.checkpointer.maybe(m.rounds)
+
         done = bool(ctrl[:, 1].all())
         return exhausted, done
 

@@ -1,6 +1,7 @@
 """Multi-rank semantics on the in-process loopback MiniCluster (the analog
 of the reference's Flink local-mode tests: real partitioning, real
 worker<->server message cycle, several subtasks in one process)."""
+import pytest
 import torch
 
 from flink_parameter_server_amd import (PSConfig, PSEngine, ShardedTable,
@@ -148,11 +149,13 @@ def test_range_partitioner_across_ranks():
     assert all(v == world for v in merged.values())
 
 
-def test_randomized_world4_stress():
-    """4 ranks, 25 randomized rounds of interleaved pulls/pushes.
-    '+' server semantics are order-independent, so the final table must
-    equal the global per-id push totals, and every pull must have been
-    answered exactly once (in-order feedback edge)."""
+@pytest.mark.parametrize("feedback", ["fresh", "merged"])
+def test_randomized_world4_stress(feedback):
+    """4 ranks, 25 randomized rounds of interleaved pulls/pushes, under
+    both feedback schedules.  '+' server semantics are
+    order-independent, so the final table must equal the global per-id
+    push totals, and every pull must have been answered exactly once
+    (in-order feedback edge)."""
     world, n_ids, dim, rounds = 4, 57, 3, 25
 
     class RandomTraffic(WorkerLogic):
@@ -185,7 +188,8 @@ def test_randomized_world4_stress():
         part = ModPartitioner(n_ids, world)
         table = ShardedTable(part, rank, dim, "cpu")
         logic = RandomTraffic(rank)
-        eng = PSEngine(PSConfig(num_ids=n_ids, dim=dim), logic,
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=dim,
+                                feedback=feedback), logic,
                        TableServerLogic(table), transport, part)
         eng.run([None or r for r in range(rounds)])
         assert all(v == 0 for v in logic.pending.values()), \

@@ -222,3 +222,62 @@ def test_multiworker_singleshard_bucket_padding():
             assert not t  # empty shard dumps nothing
         counts.update(t)
     assert counts == {0: 1.0, 1: 1.0, 4: 1.0, 5: 1.0}
+
+
+def test_merged_feedback_counter_with_codec():
+    """feedback='merged' (answers ride the next round's single exchange)
+    with the entity codec on every segment: same final counts, answers
+    delivered exactly once."""
+
+    def job(rank, transport):
+        wl = CounterLogic()
+        data = [torch.tensor([rank, 2, 5]), torch.tensor([2])]
+        eng = PSEngine(PSConfig(num_ids=8, dim=1, feedback="merged"), wl,
+                       _server(), SerializingTransport(transport))
+        outs = eng.run(iter(data))
+        return wl.answers, outs
+
+    results = run_loopback(2, job)
+    counts = {}
+    for (answers, (_, souts)) in results:
+        assert answers == 4  # 3 + 1 unique ids answered across rounds
+        counts.update(_count_table(souts))
+    assert counts[2] == 4.0 and counts[5] == 2.0
+    assert counts[0] == 1.0 and counts[1] == 1.0
+
+
+def test_merged_feedback_out_of_order_server():
+    """merged + a non-in-order server: answer ids travel on the merged
+    exchange's extra channel (C==5 codec path)."""
+    from flink_parameter_server_amd import ServerLogic
+
+    class ReversedServer(ServerLogic):
+        def __init__(self):
+            self.table = {}
+
+        def on_pull_recv(self, ids, worker, ps):
+            rev = torch.flip(ids, [0])
+            vals = torch.tensor(
+                [[self.table.get(int(i), 0.0)] for i in rev])
+            ps.answer_pull(rev, vals, worker)
+
+        def on_push_recv(self, ids, deltas, ps):
+            for k, i in enumerate(ids.tolist()):
+                self.table[i] = self.table.get(i, 0.0) + float(deltas[k])
+
+        def close(self, ps):
+            for i, v in sorted(self.table.items()):
+                ps.output((i, v))
+
+    def job(rank, transport):
+        wl = CounterLogic()
+        data = [torch.tensor([0, 1, 2, 3]), torch.tensor([1, 2])]
+        eng = PSEngine(PSConfig(num_ids=4, dim=1, feedback="merged"), wl,
+                       ReversedServer(), SerializingTransport(transport))
+        return eng.run(iter(data))
+
+    results = run_loopback(2, job)
+    counts = {}
+    for _, souts in results:
+        counts.update({i: v for i, v in souts})
+    assert counts == {0: 2.0, 1: 4.0, 2: 4.0, 3: 2.0}
