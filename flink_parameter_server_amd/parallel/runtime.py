@@ -9,25 +9,30 @@ quiescence timeout.
 
 MI355X design: one process per GPU; each rank hosts one
 worker role AND one server-shard role.  The free-running record flow
-becomes *rounds*: each round every rank
+becomes *rounds* with two feedback-edge schedules (``PSConfig.feedback``):
 
-  1. delivers last round's pull answers to ``WorkerLogic.on_pull_recv``
-  2. consumes one data batch into ``WorkerLogic.on_recv`` (unless SSP-
-     stalled or out of data)
-  3. coalesces + buckets this round's pulls/pushes by destination shard
-  4. exchanges them (grouped send/recv alltoallv — RCCL over xGMI on GPU)
-  5. runs the server role: apply pushes, gather + answer pulls
-  6. exchanges the answers back (the reference's feedback edge)
-  7. allgathers a tiny control vector (clock, done, in-flight) — this
-     replaces both the SSP clock (new capability) and the reference's
-     quiescence-timeout termination with an explicit all-drained test.
+``"fresh"`` (default) — per round: deliver last round's answers ->
+consume a batch -> bucket this round's pulls/pushes -> exchange them
+(grouped send/recv alltoallv; the SSP clock + drain control rides the
+counts all_gather) -> server phase (apply pushes, gather answers) ->
+exchange the answers back (the reference's feedback edge; in-order
+servers elide the answer ids AND their counts — the puller reconstructs
+both from its own sent segments).  One all_gather + two grouped p2p
+phases per round; a pull is answered with the freshest possible values,
+one round after it was requested.
 
-Asynchrony: an answer requested in round r arrives at the start of round
-r+1, and the push it provokes is applied in round r+1 — the worker never
-blocks on the server, which is the reference's async pull/push contract;
-staleness comes from round depth.  ``pull_limit`` bounds distinct
+``"merged"`` — the answers ride the NEXT round's single exchange, so a
+round is exactly one counts+control all_gather + one grouped p2p phase.
++1 round of answer latency / staleness; the throughput option for large
+multi-GPU jobs where per-round collective latency is the scaling tax.
+
+Asynchrony (both modes): the worker never blocks on the server — the
+reference's async pull/push contract; staleness comes from round/pipeline
+depth (the SSP knob bounds it on top).  ``pull_limit`` bounds distinct
 outstanding pull ids exactly like ``WorkerLogic.addPullLimiter`` (excess
-pulls queue locally).
+pulls queue locally).  Termination is an explicit all-drained consensus
+carried by the control lane (replacing the reference's
+``iterationWaitTime`` quiescence timeout).
 """
 from __future__ import annotations
 
