@@ -491,53 +491,7 @@ __global__ __launch_bounds__(256) void k_seg_fused(
 
 #pragma unroll
     for (int e = 0; e < EPL; ++e) acc[e] = make_float4(0, 0, 0, 0);
-
-    // fast path: the whole chunk is one segment (the common case for
-    // zipf-headed key distributions, where segments are much longer
-    // than a chunk) — no per-sample key compare, count known up front,
-    // ownership decided by peeking the chunk's neighbors.
-    if (rec[end - 1].x == cur) {
-      load_kv();
-      left_owned = (start == 0) || (rec[start - 1].x != cur);
-      const bool right_owned = (end == n) || (rec[end].x != cur);
-      for (int64_t j = start; j < end; ++j) {
-        const int4 rj = rec[j];
-        const int64_t sr = (int64_t)rj.y * dim4;
-        float4 sv[EPL];
-#pragma unroll
-        for (int e = 0; e < EPL; ++e) {
-          int c = lane + e * GP;
-          sv[e] = (c < dim4) ? other[sr + c] : make_float4(0, 0, 0, 0);
-        }
-        float g;
-        if (MODE == 2) {
-          g = __int_as_float(rj.z);
-        } else {
-          float dot = 0.f;
-#pragma unroll
-          for (int e = 0; e < EPL; ++e)
-            dot += kv[e].x * sv[e].x + kv[e].y * sv[e].y +
-                   kv[e].z * sv[e].z + kv[e].w * sv[e].w;
-          dot = group_sum<GP>(dot);
-          g = (MODE == 0)
-                  ? (__int_as_float(rj.z) - dot)
-                  : lrc * (__int_as_float(rj.z) -
-                           1.0f / (1.0f + __expf(-dot)));
-          if (lane == 0) rec[j].z = __float_as_int(g);
-        }
-#pragma unroll
-        for (int e = 0; e < EPL; ++e) {
-          acc[e].x += g * sv[e].x; acc[e].y += g * sv[e].y;
-          acc[e].z += g * sv[e].z; acc[e].w += g * sv[e].w;
-        }
-      }
-      cnt = (int)(end - start);
-      flush(right_owned);
-      continue;
-    }
-
     load_kv();
-    if (start > 0) left_owned = (rec[start - 1].x != cur);
     for (int64_t j = start; j < end; ++j) {
       const int4 rj = rec[j];
       if (rj.x != cur) {
@@ -577,7 +531,7 @@ __global__ __launch_bounds__(256) void k_seg_fused(
       }
       ++cnt;
     }
-    flush(end == n || rec[end].x != cur);  // peek: segment ends here?
+    flush(end == n);  // right-bounded only at the array end
   }
 }
 
