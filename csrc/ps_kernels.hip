@@ -116,6 +116,24 @@ __global__ __launch_bounds__(256) void k_gather_vec4(
   }
 }
 
+// streaming variant: working set >> L2, so skip cache retention on the
+// row traffic (guide: nt on streamed data one CU reads once).  The
+// nontemporal builtins want clang vector types, not HIP_vector_type.
+typedef float v4f __attribute__((ext_vector_type(4)));
+
+__global__ __launch_bounds__(256) void k_gather_vec4_nt(
+    const float4* __restrict__ table, const int64_t* __restrict__ rows,
+    float4* __restrict__ out, int64_t n, int64_t dim4) {
+  int64_t total = n * dim4;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = i / dim4, c = i - j * dim4;
+    v4f v = __builtin_nontemporal_load(
+        (const v4f*)&table[rows[j] * dim4 + c]);
+    __builtin_nontemporal_store(v, (v4f*)&out[i]);
+  }
+}
+
 __global__ __launch_bounds__(256) void k_gather(
     const float* __restrict__ table, const int64_t* __restrict__ rows,
     float* __restrict__ out, int64_t n, int64_t dim) {
@@ -157,6 +175,21 @@ __global__ __launch_bounds__(256) void k_apply_add_unique_vec4(
     float4 a = table[t], d = deltas[i];
     a.x += d.x; a.y += d.y; a.z += d.z; a.w += d.w;
     table[t] = a;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_apply_add_unique_vec4_nt(
+    float4* __restrict__ table, const int64_t* __restrict__ rows,
+    const float4* __restrict__ deltas, int64_t n, int64_t dim4) {
+  int64_t total = n * dim4;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = i / dim4, c = i - j * dim4;
+    int64_t t = rows[j] * dim4 + c;
+    v4f a = __builtin_nontemporal_load((const v4f*)&table[t]);
+    v4f d = __builtin_nontemporal_load((const v4f*)&deltas[i]);
+    a += d;
+    __builtin_nontemporal_store(a, (v4f*)&table[t]);
   }
 }
 
@@ -658,7 +691,10 @@ torch::Tensor gather(torch::Tensor table, torch::Tensor rows) {
   if (!n) return out;
   if (dim % 4 == 0) {
     int64_t dim4 = dim / 4;
-    hipLaunchKernelGGL(k_gather_vec4, dim3(grid_for(n * dim4, 256)),
+    // streamed traffic far beyond L2: nontemporal variant
+    bool streaming = n * dim * 4 > (32ll << 20);
+    auto* kfn = streaming ? k_gather_vec4_nt : k_gather_vec4;
+    hipLaunchKernelGGL(kfn, dim3(grid_for(n * dim4, 256)),
                        dim3(256), 0, cur_stream(),
                        (const float4*)table.data_ptr<float>(),
                        rows.data_ptr<int64_t>(),
@@ -681,8 +717,10 @@ void apply_add(torch::Tensor table, torch::Tensor rows,
   if (!n) return;
   if (unique && dim % 4 == 0) {
     int64_t dim4 = dim / 4;
-    hipLaunchKernelGGL(k_apply_add_unique_vec4,
-                       dim3(grid_for(n * dim4, 256)), dim3(256), 0,
+    bool streaming = n * dim * 4 > (32ll << 20);
+    auto* kfn = streaming ? k_apply_add_unique_vec4_nt
+                          : k_apply_add_unique_vec4;
+    hipLaunchKernelGGL(kfn, dim3(grid_for(n * dim4, 256)), dim3(256), 0,
                        cur_stream(), (float4*)table.data_ptr<float>(),
                        rows.data_ptr<int64_t>(),
                        (const float4*)deltas.data_ptr<float>(), n, dim4);
