@@ -33,7 +33,7 @@ Parameter-mapping notes:
 """
 from __future__ import annotations
 
-from typing import Any, Iterable, List, Optional, Sequence, Tuple, Union
+from typing import Any, Iterable, List, Optional, Tuple
 
 import torch
 

@@ -26,7 +26,7 @@ from __future__ import annotations
 
 import collections
 from abc import ABC, abstractmethod
-from typing import Any, Callable, Generic, List, Optional, Tuple, TypeVar
+from typing import Any, Callable, Optional, TypeVar
 
 import torch
 
@@ -50,11 +50,15 @@ class ParameterServerClient(ABC):
         """Request current values for ``ids`` (async, answered later).
         ``unique=True`` asserts the caller already de-duplicated ``ids``
         (sorted or not), letting the engine skip its per-round coalescing
-        pass when this is the round's only pull."""
+        pass when this is the round's only pull.  The tensor must not be
+        mutated after the call (async send; re-sending one cached id
+        buffer every round is encouraged — the engine memoizes its
+        shard bucketing by buffer identity)."""
 
     @abstractmethod
     def push(self, ids: torch.Tensor, deltas: torch.Tensor) -> None:
-        """Send delta updates for ``ids`` (async, no reply)."""
+        """Send delta updates for ``ids`` (async, no reply; the tensors
+        must not be mutated after the call)."""
 
     @abstractmethod
     def output(self, out: Any) -> None:

@@ -13,7 +13,7 @@ HIP kernel K8) and pushes deltas for center and context rows.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 

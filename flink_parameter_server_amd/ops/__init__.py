@@ -7,7 +7,7 @@ in-tree extension; a missing extension raises (never a silent fallback).
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

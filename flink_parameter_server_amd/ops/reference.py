@@ -21,7 +21,7 @@ cites are file/symbol level):
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

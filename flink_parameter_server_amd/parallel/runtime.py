@@ -147,6 +147,7 @@ class PSEngine:
         self._prev_sent_pulls: List[Optional[torch.Tensor]] = \
             [None] * self.world
         self._stalled = False
+        self._bucket_cache: Dict[Tuple[int, int], Any] = {}
         self._clock = 0
         self._min_clock = 0
 
@@ -583,11 +584,26 @@ class PSEngine:
     def _shard_counts(self, ids: torch.Tensor):
         """bucket_by_shard with counts padded to world size (the
         partitioner may define fewer shards than ranks when
-        ps_parallelism < worker_parallelism — algorithms.py)."""
+        ps_parallelism < worker_parallelism — algorithms.py).
+
+        Memoized by buffer identity: steady-state workers re-send one
+        cached id tensor every round (dense-pull MF sends the same
+        arange; pushes reuse the pulled uids), so the argsort/bincount
+        bucketing runs once, not every round.  Held references keep the
+        cached storages alive, so a pointer can never alias a different
+        live tensor."""
+        key = (ids.data_ptr(), ids.numel())
+        hit = self._bucket_cache.get(key)
+        if hit is not None:
+            _, order, counts = hit
+            return order, counts, None
         order, counts, shards = bucket_by_shard(ids, self.part)
         if counts.numel() < self.world:
             counts = torch.cat([counts, counts.new_zeros(
                 self.world - counts.numel())])
+        if len(self._bucket_cache) >= 8:
+            self._bucket_cache.clear()
+        self._bucket_cache[key] = (ids, order, counts)
         return order, counts, shards
 
 

@@ -11,7 +11,7 @@ of the id), updated by HIP scatter-apply kernels.
 """
 from __future__ import annotations
 
-from typing import Callable, Optional, Tuple
+from typing import Tuple
 
 import torch
 
