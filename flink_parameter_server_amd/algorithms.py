@@ -118,6 +118,7 @@ def _ps_mf(ratings, n_users, n_items, num_factors, range_min, range_max,
             reg=reg, range_min=range_min, range_max=range_max,
             negative_sample_rate=negative_sample_rate, n_items=n_items,
             pull_limit=pull_limit, output_user_factors=output_user_factors,
+            output_user_factors_on_close=True,
             seed=seed + rank)
         workers[rank] = wl
         return wl

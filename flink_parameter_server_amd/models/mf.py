@@ -83,6 +83,7 @@ class OnlineMFWorkerLogic(WorkerLogic):
                  device: torch.device = torch.device("cpu"),
                  pull_limit: Optional[int] = None,
                  output_user_factors: bool = False,
+                 output_user_factors_on_close: bool = False,
                  seed: int = 1):
         self.n_factors = int(n_factors)
         self.lr = float(learning_rate)
@@ -92,6 +93,8 @@ class OnlineMFWorkerLogic(WorkerLogic):
         self.device = torch.device(device)
         self.pull_limit = pull_limit
         self.output_user_factors = output_user_factors
+        self.output_on_close = (output_user_factors or
+                                output_user_factors_on_close)
         self._neg_counter = 0
         self._seed = seed
         # Worker-local user table (reference: per-subtask user map with
@@ -199,7 +202,7 @@ class OnlineMFWorkerLogic(WorkerLogic):
                                    device=self.device)]))
 
     def close(self, client: ParameterServerClient) -> None:
-        if self.output_user_factors:
+        if self.output_on_close:
             rows = torch.arange(self.user_table.shape[0],
                                 dtype=torch.int64, device=self.device)
             client.output((rows, self.user_table.clone()))
