@@ -901,7 +901,15 @@ static void by_group4(int64_t dim4, F f) {
       f(integral_constant<int, 8>{}, integral_constant<int, 1>{});
   }
   else if (dim4 <= 16) f(integral_constant<int, 8>{}, integral_constant<int, 2>{});
-  else if (dim4 <= 32) f(integral_constant<int, 8>{}, integral_constant<int, 4>{});
+  else if (dim4 <= 32) {
+    // FPS_GP_D32=16: half the per-lane elements, twice the lanes per
+    // sample (fewer registers, one more shfl) — sweepable on GPU
+    static int gp32 = env_int("FPS_GP_D32", 8);
+    if (gp32 == 16)
+      f(integral_constant<int, 16>{}, integral_constant<int, 2>{});
+    else
+      f(integral_constant<int, 8>{}, integral_constant<int, 4>{});
+  }
   else if (dim4 <= 48) f(integral_constant<int, 16>{}, integral_constant<int, 3>{});
   else if (dim4 <= 80) f(integral_constant<int, 16>{}, integral_constant<int, 5>{});
   else { TORCH_CHECK(dim4 <= 128, "row dim > 512 unsupported");
