@@ -820,7 +820,7 @@ static torch::Tensor build_rec_t(torch::Tensor keys, torch::Tensor slots,
   if (m > 0 && m <= PACK_MAX_M) {
     if (!hist.defined()) {
       hist = torch::zeros({m}, opts_i32);
-      int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+      int hg = (int)std::min<int64_t>((n + 1023) / 1024, 1024);
       hipLaunchKernelGGL((k_key_hist<TK>), dim3(std::max(hg, 1)),
                          dim3(256), (size_t)m * sizeof(int), cur_stream(),
                          keys.data_ptr<TK>(), hist.data_ptr<int>(), n,
@@ -904,7 +904,7 @@ static void by_group4(int64_t dim4, F f) {
   else if (dim4 <= 32) {
     // FPS_GP_D32=16: half the per-lane elements, twice the lanes per
     // sample (fewer registers, one more shfl) — sweepable on GPU
-    static int gp32 = env_int("FPS_GP_D32", 8);
+    static int gp32 = env_int("FPS_GP_D32", 16);
     if (gp32 == 16)
       f(integral_constant<int, 16>{}, integral_constant<int, 2>{});
     else
@@ -924,7 +924,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                       const torch::Tensor& keyed,
                       const torch::Tensor& other, int64_t dim4,
                       double scale, double a, double lrc,
-                      int64_t rep_stride = 0) {
+                      int64_t rep_stride = 0, int rep_mask = 7) {
   int64_t n = rec.size(0);
   if (!n) return;
   int chunk = seg_chunk();
@@ -939,14 +939,17 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                        (const float4*)keyed.data_ptr<float>(),
                        (const float4*)other.data_ptr<float>(), n, dim4,
                        (float)scale, (float)a, (float)lrc, chunk,
-                       rep_stride / 4);
+                       rep_stride / 4, rep_mask);
   });
 }
 
-// replica count heuristic for the hot-key delta flush
+// replica count heuristic for the hot-key delta flush (power of two,
+// env-sweepable)
 static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
-  if (n < 8 * m) return 1;
-  int64_t r = 8;
+  static int64_t r_env = env_int("FPS_DELTA_REPS", 8);
+  int64_t r = r_env;
+  if (r < 1) r = 1;
+  if (n < r * m) return 1;
   if (r * m * dim * 4 > (64 << 20)) return 1;  // cap replica buffer 64 MB
   return r;
 }
@@ -987,7 +990,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     auto hist_both = torch::zeros({m + U}, opts_i32);  // one fill launch
     histA = hist_both.narrow(0, 0, m);
     histB = hist_both.narrow(0, m, U);
-    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+    int hg = (int)std::min<int64_t>((n + 1023) / 1024, 1024);
     size_t lds = (size_t)(m + U) * sizeof(int);
     if (item_slots.dtype() == torch::kInt32)
       hipLaunchKernelGGL((k_dual_hist<int>), dim3(std::max(hg, 1)),
@@ -1011,7 +1014,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   if (reps > 1) {
     auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
     seg_fused<0, true, true>(rep_buf, recA, item_vecs, user_table, dim4,
-                             lr, -lr * reg, 0.0, m * dim);
+                             lr, -lr * reg, 0.0, m * dim, (int)reps - 1);
     deltas = rep_buf.view({reps, m, dim}).sum(0);
   } else {
     deltas = torch::zeros_like(item_vecs);
@@ -1024,7 +1027,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   else if (!histB.defined()) {
     auto opts_i32 = torch::dtype(torch::kInt32).device(ratings.device());
     histB = torch::zeros({U}, opts_i32);
-    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+    int hg = (int)std::min<int64_t>((n + 1023) / 1024, 1024);
     if (user_rows.dtype() == torch::kInt32)
       hipLaunchKernelGGL((k_key_hist<int>), dim3(std::max(hg, 1)),
                          dim3(256), (size_t)U * sizeof(int), cur_stream(),
@@ -1066,7 +1069,7 @@ torch::Tensor sgns_step(torch::Tensor vecs, torch::Tensor center_slots,
   if (m <= PACK_MAX_M) {
     auto opts_i32 = torch::dtype(torch::kInt32).device(vecs.device());
     histD = torch::zeros({m}, opts_i32);
-    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+    int hg = (int)std::min<int64_t>((n + 1023) / 1024, 1024);
     if (ctx_slots.dtype() == torch::kInt32)
       hipLaunchKernelGGL((k_key_hist<int>), dim3(std::max(hg, 1)),
                          dim3(256), (size_t)m * sizeof(int), cur_stream(),
