@@ -461,8 +461,8 @@ __global__ __launch_bounds__(256) void k_seg_fused(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride) {
-  const int64_t rep_off = (int64_t)(blockIdx.x & 7) * rep_stride;
+    int64_t rep_stride, int rep_mask) {
+  const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
   const int64_t gpb = blockDim.x / GP;
   const int64_t g0 = (int64_t)blockIdx.x * gpb + threadIdx.x / GP;
@@ -820,7 +820,7 @@ static torch::Tensor build_rec_t(torch::Tensor keys, torch::Tensor slots,
   if (m > 0 && m <= PACK_MAX_M) {
     if (!hist.defined()) {
       hist = torch::zeros({m}, opts_i32);
-      int hg = (int)std::min<int64_t>((n + 1023) / 1024, 1024);
+      int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
       hipLaunchKernelGGL((k_key_hist<TK>), dim3(std::max(hg, 1)),
                          dim3(256), (size_t)m * sizeof(int), cur_stream(),
                          keys.data_ptr<TK>(), hist.data_ptr<int>(), n,
@@ -990,7 +990,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     auto hist_both = torch::zeros({m + U}, opts_i32);  // one fill launch
     histA = hist_both.narrow(0, 0, m);
     histB = hist_both.narrow(0, m, U);
-    int hg = (int)std::min<int64_t>((n + 1023) / 1024, 1024);
+    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
     size_t lds = (size_t)(m + U) * sizeof(int);
     if (item_slots.dtype() == torch::kInt32)
       hipLaunchKernelGGL((k_dual_hist<int>), dim3(std::max(hg, 1)),
@@ -1027,7 +1027,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   else if (!histB.defined()) {
     auto opts_i32 = torch::dtype(torch::kInt32).device(ratings.device());
     histB = torch::zeros({U}, opts_i32);
-    int hg = (int)std::min<int64_t>((n + 1023) / 1024, 1024);
+    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
     if (user_rows.dtype() == torch::kInt32)
       hipLaunchKernelGGL((k_key_hist<int>), dim3(std::max(hg, 1)),
                          dim3(256), (size_t)U * sizeof(int), cur_stream(),
@@ -1069,7 +1069,7 @@ torch::Tensor sgns_step(torch::Tensor vecs, torch::Tensor center_slots,
   if (m <= PACK_MAX_M) {
     auto opts_i32 = torch::dtype(torch::kInt32).device(vecs.device());
     histD = torch::zeros({m}, opts_i32);
-    int hg = (int)std::min<int64_t>((n + 1023) / 1024, 1024);
+    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
     if (ctx_slots.dtype() == torch::kInt32)
       hipLaunchKernelGGL((k_key_hist<int>), dim3(std::max(hg, 1)),
                          dim3(256), (size_t)m * sizeof(int), cur_stream(),
