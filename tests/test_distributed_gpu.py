@@ -13,7 +13,7 @@ import torch.multiprocessing as mp
 pytestmark = pytest.mark.gpu
 
 
-def _job(rank, world, port, outdir):
+def _job(rank, world, port, outdir, feedback):
     import torch.distributed as dist
     dist.init_process_group("gloo", init_method=f"tcp://127.0.0.1:{port}",
                             rank=rank, world_size=world)
@@ -42,7 +42,8 @@ def _job(rank, world, port, outdir):
                                  range_min=-0.1, range_max=0.1,
                                  n_items=n_items, device=dev)
         transport = TorchDistTransport(device=torch.device("cpu"))
-        eng = PSEngine(PSConfig(num_ids=n_items, dim=rank_k), wl,
+        eng = PSEngine(PSConfig(num_ids=n_items, dim=rank_k,
+                                feedback=feedback), wl,
                        TableServerLogic(table, dump_on_close=False,
                                         pushes_unique=True),
                        transport, part, device=dev)
@@ -75,10 +76,13 @@ def _job(rank, world, port, outdir):
 
 
 @pytest.mark.timeout(300)
-def test_two_rank_mf_on_gpu_converges():
-    world, port = 2, 29517
+@pytest.mark.parametrize("feedback,port", [("fresh", 29517),
+                                           ("merged", 29518)])
+def test_two_rank_mf_on_gpu_converges(feedback, port):
+    world = 2
     with tempfile.TemporaryDirectory() as d:
-        mp.spawn(_job, args=(world, port, d), nprocs=world, join=True)
+        mp.spawn(_job, args=(world, port, d, feedback), nprocs=world,
+                 join=True)
         for r in range(world):
             res = torch.load(os.path.join(d, f"rank{r}.pt"),
                              weights_only=False)
