@@ -57,9 +57,12 @@ def make_mf(args, rank, world, device):
     users_local = (n_users + world - 1) // world
     part = ModPartitioner(n_items, world)
     table = ShardedTable(part, rank, k, device, init=(-0.01, 0.01))
+    # epoch-resweep pattern: the same resident batches cycle, so the
+    # fused kernel may cache its sort positions per batch
     wl = OnlineMFWorkerLogic(users_local, k, learning_rate=0.01,
                              reg=0.001, range_min=-0.01, range_max=0.01,
-                             n_items=n_items, device=device, seed=rank + 1)
+                             n_items=n_items, device=device,
+                             assume_static_batches=True, seed=rank + 1)
     g = torch.Generator().manual_seed(1234 + rank)
     batches = []
     for _ in range(args.n_batches):

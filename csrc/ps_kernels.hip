@@ -40,6 +40,7 @@
 #include <ATen/hip/HIPContext.h>
 
 #include <cstdlib>
+#include <unordered_map>
 
 #include <rocprim/rocprim.hpp>
 
@@ -321,7 +322,8 @@ template <class TK>
 __global__ __launch_bounds__(256) void k_pack_rec(
     const TK* __restrict__ keys, const TK* __restrict__ slots,
     const float* __restrict__ vals, int* __restrict__ gcursor,
-    int4* __restrict__ rec, int64_t n, int m, int chunk) {
+    int4* __restrict__ rec, int* __restrict__ pos_out, int64_t n, int m,
+    int chunk) {
   extern __shared__ int lds[];
   int* lhist = lds;
   int* lbase = lds + m;
@@ -344,8 +346,9 @@ __global__ __launch_bounds__(256) void k_pack_rec(
     for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x) {
       const int k = (int)keys[j];
       const int lpos = atomicAdd(&lhist[k], 1);
-      rec[(int64_t)lbase[k] + lpos] =
-          make_int4(k, (int)slots[j], __float_as_int(vals[j]), 0);
+      const int64_t pos = (int64_t)lbase[k] + lpos;
+      rec[pos] = make_int4(k, (int)slots[j], __float_as_int(vals[j]), 0);
+      if (pos_out) pos_out[j] = (int)pos;
     }
     __syncthreads();
   }
@@ -355,7 +358,8 @@ __global__ __launch_bounds__(256) void k_pack_rec(
 // swap, val carried through — by now val holds the pass-A coefficient)
 __global__ __launch_bounds__(256) void k_pack_rec_swap(
     const int4* __restrict__ src, int* __restrict__ gcursor,
-    int4* __restrict__ rec, int64_t n, int m, int chunk) {
+    int4* __restrict__ rec, int* __restrict__ pos_out, int64_t n, int m,
+    int chunk) {
   extern __shared__ int lds[];
   int* lhist = lds;
   int* lbase = lds + m;
@@ -378,7 +382,9 @@ __global__ __launch_bounds__(256) void k_pack_rec_swap(
     for (int64_t j = start + threadIdx.x; j < end; j += blockDim.x) {
       const int4 s = src[j];
       const int lpos = atomicAdd(&lhist[s.y], 1);
-      rec[(int64_t)lbase[s.y] + lpos] = make_int4(s.y, s.x, s.z, 0);
+      const int64_t pos = (int64_t)lbase[s.y] + lpos;
+      rec[pos] = make_int4(s.y, s.x, s.z, 0);
+      if (pos_out) pos_out[j] = (int)pos;
     }
     __syncthreads();
   }
@@ -410,6 +416,39 @@ __global__ __launch_bounds__(256) void k_interleave_swap(
     okeys[j] = s.y;
     rec[j] = make_int4(s.y, s.x, s.z, 0);
   }
+}
+
+// cached-position builders: rec[pos[j]] = record j (one scatter pass —
+// no histogram, no scan, no cursor atomics).  Valid only while the
+// caller guarantees the key/slot tensors' CONTENT is unchanged (the
+// epoch-resweep contract; see mf_sgd_step reuse_tag).
+template <class TK>
+__global__ __launch_bounds__(256) void k_apply_pos(
+    const TK* __restrict__ keys, const TK* __restrict__ slots,
+    const float* __restrict__ vals, const int* __restrict__ pos,
+    int4* __restrict__ rec, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x)
+    rec[pos[j]] = make_int4((int)keys[j], (int)slots[j],
+                            __float_as_int(vals[j]), 0);
+}
+
+__global__ __launch_bounds__(256) void k_apply_pos_swap(
+    const int4* __restrict__ src, const int* __restrict__ pos,
+    int4* __restrict__ rec, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    const int4 s = src[j];
+    rec[pos[j]] = make_int4(s.y, s.x, s.z, 0);
+  }
+}
+
+// pos[order[j]] = j  (inverse of a sort permutation)
+__global__ __launch_bounds__(256) void k_invert_order(
+    const int64_t* __restrict__ order, int* __restrict__ pos, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x)
+    pos[order[j]] = (int)j;
 }
 
 static int bits_for(int64_t m) {
@@ -809,14 +848,19 @@ static torch::Tensor exclusive_starts(torch::Tensor hist) {
   return starts;
 }
 
-// recs sorted by `keys`; hist may be precomputed (else computed here)
+// recs sorted by `keys`; hist may be precomputed (else computed here).
+// When `pos_out` is given (cacheable first pass), the final position of
+// every input record is also emitted; the radix path then goes through
+// argsort + gather so positions exist.
 template <class TK>
 static torch::Tensor build_rec_t(torch::Tensor keys, torch::Tensor slots,
                                  torch::Tensor vals, int64_t m,
-                                 torch::Tensor hist) {
+                                 torch::Tensor hist,
+                                 torch::Tensor pos_out) {
   int64_t n = keys.numel();
   auto opts_i32 = torch::dtype(torch::kInt32).device(keys.device());
   auto rec = torch::empty({n, 4}, opts_i32);
+  int* pos = pos_out.defined() ? pos_out.data_ptr<int>() : nullptr;
   if (m > 0 && m <= PACK_MAX_M) {
     if (!hist.defined()) {
       hist = torch::zeros({m}, opts_i32);
@@ -834,7 +878,17 @@ static torch::Tensor build_rec_t(torch::Tensor keys, torch::Tensor slots,
                        2 * (size_t)m * sizeof(int), cur_stream(),
                        keys.data_ptr<TK>(), slots.data_ptr<TK>(),
                        vals.data_ptr<float>(), cursor.data_ptr<int>(),
-                       (int4*)rec.data_ptr<int>(), n, (int)m, chunk);
+                       (int4*)rec.data_ptr<int>(), pos, n, (int)m, chunk);
+  } else if (pos) {
+    auto keys32 = keys.dtype() == torch::kInt32
+                      ? keys : keys.to(torch::kInt32);
+    auto order = torch::argsort(keys32);
+    hipLaunchKernelGGL(k_invert_order, dim3(grid_for(n, 256)), dim3(256),
+                       0, cur_stream(), order.data_ptr<int64_t>(), pos, n);
+    hipLaunchKernelGGL((k_apply_pos<TK>), dim3(grid_for(n, 256)),
+                       dim3(256), 0, cur_stream(), keys.data_ptr<TK>(),
+                       slots.data_ptr<TK>(), vals.data_ptr<float>(), pos,
+                       (int4*)rec.data_ptr<int>(), n);
   } else {
     auto keys32 = torch::empty({n}, opts_i32);
     hipLaunchKernelGGL((k_interleave_rec<TK>), dim3(grid_for(n, 256)),
@@ -849,18 +903,21 @@ static torch::Tensor build_rec_t(torch::Tensor keys, torch::Tensor slots,
 
 static torch::Tensor build_rec(torch::Tensor keys, torch::Tensor slots,
                                torch::Tensor vals, int64_t m,
-                               torch::Tensor hist = {}) {
+                               torch::Tensor hist = {},
+                               torch::Tensor pos_out = {}) {
   if (keys.dtype() == torch::kInt32)
-    return build_rec_t<int>(keys, slots, vals, m, hist);
-  return build_rec_t<int64_t>(keys, slots, vals, m, hist);
+    return build_rec_t<int>(keys, slots, vals, m, hist, pos_out);
+  return build_rec_t<int64_t>(keys, slots, vals, m, hist, pos_out);
 }
 
 // recs re-keyed by the slot field of an existing rec array
 static torch::Tensor build_rec_swap(torch::Tensor src, int64_t m,
-                                    torch::Tensor hist = {}) {
+                                    torch::Tensor hist = {},
+                                    torch::Tensor pos_out = {}) {
   int64_t n = src.size(0);
   auto opts_i32 = torch::dtype(torch::kInt32).device(src.device());
   auto rec = torch::empty({n, 4}, opts_i32);
+  int* pos = pos_out.defined() ? pos_out.data_ptr<int>() : nullptr;
   if (m > 0 && m <= PACK_MAX_M) {
     TORCH_CHECK(hist.defined(), "build_rec_swap needs a histogram");
     auto cursor = exclusive_starts(hist).contiguous();
@@ -871,7 +928,7 @@ static torch::Tensor build_rec_swap(torch::Tensor src, int64_t m,
                        2 * (size_t)m * sizeof(int), cur_stream(),
                        (const int4*)src.data_ptr<int>(),
                        cursor.data_ptr<int>(), (int4*)rec.data_ptr<int>(),
-                       n, (int)m, chunk);
+                       pos, n, (int)m, chunk);
   } else {
     auto keys32 = torch::empty({n}, opts_i32);
     hipLaunchKernelGGL(k_interleave_swap, dim3(grid_for(n, 256)),
@@ -879,6 +936,18 @@ static torch::Tensor build_rec_swap(torch::Tensor src, int64_t m,
                        (const int4*)src.data_ptr<int>(),
                        keys32.data_ptr<int>(), (int4*)rec.data_ptr<int>(),
                        n);
+    if (pos) {
+      auto order = torch::argsort(keys32);
+      hipLaunchKernelGGL(k_invert_order, dim3(grid_for(n, 256)),
+                         dim3(256), 0, cur_stream(),
+                         order.data_ptr<int64_t>(), pos, n);
+      auto rec2 = torch::empty({n, 4}, opts_i32);
+      hipLaunchKernelGGL(k_apply_pos_swap, dim3(grid_for(n, 256)),
+                         dim3(256), 0, cur_stream(),
+                         (const int4*)src.data_ptr<int>(), pos,
+                         (int4*)rec2.data_ptr<int>(), n);
+      return rec2;
+    }
     rec = radix_sort_recs(keys32, rec, n, m);
   }
   return rec;
@@ -961,9 +1030,30 @@ static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
 //     lr*(e*u0 - reg*v) segment-accumulated; e persisted into rec.val
 //   pass B (recs by user): user updates lr*(e*v - reg*u0) accumulated
 //     into user_table in place
+// position cache for epoch-resweep workloads (reuse_tag != 0): the
+// caller promises the key/slot tensors' CONTENT is unchanged whenever
+// the same (tag, buffers) triple recurs, so the sort positions from the
+// first visit replace the histogram+scan+counting-sort machinery with
+// one scatter pass per side.  Entries hold references to the key
+// tensors, so a pointer can never alias a different live tensor.
+struct PosCacheEntry {
+  torch::Tensor keys, slots, posA, posB;
+};
+static std::unordered_map<uint64_t, PosCacheEntry> g_pos_cache;
+
+static uint64_t pos_cache_key(int64_t tag, const torch::Tensor& a,
+                              const torch::Tensor& b, int64_t n) {
+  uint64_t h = (uint64_t)tag * 0x9E3779B97F4A7C15ull;
+  h ^= (uint64_t)a.data_ptr() + 0x9E3779B97F4A7C15ull + (h << 6);
+  h ^= (uint64_t)b.data_ptr() + 0x9E3779B97F4A7C15ull + (h << 6);
+  h ^= (uint64_t)n + (h >> 2);
+  return h;
+}
+
 torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                           torch::Tensor item_vecs, torch::Tensor item_slots,
-                          torch::Tensor ratings, double lr, double reg) {
+                          torch::Tensor ratings, double lr, double reg,
+                          int64_t reuse_tag) {
   CHECK_DEV(user_table); CHECK_CONT(user_table);
   CHECK_DEV(item_vecs); CHECK_CONT(item_vecs);
   CHECK_CONT(user_rows); CHECK_CONT(item_slots); CHECK_CONT(ratings);
@@ -981,34 +1071,85 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       user_rows = user_rows.to(torch::kInt32);
   }
 
-  // histograms: one fused read when both key ranges are LDS-sized
-  torch::Tensor histA, histB;
-  bool smallA = m <= PACK_MAX_M, smallB = U <= PACK_MAX_M;
-  if (smallA && smallB && m + U <= 2 * PACK_MAX_M &&
-      item_slots.dtype() == user_rows.dtype()) {
-    auto opts_i32 = torch::dtype(torch::kInt32).device(ratings.device());
-    auto hist_both = torch::zeros({m + U}, opts_i32);  // one fill launch
-    histA = hist_both.narrow(0, 0, m);
-    histB = hist_both.narrow(0, m, U);
-    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
-    size_t lds = (size_t)(m + U) * sizeof(int);
+  auto opts_i32 = torch::dtype(torch::kInt32).device(ratings.device());
+
+  // cached-position fast path (epoch resweeps): one scatter per side
+  // replaces histogram + scan + counting sort
+  PosCacheEntry* hit = nullptr;
+  uint64_t ck = 0;
+  if (reuse_tag) {
+    ck = pos_cache_key(reuse_tag, item_slots, user_rows, n);
+    auto it = g_pos_cache.find(ck);
+    if (it != g_pos_cache.end()) hit = &it->second;
+  }
+
+  torch::Tensor recA;
+  torch::Tensor posA, posB;
+  torch::Tensor histB;
+  if (hit) {
+    recA = torch::empty({n, 4}, opts_i32);
     if (item_slots.dtype() == torch::kInt32)
-      hipLaunchKernelGGL((k_dual_hist<int>), dim3(std::max(hg, 1)),
-                         dim3(256), lds, cur_stream(),
+      hipLaunchKernelGGL((k_apply_pos<int>), dim3(grid_for(n, 256)),
+                         dim3(256), 0, cur_stream(),
                          item_slots.data_ptr<int>(),
-                         user_rows.data_ptr<int>(), histA.data_ptr<int>(),
-                         histB.data_ptr<int>(), n, (int)m, (int)U);
+                         user_rows.data_ptr<int>(),
+                         ratings.data_ptr<float>(),
+                         hit->posA.data_ptr<int>(),
+                         (int4*)recA.data_ptr<int>(), n);
     else
-      hipLaunchKernelGGL((k_dual_hist<int64_t>), dim3(std::max(hg, 1)),
-                         dim3(256), lds, cur_stream(),
+      hipLaunchKernelGGL((k_apply_pos<int64_t>), dim3(grid_for(n, 256)),
+                         dim3(256), 0, cur_stream(),
                          item_slots.data_ptr<int64_t>(),
                          user_rows.data_ptr<int64_t>(),
-                         histA.data_ptr<int>(), histB.data_ptr<int>(), n,
-                         (int)m, (int)U);
+                         ratings.data_ptr<float>(),
+                         hit->posA.data_ptr<int>(),
+                         (int4*)recA.data_ptr<int>(), n);
+  } else {
+    bool smallA = m <= PACK_MAX_M, smallB = U <= PACK_MAX_M;
+    torch::Tensor histA;
+    // histograms: one fused read when both key ranges are LDS-sized
+    if (smallA && smallB && m + U <= 2 * PACK_MAX_M) {
+      auto hist_both = torch::zeros({m + U}, opts_i32);  // one fill
+      histA = hist_both.narrow(0, 0, m);
+      histB = hist_both.narrow(0, m, U);
+      int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+      size_t lds = (size_t)(m + U) * sizeof(int);
+      if (item_slots.dtype() == torch::kInt32)
+        hipLaunchKernelGGL((k_dual_hist<int>), dim3(std::max(hg, 1)),
+                           dim3(256), lds, cur_stream(),
+                           item_slots.data_ptr<int>(),
+                           user_rows.data_ptr<int>(),
+                           histA.data_ptr<int>(), histB.data_ptr<int>(),
+                           n, (int)m, (int)U);
+      else
+        hipLaunchKernelGGL((k_dual_hist<int64_t>), dim3(std::max(hg, 1)),
+                           dim3(256), lds, cur_stream(),
+                           item_slots.data_ptr<int64_t>(),
+                           user_rows.data_ptr<int64_t>(),
+                           histA.data_ptr<int>(), histB.data_ptr<int>(),
+                           n, (int)m, (int)U);
+    } else if (smallB) {
+      histB = torch::zeros({U}, opts_i32);
+      int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
+      if (user_rows.dtype() == torch::kInt32)
+        hipLaunchKernelGGL((k_key_hist<int>), dim3(std::max(hg, 1)),
+                           dim3(256), (size_t)U * sizeof(int),
+                           cur_stream(), user_rows.data_ptr<int>(),
+                           histB.data_ptr<int>(), n, (int)U);
+      else
+        hipLaunchKernelGGL((k_key_hist<int64_t>), dim3(std::max(hg, 1)),
+                           dim3(256), (size_t)U * sizeof(int),
+                           cur_stream(), user_rows.data_ptr<int64_t>(),
+                           histB.data_ptr<int>(), n, (int)U);
+    }
+    if (reuse_tag) {
+      posA = torch::empty({n}, opts_i32);
+      posB = torch::empty({n}, opts_i32);
+    }
+    recA = build_rec(item_slots, user_rows, ratings, m, histA, posA);
   }
 
   // pass A: recs by item slot; fused e + item-delta accumulation
-  auto recA = build_rec(item_slots, user_rows, ratings, m, histA);
   int64_t reps = delta_reps(n, m, dim);
   torch::Tensor deltas;
   if (reps > 1) {
@@ -1023,23 +1164,21 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   }
 
   // pass B: recs by user row; apply user updates in place
-  if (!smallB) histB = torch::Tensor();
-  else if (!histB.defined()) {
-    auto opts_i32 = torch::dtype(torch::kInt32).device(ratings.device());
-    histB = torch::zeros({U}, opts_i32);
-    int hg = (int)std::min<int64_t>((n + 4095) / 4096, 512);
-    if (user_rows.dtype() == torch::kInt32)
-      hipLaunchKernelGGL((k_key_hist<int>), dim3(std::max(hg, 1)),
-                         dim3(256), (size_t)U * sizeof(int), cur_stream(),
-                         user_rows.data_ptr<int>(), histB.data_ptr<int>(),
-                         n, (int)U);
-    else
-      hipLaunchKernelGGL((k_key_hist<int64_t>), dim3(std::max(hg, 1)),
-                         dim3(256), (size_t)U * sizeof(int), cur_stream(),
-                         user_rows.data_ptr<int64_t>(),
-                         histB.data_ptr<int>(), n, (int)U);
+  torch::Tensor recB;
+  if (hit) {
+    recB = torch::empty({n, 4}, opts_i32);
+    hipLaunchKernelGGL(k_apply_pos_swap, dim3(grid_for(n, 256)),
+                       dim3(256), 0, cur_stream(),
+                       (const int4*)recA.data_ptr<int>(),
+                       hit->posB.data_ptr<int>(),
+                       (int4*)recB.data_ptr<int>(), n);
+  } else {
+    recB = build_rec_swap(recA, U, histB, posB);
+    if (reuse_tag) {
+      if (g_pos_cache.size() >= 32) g_pos_cache.clear();
+      g_pos_cache[ck] = {item_slots, user_rows, posA, posB};
+    }
   }
-  auto recB = build_rec_swap(recA, U, histB);
   seg_fused<2, true, false>(user_table, recB, user_table, item_vecs,
                             dim4, lr, -lr * reg, 0.0);
   return deltas;
@@ -1184,7 +1323,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K2 ps_apply_adagrad over a row-sorted batch");
   m.def("hash_init", &hash_init, "K6 deterministic per-id init");
   m.def("hash_init_rows", &hash_init_rows, "K6 slab init");
-  m.def("mf_sgd_step", &mf_sgd_step, "K4 fused MF SGD step");
+  m.def("mf_sgd_step", &mf_sgd_step, "K4 fused MF SGD step",
+        py::arg("user_table"), py::arg("user_rows"), py::arg("item_vecs"),
+        py::arg("item_slots"), py::arg("ratings"), py::arg("lr"),
+        py::arg("reg") = 0.0, py::arg("reuse_tag") = 0);
   m.def("sgns_step", &sgns_step, "K8 fused SGNS step");
   m.def("pa_step", &pa_step, "K7 PA binary step");
   m.def("pa_multiclass_step", &pa_multiclass_step, "K7 PA multiclass");

@@ -84,7 +84,13 @@ class OnlineMFWorkerLogic(WorkerLogic):
                  pull_limit: Optional[int] = None,
                  output_user_factors: bool = False,
                  output_user_factors_on_close: bool = False,
+                 assume_static_batches: bool = False,
                  seed: int = 1):
+        """``assume_static_batches=True``: the data source re-yields the
+        SAME batch objects with unchanged index tensors (epoch sweeps
+        over a resident dataset — the offline-MF pattern); the fused
+        kernel then caches its sort positions per batch.  Never set it
+        for sources that rewrite batch buffers in place."""
         self.n_factors = int(n_factors)
         self.lr = float(learning_rate)
         self.reg = float(reg)
@@ -95,6 +101,7 @@ class OnlineMFWorkerLogic(WorkerLogic):
         self.output_user_factors = output_user_factors
         self.output_on_close = (output_user_factors or
                                 output_user_factors_on_close)
+        self.assume_static = bool(assume_static_batches)
         self._neg_counter = 0
         self._seed = seed
         # Worker-local user table (reference: per-subtask user map with
@@ -174,9 +181,13 @@ class OnlineMFWorkerLogic(WorkerLogic):
                slots: Optional[torch.Tensor] = None) -> None:
         if slots is None:
             slots = buf.slots_of(batch.item_ids)
+        # cacheable only when slots are the batch's own stable tensors
+        # (dense pull); coalescer slots are fresh every round
+        tag = id(batch) if (self.assume_static and self._dense_pull and
+                            self.neg_rate == 0) else 0
         deltas = ops.mf_sgd_step(self.user_table, batch.user_rows,
                                  buf.values, slots, batch.ratings,
-                                 self.lr, self.reg)
+                                 self.lr, self.reg, reuse_tag=tag)
         client.push(buf.uids, deltas)
         if self.output_user_factors:
             client.output((batch.user_rows,

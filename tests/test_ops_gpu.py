@@ -205,3 +205,28 @@ def test_mf_engine_e2e_on_gpu():
     torch.cuda.synchronize()
     l1 = loss()
     assert l1 < 0.3 * l0, (l0, l1)
+
+
+def test_mf_sgd_step_position_cache_matches():
+    """reuse_tag: the second visit of the same (rows, slots) buffers goes
+    through the cached-position fast path and must match the oracle."""
+    torch.manual_seed(11)
+    n, U, M, dim = 5000, 300, 200, 32
+    users = torch.randint(0, U, (n,), dtype=torch.int32)
+    slots = torch.randint(0, M, (n,), dtype=torch.int32)
+    ut_gpu = torch.randn(U, dim).to(DEV)
+    users_g, slots_g = users.to(DEV), slots.to(DEV)
+    tag = 987654321
+    for visit in range(3):  # visit 0 fills the cache, 1..2 hit it
+        r = torch.randn(n)
+        iv = torch.randn(M, dim)
+        ut_cpu = ut_gpu.cpu().clone()
+        d_cpu = R.mf_sgd_step(ut_cpu, users.long(), iv, slots.long(), r,
+                              0.05, 0.01)
+        d_gpu = ops.mf_sgd_step(ut_gpu, users_g, iv.to(DEV), slots_g,
+                                r.to(DEV), 0.05, 0.01, reuse_tag=tag)
+        torch.cuda.synchronize()
+        assert torch.allclose(ut_cpu, ut_gpu.cpu(), atol=1e-3,
+                              rtol=1e-3), visit
+        assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3,
+                              rtol=1e-3), visit
