@@ -625,6 +625,131 @@ __global__ __launch_bounds__(256) void k_seg_fused(
   }
 }
 
+// LDS-staged variant: the BLOCK cooperatively copies its groups'
+// chunks (block-contiguous samples) into LDS with coalesced loads, and
+// the walks read records from LDS — the per-iteration dependent global
+// record load disappears from the latency chain (guide: stage tiles
+// through LDS).  Coefficient write-back (MODE 0/1) still goes to the
+// global rec array for the second pass.
+template <int GP, int EPL, int MODE, bool SELF, bool ZERO_OUT>
+__global__ __launch_bounds__(256) void k_seg_fused_lds(
+    float4* __restrict__ out, int4* __restrict__ rec,
+    const float4* __restrict__ keyed, const float4* __restrict__ other,
+    int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
+    int64_t rep_stride, int rep_mask) {
+  extern __shared__ int4 srec[];
+  const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
+  const int lane = threadIdx.x % GP;
+  const int gpb = blockDim.x / GP;
+  const int my_g = threadIdx.x / GP;
+  const int64_t nchunks = (n + chunk - 1) / chunk;
+  for (int64_t base = (int64_t)blockIdx.x * gpb; base < nchunks;
+       base += (int64_t)gridDim.x * gpb) {
+    const int64_t blk_first = base * chunk;
+    const int64_t blk_total = min((int64_t)gpb * chunk, n - blk_first);
+    __syncthreads();
+    for (int64_t t = threadIdx.x; t < blk_total; t += blockDim.x)
+      srec[t] = rec[blk_first + t];
+    __syncthreads();
+    const int64_t cidx = base + my_g;
+    if (cidx >= nchunks) continue;
+    const int64_t start = cidx * chunk;
+    const int64_t end = min(n, start + (int64_t)chunk);
+    const int4* mine = srec + (int64_t)my_g * chunk;
+    const int len = (int)(end - start);
+
+    float4 acc[EPL], kv[EPL];
+    int cnt = 0;
+    int cur = mine[0].x;
+    bool left_owned = (start == 0);
+
+    auto load_kv = [&]() {
+      if (MODE == 2 && !SELF) return;
+      const int64_t kr = (int64_t)cur * dim4;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) {
+        int c = lane + e * GP;
+        kv[e] = (c < dim4) ? keyed[kr + c] : make_float4(0, 0, 0, 0);
+      }
+    };
+    auto flush = [&](bool right_owned) {
+      const bool excl = left_owned && right_owned;
+      const int64_t orow = rep_off + (int64_t)cur * dim4;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) {
+        int c = lane + e * GP;
+        if (c < dim4) {
+          float4 t = acc[e];
+          t.x *= scale; t.y *= scale; t.z *= scale; t.w *= scale;
+          if (SELF) {
+            const float ac = a * cnt;
+            t.x += ac * kv[e].x; t.y += ac * kv[e].y;
+            t.z += ac * kv[e].z; t.w += ac * kv[e].w;
+          }
+          if (excl) {
+            if (ZERO_OUT) {
+              out[orow + c] = t;
+            } else {
+              float4 o = out[orow + c];
+              o.x += t.x; o.y += t.y; o.z += t.z; o.w += t.w;
+              out[orow + c] = o;
+            }
+          } else {
+            float* o = (float*)&out[orow + c];
+            atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
+            atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
+          }
+        }
+        acc[e] = make_float4(0, 0, 0, 0);
+      }
+    };
+
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) acc[e] = make_float4(0, 0, 0, 0);
+    load_kv();
+    for (int lj = 0; lj < len; ++lj) {
+      const int4 rj = mine[lj];
+      if (rj.x != cur) {
+        flush(true);
+        left_owned = true;
+        cur = rj.x;
+        cnt = 0;
+        load_kv();
+      }
+      const int64_t sr = (int64_t)rj.y * dim4;
+      float4 sv[EPL];
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) {
+        int c = lane + e * GP;
+        sv[e] = (c < dim4) ? other[sr + c] : make_float4(0, 0, 0, 0);
+      }
+      float g;
+      if (MODE == 2) {
+        g = __int_as_float(rj.z);
+      } else {
+        float dot = 0.f;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e)
+          dot += kv[e].x * sv[e].x + kv[e].y * sv[e].y +
+                 kv[e].z * sv[e].z + kv[e].w * sv[e].w;
+        dot = group_sum<GP>(dot);
+        g = (MODE == 0)
+                ? (__int_as_float(rj.z) - dot)
+                : lrc * (__int_as_float(rj.z) -
+                         1.0f / (1.0f + __expf(-dot)));
+        if (lane == 0) ((int*)&rec[start + lj])[2] = __float_as_int(g);
+      }
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) {
+        acc[e].x += g * sv[e].x; acc[e].y += g * sv[e].y;
+        acc[e].z += g * sv[e].z; acc[e].w += g * sv[e].w;
+      }
+      ++cnt;
+    }
+    flush(end == n);
+  }
+}
+
 // ---------------------------------------------------------------------
 // K7 (binary) / K9: PA and LR share the sparse-margin skeleton.
 // One wave per instance; lanes stride the instance's nnz.
@@ -1016,9 +1141,25 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
   if (!n) return;
   int chunk = seg_chunk();
   int64_t nchunks = (n + chunk - 1) / chunk;
+  static int use_lds = env_int("FPS_SEG_LDS", 0);
   by_group4(dim4, [&](auto tag, auto etag) {
     constexpr int GP = decltype(tag)::value;
     constexpr int EPL = decltype(etag)::value;
+    if (use_lds) {
+      size_t lds = (size_t)(256 / GP) * chunk * sizeof(int4);
+      if (lds <= 64 * 1024) {
+        hipLaunchKernelGGL(
+            (k_seg_fused_lds<GP, EPL, MODE, SELF, ZERO_OUT>),
+            dim3(grid_for(nchunks * GP, 256)), dim3(256), lds,
+            cur_stream(), (float4*)out.data_ptr<float>(),
+            (int4*)rec.data_ptr<int>(),
+            (const float4*)keyed.data_ptr<float>(),
+            (const float4*)other.data_ptr<float>(), n, dim4,
+            (float)scale, (float)a, (float)lrc, chunk, rep_stride / 4,
+            rep_mask);
+        return;
+      }
+    }
     hipLaunchKernelGGL((k_seg_fused<GP, EPL, MODE, SELF, ZERO_OUT>),
                        dim3(grid_for(nchunks * GP, 256)), dim3(256), 0,
                        cur_stream(), (float4*)out.data_ptr<float>(),
