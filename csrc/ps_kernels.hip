@@ -1141,7 +1141,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
   if (!n) return;
   int chunk = seg_chunk();
   int64_t nchunks = (n + chunk - 1) / chunk;
-  static int use_lds = env_int("FPS_SEG_LDS", 0);
+  static int use_lds = env_int("FPS_SEG_LDS", 1);
   by_group4(dim4, [&](auto tag, auto etag) {
     constexpr int GP = decltype(tag)::value;
     constexpr int EPL = decltype(etag)::value;
