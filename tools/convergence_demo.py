@@ -1,0 +1,102 @@
+#!/usr/bin/env python3
+"""Convergence-at-speed evidence: train rank-32 MF on a PLANTED
+MovieLens-1M-shaped low-rank matrix through the full engine (pull ->
+fused HIP step -> push -> scatter-apply) and log the loss curve with
+wall-clock — demonstrating the benchmarked throughput does real
+optimization, not just data movement.
+
+    python tools/convergence_demo.py            # GPU (or CPU fallback)
+
+Writes one JSON line (curve + rates) to stdout; redirect into
+profiles/ for the record.
+"""
+import json
+import math
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from flink_parameter_server_amd import (PSConfig, PSEngine, ShardedTable,
+                                        TableServerLogic)
+from flink_parameter_server_amd.models.mf import MFBatch, OnlineMFWorkerLogic
+from flink_parameter_server_amd.parallel.partitioner import ModPartitioner
+
+
+def main():
+    dev = torch.device("cuda" if torch.cuda.is_available() else "cpu")
+    if dev.type == "cuda":
+        n_users, n_items, k = 6040, 3952, 32
+        n_ratings, batch, epochs = 4 << 20, 1 << 17, 30
+    else:  # CPU smoke: scaled shape with the same rating density
+        n_users, n_items, k = 600, 400, 32
+        n_ratings, batch, epochs = 1 << 16, 1 << 12, 30
+
+    g = torch.Generator().manual_seed(7)
+    U = torch.randn(n_users, k, generator=g) / math.sqrt(k)
+    V = torch.randn(n_items, k, generator=g) / math.sqrt(k)
+    users = torch.randint(0, n_users, (n_ratings,), generator=g)
+    zipf = torch.rand(n_ratings, generator=g).pow(3.0)
+    items = (n_items * zipf).long().clamp_(0, n_items - 1)
+    ratings = (U[users] * V[items]).sum(1)
+    var = float(ratings.var())
+
+    batches = [MFBatch(users[s:s + batch].to(dev, torch.int32),
+                       items[s:s + batch].to(dev, torch.int32),
+                       ratings[s:s + batch].to(dev))
+               for s in range(0, n_ratings, batch)]
+
+    part = ModPartitioner(n_items, 1)
+    table = ShardedTable(part, 0, k, dev, init=(-0.15, 0.15))
+    wl = OnlineMFWorkerLogic(n_users, k, learning_rate=0.05, reg=1e-4,
+                             range_min=-0.15, range_max=0.15,
+                             n_items=n_items, device=dev,
+                             assume_static_batches=True)
+    eng = PSEngine(PSConfig(num_ids=n_items, dim=k), wl,
+                   TableServerLogic(table, dump_on_close=False,
+                                    pushes_unique=True),
+                   None, part, device=dev)
+
+    def mse():
+        tot = 0.0
+        for b in batches:
+            pred = (wl.user_table[b.user_rows.long()] *
+                    table.data[b.item_ids.long()]).sum(1)
+            tot += float(((pred - b.ratings) ** 2).sum())
+        return tot / n_ratings
+
+    if dev.type == "cuda":
+        torch.cuda.synchronize()
+    curve = [{"epoch": 0, "mse": mse(), "sec": 0.0}]
+    t0 = time.perf_counter()
+    it = iter([])
+    trained = 0
+    for ep in range(1, epochs + 1):
+        for b in batches:
+            eng._round(iter([b] * 1), False)
+            trained += len(b)
+        # drain the 1-round pipeline so the epoch's pushes are applied
+        eng._round(iter([]), True)
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+        curve.append({"epoch": ep, "mse": mse(),
+                      "sec": round(time.perf_counter() - t0, 3)})
+    elapsed = time.perf_counter() - t0
+    out = {
+        "what": "planted rank-32 MF, MovieLens-1M shape, full PS engine",
+        "device": str(dev), "rating_variance": var,
+        "n_ratings": n_ratings, "epochs": epochs,
+        "samples_trained": trained,
+        "samples_per_sec": trained / elapsed,
+        "final_mse_over_var": curve[-1]["mse"] / var,
+        "curve": curve,
+    }
+    print(json.dumps(out))
+    assert curve[-1]["mse"] < 0.1 * var, "did not converge"
+
+
+if __name__ == "__main__":
+    main()
