@@ -40,7 +40,8 @@
 #include <ATen/hip/HIPContext.h>
 
 #include <cstdlib>
-#include <unordered_map>
+#include <map>
+#include <tuple>
 
 #include <rocprim/rocprim.hpp>
 
@@ -1198,15 +1199,15 @@ static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
 struct PosCacheEntry {
   torch::Tensor keys, slots, posA, posB, cntB;
 };
-static std::unordered_map<uint64_t, PosCacheEntry> g_pos_cache;
+// exact composite key (no hash collisions): the held tensor references
+// inside each entry keep the keyed storages alive, so a pointer can
+// never alias a different live tensor.
+using PosCacheKey = std::tuple<int64_t, const void*, const void*, int64_t>;
+static std::map<PosCacheKey, PosCacheEntry> g_pos_cache;
 
-static uint64_t pos_cache_key(int64_t tag, const torch::Tensor& a,
-                              const torch::Tensor& b, int64_t n) {
-  uint64_t h = (uint64_t)tag * 0x9E3779B97F4A7C15ull;
-  h ^= (uint64_t)a.data_ptr() + 0x9E3779B97F4A7C15ull + (h << 6);
-  h ^= (uint64_t)b.data_ptr() + 0x9E3779B97F4A7C15ull + (h << 6);
-  h ^= (uint64_t)n + (h >> 2);
-  return h;
+static PosCacheKey pos_cache_key(int64_t tag, const torch::Tensor& a,
+                                 const torch::Tensor& b, int64_t n) {
+  return {tag, a.data_ptr(), b.data_ptr(), n};
 }
 
 torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
@@ -1235,7 +1236,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   // cached-position fast path (epoch resweeps): one scatter per side
   // replaces histogram + scan + counting sort
   PosCacheEntry* hit = nullptr;
-  uint64_t ck = 0;
+  PosCacheKey ck{};
   if (reuse_tag) {
     ck = pos_cache_key(reuse_tag, item_slots, user_rows, n);
     auto it = g_pos_cache.find(ck);
