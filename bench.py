@@ -215,6 +215,10 @@ def main():
     xgmi = transport.allgather_int([eng.metrics.bytes_sent])
     xgmi_bytes = int(xgmi.sum())
 
+    # the throughput number is only valid if training stayed finite
+    finite = bool(torch.isfinite(table.data).all())
+    assert finite, "parameter table diverged (non-finite values)"
+
     if rank == 0:
         k = meta.get("factors", cfg.dim)
         result = {
