@@ -59,7 +59,7 @@ def make_mf(args, rank, world, device):
     table = ShardedTable(part, rank, k, device, init=(-0.01, 0.01))
     # epoch-resweep pattern: the same resident batches cycle, so the
     # fused kernel may cache its sort positions per batch
-    wl = OnlineMFWorkerLogic(users_local, k, learning_rate=0.01,
+    wl = OnlineMFWorkerLogic(users_local, k, learning_rate=1e-4,
                              reg=0.001, range_min=-0.01, range_max=0.01,
                              n_items=n_items, device=device,
                              assume_static_batches=True, seed=rank + 1)
@@ -69,7 +69,11 @@ def make_mf(args, rank, world, device):
         users = torch.randint(0, users_local, (args.batch,), generator=g)
         u = torch.rand(args.batch, generator=g)
         items = (n_items * u.pow(3.0)).long().clamp_(0, n_items - 1)
-        ratings = torch.rand(args.batch, generator=g) * 4.0 + 1.0
+        # centered ratings + small lr keep the SUM-aggregated hot-key
+        # updates (zipf head ~6% of a 2^20 batch) inside the stable
+        # regime for arbitrarily long runs — bench.py asserts the table
+        # stayed finite after the timed region
+        ratings = torch.randn(args.batch, generator=g) * 0.25
         # int32 indices: the fused kernels take them directly, halving
         # the metadata bytes of the histogram/pack passes
         batches.append(MFBatch(users.to(device, torch.int32),
