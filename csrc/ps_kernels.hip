@@ -450,6 +450,20 @@ __global__ __launch_bounds__(256) void k_apply_pos_swap(
 // DURING concurrent chunk flushes — a measured 0.04-scale divergence on
 // chunk-spanning segments; this form is race-free and bit-equal to the
 // oracle's per-occurrence subtraction.)
+// mean-aggregated user update: u = u*(1 - lr*reg*[cnt>0]) + acc/cnt
+// (per-key averaged mini-batch — effective per-row step stays lr no
+// matter how often the row appears in the batch)
+__global__ __launch_bounds__(256) void k_apply_mean_update(
+    float* __restrict__ table, const float* __restrict__ acc,
+    const int* __restrict__ cnt, int64_t rows, int64_t dim, float decay) {
+  int64_t total = rows * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = cnt[i / dim];
+    if (c) table[i] = table[i] * (1.0f - decay) + acc[i] / c;
+  }
+}
+
 __global__ __launch_bounds__(256) void k_scale_rows_by_count(
     float* __restrict__ table, const int* __restrict__ cnt, int64_t rows,
     int64_t dim, float coef) {
@@ -1213,7 +1227,7 @@ static PosCacheKey pos_cache_key(int64_t tag, const torch::Tensor& a,
 torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                           torch::Tensor item_vecs, torch::Tensor item_slots,
                           torch::Tensor ratings, double lr, double reg,
-                          int64_t reuse_tag) {
+                          int64_t reuse_tag, bool mean_users) {
   CHECK_DEV(user_table); CHECK_CONT(user_table);
   CHECK_DEV(item_vecs); CHECK_CONT(item_vecs);
   CHECK_CONT(user_rows); CHECK_CONT(item_slots); CHECK_CONT(ratings);
@@ -1337,7 +1351,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                        hit->posB.data_ptr<int>(),
                        (int4*)recB.data_ptr<int>(), n);
   } else {
-    if (!histB.defined() && reg != 0.0) {
+    if (!histB.defined() && (reg != 0.0 || mean_users)) {
       // radix-path fallback: per-user counts via bincount
       histB = torch::bincount(user_rows.dtype() == torch::kInt32
                                   ? user_rows.to(torch::kInt64)
@@ -1349,6 +1363,20 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       if (g_pos_cache.size() >= 32) g_pos_cache.clear();
       g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB};
     }
+  }
+  if (mean_users) {
+    TORCH_CHECK(cntB.defined() || histB.defined(),
+                "MF mean path needs per-user counts");
+    auto& cc = cntB.defined() ? cntB : histB;
+    auto acc = torch::zeros_like(user_table);
+    seg_fused<2, false, true>(acc, recB, user_table, item_vecs, dim4,
+                              lr, 0.0, 0.0);
+    hipLaunchKernelGGL(k_apply_mean_update,
+                       dim3(grid_for(U * dim, 256)), dim3(256), 0,
+                       cur_stream(), user_table.data_ptr<float>(),
+                       acc.data_ptr<float>(), cc.data_ptr<int>(), U, dim,
+                       (float)(lr * reg));
+    return deltas;
   }
   if (reg != 0.0) {
     TORCH_CHECK(cntB.defined() || histB.defined(),
@@ -1506,7 +1534,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mf_sgd_step", &mf_sgd_step, "K4 fused MF SGD step",
         py::arg("user_table"), py::arg("user_rows"), py::arg("item_vecs"),
         py::arg("item_slots"), py::arg("ratings"), py::arg("lr"),
-        py::arg("reg") = 0.0, py::arg("reuse_tag") = 0);
+        py::arg("reg") = 0.0, py::arg("reuse_tag") = 0,
+        py::arg("mean_users") = false);
   m.def("sgns_step", &sgns_step, "K8 fused SGNS step");
   m.def("pa_step", &pa_step, "K7 PA binary step");
   m.def("pa_multiclass_step", &pa_multiclass_step, "K7 PA multiclass");

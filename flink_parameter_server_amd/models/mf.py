@@ -85,12 +85,21 @@ class OnlineMFWorkerLogic(WorkerLogic):
                  output_user_factors: bool = False,
                  output_user_factors_on_close: bool = False,
                  assume_static_batches: bool = False,
+                 aggregate: str = "sum",
                  seed: int = 1):
         """``assume_static_batches=True``: the data source re-yields the
         SAME batch objects with unchanged index tensors (epoch sweeps
         over a resident dataset — the offline-MF pattern); the fused
         kernel then caches its sort positions per batch.  Never set it
-        for sources that rewrite batch buffers in place."""
+        for sources that rewrite batch buffers in place.
+
+        ``aggregate``: "sum" (reference parity — a batch's updates for a
+        row are the SUM over its occurrences, so a hot key's effective
+        step scales with its batch multiplicity) or "mean" (new
+        capability — item deltas AND user updates averaged per row,
+        keeping the effective per-row step at lr regardless of batch
+        composition; the stable choice for large batches over skewed
+        keys)."""
         self.n_factors = int(n_factors)
         self.lr = float(learning_rate)
         self.reg = float(reg)
@@ -102,6 +111,10 @@ class OnlineMFWorkerLogic(WorkerLogic):
         self.output_on_close = (output_user_factors or
                                 output_user_factors_on_close)
         self.assume_static = bool(assume_static_batches)
+        if aggregate not in ("sum", "mean"):
+            raise ValueError(f"unknown aggregate {aggregate!r}")
+        self.aggregate = aggregate
+        self._cnt_cache: dict = {}
         self._neg_counter = 0
         self._seed = seed
         # Worker-local user table (reference: per-subtask user map with
@@ -187,11 +200,31 @@ class OnlineMFWorkerLogic(WorkerLogic):
                             self.neg_rate == 0) else 0
         deltas = ops.mf_sgd_step(self.user_table, batch.user_rows,
                                  buf.values, slots, batch.ratings,
-                                 self.lr, self.reg, reuse_tag=tag)
+                                 self.lr, self.reg, reuse_tag=tag,
+                                 mean_users=self.aggregate == "mean")
+        if self.aggregate == "mean":
+            deltas = deltas * self._inv_counts(batch, slots,
+                                               buf.uids.numel())
         client.push(buf.uids, deltas)
         if self.output_user_factors:
             client.output((batch.user_rows,
                            self.user_table[batch.user_rows].clone()))
+
+    def _inv_counts(self, batch: MFBatch, slots: torch.Tensor,
+                    m: int) -> torch.Tensor:
+        """(m, 1) reciprocal occurrence counts of each pulled row in the
+        batch (cached per static batch object)."""
+        if self.assume_static:
+            hit = self._cnt_cache.get(id(batch))
+            if hit is not None and hit[0] is slots:
+                return hit[1]
+        cnt = torch.bincount(slots.long(), minlength=m).clamp_(min=1)
+        inv = cnt.reciprocal().to(torch.float32).unsqueeze(1)
+        if self.assume_static:
+            if len(self._cnt_cache) > 64:
+                self._cnt_cache.clear()
+            self._cnt_cache[id(batch)] = (slots, inv)
+        return inv
 
     def _with_negatives(self, data: MFBatch) -> MFBatch:
         """rate-N hash-RNG negative items per positive, rating 0

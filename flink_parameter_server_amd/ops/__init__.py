@@ -86,7 +86,8 @@ def hash_init_rows(out: torch.Tensor, first_id: int, stride: int,
 def mf_sgd_step(user_table: torch.Tensor, user_rows: torch.Tensor,
                 item_vecs: torch.Tensor, item_slots: torch.Tensor,
                 ratings: torch.Tensor, lr: float, reg: float = 0.0,
-                reuse_tag: int = 0) -> torch.Tensor:
+                reuse_tag: int = 0,
+                mean_users: bool = False) -> torch.Tensor:
     """``reuse_tag != 0`` promises that whenever the same (tag,
     user_rows buffer, item_slots buffer) recur, their CONTENT is
     unchanged (epoch resweeps over resident batches) — the HIP path then
@@ -94,10 +95,11 @@ def mf_sgd_step(user_table: torch.Tensor, user_rows: torch.Tensor,
     if _cuda(user_table):
         return hip.ext().mf_sgd_step(user_table, user_rows, item_vecs,
                                      item_slots, ratings, lr, reg,
-                                     reuse_tag)
+                                     reuse_tag, mean_users)
     # torch index ops need int64; the HIP path takes int32 directly
     return reference.mf_sgd_step(user_table, user_rows.long(), item_vecs,
-                                 item_slots.long(), ratings, lr, reg)
+                                 item_slots.long(), ratings, lr, reg,
+                                 mean_users=mean_users)
 
 
 def pa_step(indptr: torch.Tensor, feat_slots: torch.Tensor,

@@ -119,8 +119,8 @@ def apply_adagrad(table: torch.Tensor, accum: torch.Tensor,
 
 def mf_sgd_step(user_table: torch.Tensor, user_rows: torch.Tensor,
                 item_vecs: torch.Tensor, item_slots: torch.Tensor,
-                ratings: torch.Tensor, lr: float, reg: float = 0.0
-                ) -> torch.Tensor:
+                ratings: torch.Tensor, lr: float, reg: float = 0.0,
+                mean_users: bool = False) -> torch.Tensor:
     """One SGD pass over a batch of ratings.
 
     user_table : (U, k) worker-local user factors, updated in place
@@ -133,11 +133,20 @@ def mf_sgd_step(user_table: torch.Tensor, user_rows: torch.Tensor,
     u = user_table.index_select(0, user_rows)
     v = item_vecs.index_select(0, item_slots)
     e = ratings - (u * v).sum(dim=1)
-    du = lr * (e.unsqueeze(1) * v - reg * u)
     di = lr * (e.unsqueeze(1) * u - reg * v)
-    user_table.index_add_(0, user_rows, du)
     item_deltas = torch.zeros_like(item_vecs)
     item_deltas.index_add_(0, item_slots, di)
+    if mean_users:
+        # per-user averaged update: u = u*(1 - lr*reg*[cnt>0]) + acc/cnt
+        acc = torch.zeros_like(user_table)
+        acc.index_add_(0, user_rows, lr * e.unsqueeze(1) * v)
+        cnt = torch.bincount(user_rows,
+                             minlength=user_table.shape[0]).unsqueeze(1)
+        decay = torch.where(cnt > 0, 1.0 - lr * reg, 1.0)
+        user_table.mul_(decay).add_(acc / cnt.clamp(min=1))
+    else:
+        du = lr * (e.unsqueeze(1) * v - reg * u)
+        user_table.index_add_(0, user_rows, du)
     return item_deltas
 
 

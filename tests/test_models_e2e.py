@@ -7,7 +7,8 @@ import torch
 from flink_parameter_server_amd import (PSConfig, PSEngine, ShardedTable,
                                         TableServerLogic)
 from flink_parameter_server_amd.parallel.partitioner import ModPartitioner
-from flink_parameter_server_amd.models.mf import (OnlineMFWorkerLogic,
+from flink_parameter_server_amd.models.mf import (MFBatch,
+                                                  OnlineMFWorkerLogic,
                                                   epoch_iterator)
 from flink_parameter_server_amd.models.pa import PAWorkerLogic
 from flink_parameter_server_amd.models.lr import LRWorkerLogic
@@ -206,3 +207,48 @@ def test_full_range_gather_aliases_slab():
     sub = table.gather(part_ids)
     assert sub.data_ptr() != table.data.data_ptr()
     assert torch.equal(sub, table.data[[3, 5]])
+
+
+def test_mean_item_delta_stable_where_sum_diverges():
+    """item_delta='mean' (new capability): with a zipf-hot item and a
+    large batch, SUM-aggregated deltas scale the hot item's effective
+    step by its multiplicity and blow up; MEAN aggregation keeps the
+    model bounded and fitting."""
+    n_users, n_items, rank = 50, 40, 4
+    g = torch.Generator().manual_seed(2)
+    n = 20000
+    users = torch.randint(0, n_users, (n,), generator=g)
+    items = (n_items * torch.rand(n, generator=g).pow(4.0)).long() \
+        .clamp_(0, n_items - 1)  # very hot head
+    ratings = torch.rand(n, generator=g) * 4.0 + 1.0  # uncentered
+    batches = [MFBatch(users[s:s + 10000], items[s:s + 10000],
+                       ratings[s:s + 10000]) for s in (0, 10000)]
+
+    def run(mode):
+        part = ModPartitioner(n_items, 1)
+        table = ShardedTable(part, 0, rank, "cpu", init=(-0.1, 0.1))
+        wl = OnlineMFWorkerLogic(n_users, rank, learning_rate=0.05,
+                                 n_items=n_items, aggregate=mode)
+        eng = PSEngine(PSConfig(num_ids=n_items, dim=rank), wl,
+                       TableServerLogic(table), None, part)
+        eng.run(epoch_iterator(batches, 10))
+        return float(table.data.abs().max())
+
+    assert not (run("sum") < 1e3)      # sum path blows up (or NaN)
+    assert run("mean") < 1e3           # mean path stays bounded
+
+
+def test_mean_item_delta_converges():
+    n_users, n_items, rank = 60, 50, 4
+    batches, _, _ = synth.planted_mf_ratings(n_users, n_items, rank,
+                                             3000, 512, seed=5)
+    part = ModPartitioner(n_items, 1)
+    table = ShardedTable(part, 0, rank, "cpu", init=(-0.1, 0.1))
+    wl = OnlineMFWorkerLogic(n_users, rank, learning_rate=0.3,
+                             n_items=n_items, aggregate="mean")
+    eng = PSEngine(PSConfig(num_ids=n_items, dim=rank), wl,
+                   TableServerLogic(table), None, part)
+    l0 = _mf_loss(wl, table, batches)
+    eng.run(epoch_iterator(batches, 10))
+    l1 = _mf_loss(wl, table, batches)
+    assert l1 < 0.5 * l0
