@@ -230,3 +230,22 @@ def test_mf_sgd_step_position_cache_matches():
                               rtol=1e-3), visit
         assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3,
                               rtol=1e-3), visit
+
+
+def test_mf_sgd_step_mean_users_matches():
+    torch.manual_seed(13)
+    n, U, M, dim = 4000, 150, 120, 32
+    users = torch.randint(0, U, (n,))
+    slots = torch.randint(0, M, (n,))
+    r = torch.randn(n)
+    iv = torch.randn(M, dim)
+    ut_cpu = torch.randn(U, dim)
+    ut_gpu = ut_cpu.to(DEV)
+    d_cpu = R.mf_sgd_step(ut_cpu, users, iv, slots, r, 0.05, 0.01,
+                          mean_users=True)
+    d_gpu = ops.mf_sgd_step(ut_gpu, users.to(DEV), iv.to(DEV),
+                            slots.to(DEV), r.to(DEV), 0.05, 0.01,
+                            mean_users=True)
+    torch.cuda.synchronize()
+    assert torch.allclose(ut_cpu, ut_gpu.cpu(), atol=1e-3, rtol=1e-3)
+    assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3, rtol=1e-3)
