@@ -26,14 +26,26 @@
 //    keyed row is loaded once per segment into registers, so the
 //    former standalone pair kernel and its gathers disappear; the
 //    coefficient is written back into rec.val for the second pass.
+//  * the walks read their records from LDS: each block cooperatively
+//    stages its chunks (block-contiguous samples) with coalesced loads,
+//    removing the dependent global record load from the latency chain
+//    (measured +8.5% on the MF32 step).
 //  * duplicate keys accumulate in registers per (segment x chunk) and
 //    flush with ONE atomicAdd per column — zipfian hot keys otherwise
 //    serialize device-scope atomics (measured: first-cut one-kernel MF
 //    step ran at ~300 GB/s — profiles/mf32_r01_baseline_atomics.md).
-//    REPS replica spreading (blockIdx & 7) further splits the hottest
-//    rows across 8 buffer copies the host sums.
+//    REPS replica spreading (blockIdx & rep_mask) splits the hottest
+//    rows across up to 8 buffer copies the host sums; near-unique keys
+//    take the opposite path — a segment provably interior to one chunk
+//    flushes with plain vector stores.
+//  * epoch-resweep workloads (reuse_tag) cache the sort positions per
+//    batch: revisits replace histogram + scan + counting sort with one
+//    scatter pass per side (measured +70% on the MF32 step).
 //  * numerics: f32 master values end to end; the semantics oracle is
-//    ops/reference.py and tests compare against it elementwise.
+//    ops/reference.py and tests compare against it elementwise.  The
+//    MF L2 term applies analytically from per-row counts (race-free;
+//    see k_scale_rows_by_count), and mean_users gives per-row averaged
+//    mini-batch updates for stability under skewed keys.
 
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
