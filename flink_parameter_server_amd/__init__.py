@@ -25,7 +25,8 @@ from .transform import (transform, transform_local, transform_simple,
 from .messaging import (CountFlush, FlushAlways, FlushStrategy, Pull,
                         PullAnswer, Push, PSToWorker, SerializingTransport,
                         TimeFlush, WorkerToPS)
-from .algorithms import (ps_offline_mf, ps_online_mf, transform_binary,
+from .algorithms import (ps_offline_mf, ps_online_mf, train_lr,
+                         train_sgns, transform_binary,
                          transform_multiclass)
 from .utils.checkpoint import (PeriodicCheckpointer, latest_checkpoint,
                                load_shard, save_shard)

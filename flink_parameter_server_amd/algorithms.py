@@ -316,3 +316,95 @@ def transform_multiclass(data, feature_count: int, n_classes: int,
                          ps_parallelism, pa_method, C, pull_limit,
                          range_partitioning, batch_size,
                          n_classes=n_classes)
+
+
+# ---------------------------------------------------------------------
+# New-capability entry points (no reference counterpart; same in-process
+# execution shape as the parity entries above)
+# ---------------------------------------------------------------------
+
+def train_sgns(pairs, vocab_size: int, dim: int = 300,
+               learning_rate: float = 0.025, negatives: int = 5,
+               worker_parallelism: int = 1,
+               ps_parallelism: Optional[int] = None,
+               pull_limit: Optional[int] = None,
+               batch_size: int = 1024, seed: int = 11):
+    """Word2vec SGNS over the PS: positive skip-gram ``pairs`` (a
+    (centers, contexts) tensor pair or an iterable of (center, context)
+    ids), one 2V-row embedding table sharded like any parameter.
+    Returns the per-shard ``(row_ids, vectors)`` dumps — rows [0, V) are
+    center embeddings, [V, 2V) context embeddings."""
+    from .models.sgns import SGNSBatch, SGNSWorkerLogic
+
+    world = int(worker_parallelism)
+    shards = int(ps_parallelism or world)
+    if shards > world:
+        raise ValueError("ps_parallelism > worker_parallelism needs one "
+                         "rank per shard; raise worker_parallelism")
+    if (isinstance(pairs, (tuple, list)) and len(pairs) == 2
+            and torch.is_tensor(pairs[0])):
+        centers, contexts = pairs[0].long(), pairs[1].long()
+    else:
+        t = torch.tensor([(int(c), int(o)) for c, o in pairs],
+                         dtype=torch.int64).reshape(-1, 2)
+        centers, contexts = t[:, 0], t[:, 1]
+    per_rank: List[List[SGNSBatch]] = []
+    for r in range(world):
+        c, o = centers[r::world], contexts[r::world]
+        per_rank.append([SGNSBatch(c[s:s + batch_size],
+                                   o[s:s + batch_size])
+                         for s in range(0, int(c.numel()), batch_size)])
+    part = make_partitioner("mod", 2 * vocab_size, shards)
+    cfg = PSConfig(n_workers=world, n_shards=shards,
+                   num_ids=2 * vocab_size, dim=dim,
+                   pull_limit=pull_limit)
+
+    def make_server(rank: int) -> TableServerLogic:
+        table = ShardedTable(part, rank, dim, torch.device("cpu"),
+                             init=(-0.5 / dim, 0.5 / dim), seed=seed)
+        return TableServerLogic(table, pushes_unique=True)
+
+    results = transform_local(
+        world, lambda r: iter(per_rank[r]),
+        lambda r: SGNSWorkerLogic(vocab_size, dim, learning_rate,
+                                  negatives=negatives,
+                                  pull_limit=pull_limit, seed=seed + r),
+        make_server, cfg, part_factory=lambda rank: part)
+    return [o for _, souts in results for o in souts]
+
+
+def train_lr(data, feature_count: int, learning_rate: float = 0.05,
+             worker_parallelism: int = 1,
+             ps_parallelism: Optional[int] = None,
+             pull_limit: Optional[int] = None,
+             batch_size: int = 256, eps: float = 1e-8):
+    """Sparse logistic regression with server-side Adagrad (configs[3]).
+    ``data``: iterable of ``(features, label)`` with label in {0, 1} or
+    None (unlabeled -> probability prediction).  Returns
+    ``(predictions, weight_dumps)``."""
+    from .models.lr import LRWorkerLogic
+
+    world = int(worker_parallelism)
+    shards = int(ps_parallelism or world)
+    if shards > world:
+        raise ValueError("ps_parallelism > worker_parallelism needs one "
+                         "rank per shard; raise worker_parallelism")
+    per_rank = _instances_to_batches(data, feature_count, world,
+                                     batch_size, multiclass=False)
+    part = make_partitioner("mod", feature_count, shards)
+    cfg = PSConfig(n_workers=world, n_shards=shards,
+                   num_ids=feature_count, dim=1, pull_limit=pull_limit)
+
+    def make_server(rank: int) -> TableServerLogic:
+        table = ShardedTable(part, rank, 1, torch.device("cpu"),
+                             optimizer="adagrad", lr=learning_rate,
+                             eps=eps)
+        return TableServerLogic(table, pushes_unique=True)
+
+    results = transform_local(
+        world, lambda r: iter(per_rank[r]),
+        lambda r: LRWorkerLogic(pull_limit=pull_limit,
+                                emit_predictions=True,
+                                n_features=feature_count),
+        make_server, cfg, part_factory=lambda rank: part)
+    return _merged(results)

@@ -74,6 +74,8 @@ class LRWorkerLogic(WorkerLogic):
         if self.emit_predictions:
             unlabeled = torch.isnan(batch.labels)
             if bool(unlabeled.any()):
-                client.output((torch.nonzero(unlabeled).reshape(-1),
-                               probs[unlabeled]))
+                rows = torch.nonzero(unlabeled).reshape(-1)
+                out_ids = batch.ids[rows] if batch.ids is not None \
+                    else rows
+                client.output((out_ids, probs[unlabeled]))
         client.push(buf.uids, grads.reshape(-1, 1))

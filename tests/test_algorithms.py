@@ -159,3 +159,42 @@ def test_sgd_updater_record_mode():
     du, di = upd.delta(2.0, u, v)
     assert torch.allclose(du, torch.tensor([0.0, 1.0]))
     assert torch.allclose(di, torch.tensor([1.0, 0.0]))
+
+
+def test_train_sgns_entry_runs():
+    from flink_parameter_server_amd.algorithms import train_sgns
+    g = torch.Generator().manual_seed(4)
+    centers = torch.randint(0, 50, (2000,), generator=g)
+    contexts = (centers + torch.randint(1, 5, (2000,), generator=g)) % 50
+    dumps = train_sgns((centers, contexts), vocab_size=50, dim=16,
+                       worker_parallelism=2, batch_size=256)
+    rows = sum(int(ids.numel()) for ids, _ in dumps)
+    assert rows == 100  # 2V rows dumped across shards
+    full = torch.zeros(100, 16)
+    for ids, vecs in dumps:
+        if ids.numel():
+            full[ids] = vecs
+    assert torch.isfinite(full).all() and float(full.abs().sum()) > 0
+
+
+def test_train_lr_entry_learns():
+    from flink_parameter_server_amd.algorithms import train_lr
+    n, n_feat = 1500, 60
+    g = torch.Generator().manual_seed(6)
+    w = torch.randn(n_feat, generator=g)
+    insts, truths = [], {}
+    for j in range(n):
+        fids = torch.randperm(n_feat, generator=g)[:8]
+        vals = torch.randn(8, generator=g)
+        y = 1.0 if float(w[fids] @ vals) > 0 else 0.0
+        insts.append((list(zip(fids.tolist(), vals.tolist())),
+                      y if j < n - 200 else None))
+        truths[j] = y
+    preds, dumps = train_lr(insts, n_feat, worker_parallelism=2,
+                            batch_size=128)
+    got = {}
+    for ids, p in preds:
+        got.update(zip(ids.tolist(), p.tolist()))
+    assert len(got) == 200
+    acc = sum((got[i] >= 0.5) == (truths[i] == 1.0) for i in got) / len(got)
+    assert acc > 0.75, acc
