@@ -496,13 +496,6 @@ __global__ __launch_bounds__(256) void k_invert_order(
     pos[order[j]] = (int)j;
 }
 
-__global__ __launch_bounds__(256) void k_invert_i32(
-    const int* __restrict__ pos, int* __restrict__ order, int64_t n) {
-  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
-       j += (int64_t)gridDim.x * blockDim.x)
-    order[pos[j]] = (int)j;
-}
-
 static int bits_for(int64_t m) {
   int b = 1;
   while ((1ll << b) < m) ++b;
@@ -782,150 +775,6 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
     }
     flush(end == n);
   }
-}
-
-// gather-direct cached walk: when the sort order is cached
-// (reuse_tag revisits), the records are never materialized — the LDS
-// load phase gathers {key, slot, val} straight from the input arrays
-// via the cached order, eliminating both position-apply passes.
-// MODE 0 additionally emits the per-sample coefficient in SORTED order
-// (coalesced stores); MODE 2 gathers it back via a precomputed
-// sortedB->sortedA index.
-template <int GP, int EPL, int MODE, bool SELF, bool ZERO_OUT, class TIdx>
-__global__ __launch_bounds__(256) void k_seg_gather_lds(
-    float4* __restrict__ out, const TIdx* __restrict__ key_src,
-    const TIdx* __restrict__ slot_src, const float* __restrict__ val_src,
-    const int* __restrict__ order, const int* __restrict__ val_idx,
-    float* __restrict__ e_out, const float4* __restrict__ keyed,
-    const float4* __restrict__ other, int64_t n, int64_t dim4,
-    float scale, float a, float lrc, int chunk, int64_t rep_stride,
-    int rep_mask) {
-  extern __shared__ int4 srec[];
-  const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
-  const int lane = threadIdx.x % GP;
-  const int gpb = blockDim.x / GP;
-  const int my_g = threadIdx.x / GP;
-  const int64_t nchunks = (n + chunk - 1) / chunk;
-  for (int64_t base = (int64_t)blockIdx.x * gpb; base < nchunks;
-       base += (int64_t)gridDim.x * gpb) {
-    const int64_t blk_first = base * chunk;
-    const int64_t blk_total = min((int64_t)gpb * chunk, n - blk_first);
-    __syncthreads();
-    for (int64_t t = threadIdx.x; t < blk_total; t += blockDim.x) {
-      const int g = order[blk_first + t];
-      const float v = (MODE == 2)
-                          ? val_src[val_idx[blk_first + t]]
-                          : val_src[g];
-      srec[t] = make_int4((int)key_src[g], (int)slot_src[g],
-                          __float_as_int(v), 0);
-    }
-    __syncthreads();
-    const int64_t cidx = base + my_g;
-    if (cidx >= nchunks) continue;
-    const int64_t start = cidx * chunk;
-    const int64_t end = min(n, start + (int64_t)chunk);
-    const int4* mine = srec + (int64_t)my_g * chunk;
-    const int len = (int)(end - start);
-
-    float4 acc[EPL], kv[EPL];
-    int cnt = 0;
-    int cur = mine[0].x;
-    bool left_owned = (start == 0);
-
-    auto load_kv = [&]() {
-      if (MODE == 2 && !SELF) return;
-      const int64_t kr = (int64_t)cur * dim4;
-#pragma unroll
-      for (int e = 0; e < EPL; ++e) {
-        int c = lane + e * GP;
-        kv[e] = (c < dim4) ? keyed[kr + c] : make_float4(0, 0, 0, 0);
-      }
-    };
-    auto flush = [&](bool right_owned) {
-      const bool excl = left_owned && right_owned;
-      const int64_t orow = rep_off + (int64_t)cur * dim4;
-#pragma unroll
-      for (int e = 0; e < EPL; ++e) {
-        int c = lane + e * GP;
-        if (c < dim4) {
-          float4 t = acc[e];
-          t.x *= scale; t.y *= scale; t.z *= scale; t.w *= scale;
-          if (SELF) {
-            const float ac = a * cnt;
-            t.x += ac * kv[e].x; t.y += ac * kv[e].y;
-            t.z += ac * kv[e].z; t.w += ac * kv[e].w;
-          }
-          if (excl) {
-            if (ZERO_OUT) {
-              out[orow + c] = t;
-            } else {
-              float4 o = out[orow + c];
-              o.x += t.x; o.y += t.y; o.z += t.z; o.w += t.w;
-              out[orow + c] = o;
-            }
-          } else {
-            float* o = (float*)&out[orow + c];
-            atomicAdd(o + 0, t.x); atomicAdd(o + 1, t.y);
-            atomicAdd(o + 2, t.z); atomicAdd(o + 3, t.w);
-          }
-        }
-        acc[e] = make_float4(0, 0, 0, 0);
-      }
-    };
-
-#pragma unroll
-    for (int e = 0; e < EPL; ++e) acc[e] = make_float4(0, 0, 0, 0);
-    load_kv();
-    for (int lj = 0; lj < len; ++lj) {
-      const int4 rj = mine[lj];
-      if (rj.x != cur) {
-        flush(true);
-        left_owned = true;
-        cur = rj.x;
-        cnt = 0;
-        load_kv();
-      }
-      const int64_t sr = (int64_t)rj.y * dim4;
-      float4 sv[EPL];
-#pragma unroll
-      for (int e = 0; e < EPL; ++e) {
-        int c = lane + e * GP;
-        sv[e] = (c < dim4) ? other[sr + c] : make_float4(0, 0, 0, 0);
-      }
-      float g;
-      if (MODE == 2) {
-        g = __int_as_float(rj.z);
-      } else {
-        float dot = 0.f;
-#pragma unroll
-        for (int e = 0; e < EPL; ++e)
-          dot += kv[e].x * sv[e].x + kv[e].y * sv[e].y +
-                 kv[e].z * sv[e].z + kv[e].w * sv[e].w;
-        dot = group_sum<GP>(dot);
-        g = (MODE == 0)
-                ? (__int_as_float(rj.z) - dot)
-                : lrc * (__int_as_float(rj.z) -
-                         1.0f / (1.0f + __expf(-dot)));
-        if (lane == 0) e_out[start + lj] = g;  // sorted order: coalesced
-      }
-#pragma unroll
-      for (int e = 0; e < EPL; ++e) {
-        acc[e].x += g * sv[e].x; acc[e].y += g * sv[e].y;
-        acc[e].z += g * sv[e].z; acc[e].w += g * sv[e].w;
-      }
-      ++cnt;
-    }
-    flush(end == n);
-  }
-}
-
-// gather-of-gather composition: out[j] = a[b[j]]
-__global__ __launch_bounds__(256) void k_compose_idx(
-    const int* __restrict__ a, const int* __restrict__ b,
-    int* __restrict__ outv, int64_t n) {
-  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
-       j += (int64_t)gridDim.x * blockDim.x)
-    outv[j] = a[b[j]];
 }
 
 // ---------------------------------------------------------------------
@@ -1349,42 +1198,6 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
   });
 }
 
-// gather-direct cached-walk launcher (records never materialized)
-template <int MODE, bool SELF, bool ZERO_OUT>
-static void seg_gather(torch::Tensor out, torch::Tensor key_src,
-                       torch::Tensor slot_src, torch::Tensor val_src,
-                       torch::Tensor order, torch::Tensor val_idx,
-                       torch::Tensor e_out, const torch::Tensor& keyed,
-                       const torch::Tensor& other, int64_t n,
-                       int64_t dim4, double scale, double a, double lrc,
-                       int64_t rep_stride = 0, int rep_mask = 7) {
-  if (!n) return;
-  int chunk = seg_chunk();
-  int64_t nchunks = (n + chunk - 1) / chunk;
-  by_group4(dim4, [&](auto tag, auto etag) {
-    constexpr int GP = decltype(tag)::value;
-    constexpr int EPL = decltype(etag)::value;
-    size_t lds = (size_t)(256 / GP) * chunk * sizeof(int4);
-    TORCH_CHECK(lds <= 64 * 1024, "seg_gather LDS budget");
-    auto launch = [&](auto idx_tag) {
-      using TIdx = typename decltype(idx_tag)::type;
-      hipLaunchKernelGGL(
-          (k_seg_gather_lds<GP, EPL, MODE, SELF, ZERO_OUT, TIdx>),
-          dim3(grid_for(nchunks * GP, 256)), dim3(256), lds,
-          cur_stream(), (float4*)out.data_ptr<float>(),
-          key_src.data_ptr<TIdx>(), slot_src.data_ptr<TIdx>(),
-          val_src.data_ptr<float>(), order.data_ptr<int>(),
-          val_idx.defined() ? val_idx.data_ptr<int>() : nullptr,
-          e_out.defined() ? e_out.data_ptr<float>() : nullptr,
-          (const float4*)keyed.data_ptr<float>(),
-          (const float4*)other.data_ptr<float>(), n, dim4, (float)scale,
-          (float)a, (float)lrc, chunk, rep_stride / 4, rep_mask);
-    };
-    if (key_src.dtype() == torch::kInt32) launch(std::common_type<int>{});
-    else launch(std::common_type<int64_t>{});
-  });
-}
-
 // replica count heuristic for the hot-key delta flush (power of two,
 // env-sweepable)
 static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
@@ -1411,9 +1224,6 @@ static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
 // tensors, so a pointer can never alias a different live tensor.
 struct PosCacheEntry {
   torch::Tensor keys, slots, posA, posB, cntB;
-  // gather-direct walk: sorted_j -> input_j orders for both passes and
-  // the sortedB -> sortedA coefficient index
-  torch::Tensor orderA, orderB, idxB2A;
 };
 // exact composite key (no hash collisions): the held tensor references
 // inside each entry keep the keyed storages alive, so a pointer can
@@ -1457,55 +1267,6 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     ck = pos_cache_key(reuse_tag, item_slots, user_rows, n);
     auto it = g_pos_cache.find(ck);
     if (it != g_pos_cache.end()) hit = &it->second;
-  }
-
-  static int use_lds_walk = env_int("FPS_SEG_LDS", 1);
-  if (hit && use_lds_walk && hit->orderA.defined()) {
-    // gather-direct cached path: the walks gather fields straight from
-    // the input arrays via the cached sort orders — no record
-    // materialization, no position-apply passes.
-    auto opts_f32 = torch::dtype(torch::kFloat32).device(ratings.device());
-    auto e_sa = torch::empty({n}, opts_f32);
-    torch::Tensor none;
-    int64_t reps = delta_reps(n, m, dim);
-    torch::Tensor deltas;
-    if (reps > 1) {
-      auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
-      seg_gather<0, true, true>(rep_buf, item_slots, user_rows, ratings,
-                                hit->orderA, none, e_sa, item_vecs,
-                                user_table, n, dim4, lr, -lr * reg, 0.0,
-                                m * dim, (int)reps - 1);
-      deltas = rep_buf.view({reps, m, dim}).sum(0);
-    } else {
-      deltas = torch::zeros_like(item_vecs);
-      seg_gather<0, true, true>(deltas, item_slots, user_rows, ratings,
-                                hit->orderA, none, e_sa, item_vecs,
-                                user_table, n, dim4, lr, -lr * reg, 0.0);
-    }
-    if (mean_users) {
-      auto acc = torch::zeros_like(user_table);
-      seg_gather<2, false, true>(acc, user_rows, item_slots, e_sa,
-                                 hit->orderB, hit->idxB2A, none,
-                                 user_table, item_vecs, n, dim4, lr, 0.0,
-                                 0.0);
-      hipLaunchKernelGGL(k_apply_mean_update,
-                         dim3(grid_for(U * dim, 256)), dim3(256), 0,
-                         cur_stream(), user_table.data_ptr<float>(),
-                         acc.data_ptr<float>(), hit->cntB.data_ptr<int>(),
-                         U, dim, (float)(lr * reg));
-    } else {
-      if (reg != 0.0)
-        hipLaunchKernelGGL(k_scale_rows_by_count,
-                           dim3(grid_for(U * dim, 256)), dim3(256), 0,
-                           cur_stream(), user_table.data_ptr<float>(),
-                           hit->cntB.data_ptr<int>(), U, dim,
-                           (float)(-lr * reg));
-      seg_gather<2, false, false>(user_table, user_rows, item_slots,
-                                  e_sa, hit->orderB, hit->idxB2A, none,
-                                  user_table, item_vecs, n, dim4, lr,
-                                  0.0, 0.0);
-    }
-    return deltas;
   }
 
   torch::Tensor recA;
@@ -1612,29 +1373,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     recB = build_rec_swap(recA, U, histB, posB);
     if (reuse_tag) {
       if (g_pos_cache.size() >= 32) g_pos_cache.clear();
-      // gather-direct companions: sorted_j -> input_j orders and the
-      // sortedB -> sortedA coefficient index
-      auto orderA = torch::empty({n}, opts_i32);
-      auto orderB = torch::empty({n}, opts_i32);
-      auto idxB2A = torch::empty({n}, opts_i32);
-      hipLaunchKernelGGL(k_invert_i32, dim3(grid_for(n, 256)), dim3(256),
-                         0, cur_stream(), posA.data_ptr<int>(),
-                         orderA.data_ptr<int>(), n);
-      hipLaunchKernelGGL(k_invert_i32, dim3(grid_for(n, 256)), dim3(256),
-                         0, cur_stream(), posB.data_ptr<int>(),
-                         orderB.data_ptr<int>(), n);
-      hipLaunchKernelGGL(k_compose_idx, dim3(grid_for(n, 256)), dim3(256),
-                         0, cur_stream(), posA.data_ptr<int>(),
-                         orderB.data_ptr<int>(), idxB2A.data_ptr<int>(),
-                         n);
-      torch::Tensor cntB = histB;
-      if (!cntB.defined())
-        cntB = torch::bincount(user_rows.dtype() == torch::kInt32
-                                   ? user_rows.to(torch::kInt64)
-                                   : user_rows,
-                               {}, U).to(torch::kInt32);
-      g_pos_cache[ck] = {item_slots, user_rows, posA, posB, cntB,
-                        orderA, orderB, idxB2A};
+      g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB};
     }
   }
   if (mean_users) {

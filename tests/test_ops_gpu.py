@@ -249,27 +249,3 @@ def test_mf_sgd_step_mean_users_matches():
     torch.cuda.synchronize()
     assert torch.allclose(ut_cpu, ut_gpu.cpu(), atol=1e-3, rtol=1e-3)
     assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3, rtol=1e-3)
-
-
-def test_mf_sgd_step_position_cache_mean_matches():
-    """cached gather-direct walk with mean_users on revisits."""
-    torch.manual_seed(17)
-    n, U, M, dim = 4000, 200, 150, 32
-    users = torch.randint(0, U, (n,), dtype=torch.int32)
-    slots = torch.randint(0, M, (n,), dtype=torch.int32)
-    ut_gpu = torch.randn(U, dim).to(DEV)
-    users_g, slots_g = users.to(DEV), slots.to(DEV)
-    for visit in range(3):
-        r = torch.randn(n)
-        iv = torch.randn(M, dim)
-        ut_cpu = ut_gpu.cpu().clone()
-        d_cpu = R.mf_sgd_step(ut_cpu, users.long(), iv, slots.long(), r,
-                              0.05, 0.01, mean_users=True)
-        d_gpu = ops.mf_sgd_step(ut_gpu, users_g, iv.to(DEV), slots_g,
-                                r.to(DEV), 0.05, 0.01,
-                                reuse_tag=424242, mean_users=True)
-        torch.cuda.synchronize()
-        assert torch.allclose(ut_cpu, ut_gpu.cpu(), atol=1e-3,
-                              rtol=1e-3), visit
-        assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3,
-                              rtol=1e-3), visit
