@@ -510,16 +510,20 @@ static torch::Tensor radix_sort_recs(torch::Tensor keys,
   auto keys_out = torch::empty_like(keys);
   auto rec_out = torch::empty_like(rec);
   size_t tmp_bytes = 0;
-  rocprim::radix_sort_pairs(
+  hipError_t err = rocprim::radix_sort_pairs(
       nullptr, tmp_bytes, keys.data_ptr<int>(), keys_out.data_ptr<int>(),
       (const int4*)rec.data_ptr<int>(), (int4*)rec_out.data_ptr<int>(),
       n, 0, bits_for(m), cur_stream());
+  TORCH_CHECK(err == hipSuccess, "radix_sort_pairs sizing: ",
+              hipGetErrorString(err));
   auto tmp = torch::empty({(int64_t)tmp_bytes},
                           opts.dtype(torch::kUInt8));
-  rocprim::radix_sort_pairs(
+  err = rocprim::radix_sort_pairs(
       tmp.data_ptr(), tmp_bytes, keys.data_ptr<int>(),
       keys_out.data_ptr<int>(), (const int4*)rec.data_ptr<int>(),
       (int4*)rec_out.data_ptr<int>(), n, 0, bits_for(m), cur_stream());
+  TORCH_CHECK(err == hipSuccess, "radix_sort_pairs: ",
+              hipGetErrorString(err));
   return rec_out;
 }
 
