@@ -41,14 +41,16 @@ class PacedIter:
         return [0]
 
 
-def _run(staleness, fast_n=30, slow_n=10, slow_every=3):
+def _run(staleness, fast_n=30, slow_n=10, slow_every=3,
+         feedback="fresh"):
     world = 2
 
     def job(rank, transport):
         part = ModPartitioner(4, world)
         table = ShardedTable(part, rank, 1, "cpu")
         logic = ClockProbe()
-        eng = PSEngine(PSConfig(num_ids=4, dim=1, staleness=staleness),
+        eng = PSEngine(PSConfig(num_ids=4, dim=1, staleness=staleness,
+                                feedback=feedback),
                        logic, TableServerLogic(table), transport, part)
         # track max clock lead over the other rank while both active
         leads = []
@@ -99,3 +101,11 @@ def test_staleness_histogram_recorded():
     assert sum(hist.values()) == fast_m["rounds"]
     assert max(hist) <= 2 + 1  # bounded lag distribution
     assert min(hist) >= 0
+
+
+def test_ssp_bounds_clock_skew_merged_feedback():
+    (fast_c, fast_lead, fast_m), (slow_c, _, _) = _run(staleness=2,
+                                                       feedback="merged")
+    assert fast_c == 30 and slow_c == 10
+    assert fast_m["ssp_stalls"] > 0
+    assert fast_lead <= 2 + 1
