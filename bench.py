@@ -49,10 +49,19 @@ def cycle(batches):
 
 
 def make_mf(args, rank, world, device):
-    """MovieLens-1M shape: 6040 users x 3952 items (scaled by --scale);
-    rank-32 factors (or 128)."""
-    n_users = int(6040 * args.scale)
-    n_items = int(3952 * args.scale)
+    """mf32: MovieLens-1M shape (6040 users x 3952 items, scaled by
+    --scale), rank 32.  mf128: the BASELINE configs[4] shape — a
+    10^8 x 10^7 rating matrix at rank 128 (user partition ~51 GB/world
+    in HBM; MovieLens shape on CPU smoke runs), SSP staleness 4 unless
+    overridden."""
+    if args.model == "mf128" and device.type == "cuda":
+        n_users = int(10**8 * args.scale)
+        n_items = int(10**7 * args.scale)
+        if args.staleness is None:
+            args.staleness = 4
+    else:
+        n_users = int(6040 * args.scale)
+        n_items = int(3952 * args.scale)
     k = 128 if args.model == "mf128" else 32
     users_local = (n_users + world - 1) // world
     part = ModPartitioner(n_items, world)
@@ -80,7 +89,9 @@ def make_mf(args, rank, world, device):
                                items.to(device, torch.int32),
                                ratings.to(device)))
     cfg = PSConfig(num_ids=n_items, dim=k, staleness=args.staleness)
-    meta = {"model": f"SGD-MF rank{k} MovieLens-1M-shaped" +
+    shape = (f"{n_users}x{n_items}" if args.model == "mf128"
+             and device.type == "cuda" else "MovieLens-1M-shaped")
+    meta = {"model": f"SGD-MF rank{k} {shape}" +
                      (f" x{args.scale:g}" if args.scale != 1 else ""),
             "n_users": n_users, "n_items": n_items, "factors": k}
     return cfg, part, table, wl, batches, meta
