@@ -476,6 +476,36 @@ __global__ __launch_bounds__(256) void k_apply_mean_update(
   }
 }
 
+// touched-rows variant for huge sparse-touched tables: segment leaders
+// of the user-sorted recs emit (row, count) pairs, and only those rows
+// are decayed — a 10^8-row table must not be scanned per step.
+__global__ __launch_bounds__(256) void k_seg_extents(
+    const int4* __restrict__ rec, int64_t n, int* __restrict__ rows,
+    int* __restrict__ cnts, int* __restrict__ ntouch) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    const int row = rec[j].x;
+    if (j > 0 && rec[j - 1].x == row) continue;  // not the leader
+    int len = 1;
+    while (j + len < n && rec[j + len].x == row) ++len;
+    const int idx = atomicAdd(ntouch, 1);
+    rows[idx] = row;
+    cnts[idx] = len;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_scale_touched_rows(
+    float* __restrict__ table, const int* __restrict__ rows,
+    const int* __restrict__ cnts, const int* __restrict__ ntouch,
+    int64_t dim, float coef) {
+  const int64_t total = (int64_t)(*ntouch) * dim;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t t = i / dim, c = i - t * dim;
+    table[(int64_t)rows[t] * dim + c] *= 1.0f + coef * cnts[t];
+  }
+}
+
 __global__ __launch_bounds__(256) void k_scale_rows_by_count(
     float* __restrict__ table, const int* __restrict__ cnt, int64_t rows,
     int64_t dim, float coef) {
@@ -1367,13 +1397,9 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                        hit->posB.data_ptr<int>(),
                        (int4*)recB.data_ptr<int>(), n);
   } else {
-    if (!histB.defined() && (reg != 0.0 || mean_users)) {
-      // radix-path fallback: per-user counts via bincount
-      histB = torch::bincount(user_rows.dtype() == torch::kInt32
-                                  ? user_rows.to(torch::kInt64)
-                                  : user_rows,
-                              {}, U).to(torch::kInt32);
-    }
+    TORCH_CHECK(histB.defined() || !mean_users,
+                "mean_users needs per-user counts (table too large for "
+                "the LDS histogram)");
     recB = build_rec_swap(recA, U, histB, posB);
     if (reuse_tag) {
       if (g_pos_cache.size() >= 32) g_pos_cache.clear();
@@ -1395,13 +1421,29 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     return deltas;
   }
   if (reg != 0.0) {
-    TORCH_CHECK(cntB.defined() || histB.defined(),
-                "MF reg path needs per-user counts");
-    auto& cc = cntB.defined() ? cntB : histB;
-    hipLaunchKernelGGL(k_scale_rows_by_count,
-                       dim3(grid_for(U * dim, 256)), dim3(256), 0,
-                       cur_stream(), user_table.data_ptr<float>(),
-                       cc.data_ptr<int>(), U, dim, (float)(-lr * reg));
+    if (cntB.defined() || histB.defined()) {
+      auto& cc = cntB.defined() ? cntB : histB;
+      hipLaunchKernelGGL(k_scale_rows_by_count,
+                         dim3(grid_for(U * dim, 256)), dim3(256), 0,
+                         cur_stream(), user_table.data_ptr<float>(),
+                         cc.data_ptr<int>(), U, dim, (float)(-lr * reg));
+    } else {
+      // huge sparse-touched table: decay only the rows this batch
+      // touches, from the sorted recs' segment extents
+      auto trows = torch::empty({n}, opts_i32);
+      auto tcnts = torch::empty({n}, opts_i32);
+      auto ntouch = torch::zeros({1}, opts_i32);
+      hipLaunchKernelGGL(k_seg_extents, dim3(grid_for(n, 256)),
+                         dim3(256), 0, cur_stream(),
+                         (const int4*)recB.data_ptr<int>(), n,
+                         trows.data_ptr<int>(), tcnts.data_ptr<int>(),
+                         ntouch.data_ptr<int>());
+      hipLaunchKernelGGL(k_scale_touched_rows,
+                         dim3(grid_for(n * dim, 256)), dim3(256), 0,
+                         cur_stream(), user_table.data_ptr<float>(),
+                         trows.data_ptr<int>(), tcnts.data_ptr<int>(),
+                         ntouch.data_ptr<int>(), dim, (float)(-lr * reg));
+    }
   }
   seg_fused<2, false, false>(user_table, recB, user_table, item_vecs,
                              dim4, lr, 0.0, 0.0);

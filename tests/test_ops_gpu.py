@@ -249,3 +249,24 @@ def test_mf_sgd_step_mean_users_matches():
     torch.cuda.synchronize()
     assert torch.allclose(ut_cpu, ut_gpu.cpu(), atol=1e-3, rtol=1e-3)
     assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3, rtol=1e-3)
+
+
+def test_mf_sgd_step_huge_user_table_reg_matches():
+    """U beyond the LDS histogram: the touched-rows L2 decay path (no
+    full-table scan, no bincount) must match the oracle with duplicate
+    users."""
+    torch.manual_seed(19)
+    n, U, M, dim = 6000, 50000, 9000, 32
+    users = torch.randint(0, U, (n,))
+    slots = torch.randint(0, M, (n,))
+    r = torch.randn(n)
+    iv = torch.randn(M, dim)
+    ut_cpu = torch.randn(U, dim)
+    ut_gpu = ut_cpu.to(DEV)
+    d_cpu = R.mf_sgd_step(ut_cpu, users, iv, slots, r, 0.05, 0.01)
+    d_gpu = ops.mf_sgd_step(ut_gpu, users.to(DEV), iv.to(DEV),
+                            slots.to(DEV), r.to(DEV), 0.05, 0.01)
+    torch.cuda.synchronize()
+    assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3, rtol=1e-3)
+    du = (ut_cpu - ut_gpu.cpu()).abs()
+    assert float(du.max()) < 1e-3, float(du.max())
