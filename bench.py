@@ -156,8 +156,9 @@ def main():
     ap.add_argument("--scale", type=float, default=1.0,
                     help="scale factor on the MovieLens-1M shape")
     ap.add_argument("--staleness", type=int, default=None)
-    ap.add_argument("--lr-features", type=int, default=10**8,
-                    help="feature count for --model lr (1e9 on 8 GPUs)")
+    ap.add_argument("--lr-features", type=int, default=10**9,
+                    help="feature count for --model lr (the configs[3] "
+                         "shape; 8 GB of weights+accum at 1 GPU)")
     ap.add_argument("--n-batches", type=int, default=8,
                     help="distinct pre-generated batches cycled")
     ap.add_argument("--device", default=None)
