@@ -80,3 +80,28 @@ def test_transform_local_multi_rank():
             merged[i] = v
     assert len(merged) == n_ids
     assert all(v == world for v in merged.values())
+
+
+def test_transform_max_rounds_bounds_an_endless_stream():
+    import itertools
+
+    import torch
+
+    from flink_parameter_server_amd import (PSConfig, SimpleServerLogic,
+                                            WorkerLogic, transform)
+
+    class Puller(WorkerLogic):
+        def on_recv(self, data, client):
+            client.push(data, torch.ones(1, 1))
+
+        def on_pull_recv(self, ids, values, client):
+            pass
+
+    endless = (torch.tensor([0]) for _ in itertools.count())
+    server = SimpleServerLogic(lambda i: torch.zeros(1),
+                               lambda old, d: old + d)
+    wout, souts = transform(endless, Puller(), server,
+                            PSConfig(num_ids=1, dim=1), max_rounds=5)
+    # 5 rounds consumed 5 batches; close() still dumped the table
+    total = sum(float(v) for _, v in souts)
+    assert 0 < total <= 5.0
