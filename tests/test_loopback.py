@@ -205,3 +205,56 @@ def test_randomized_world4_stress(feedback):
         merged.update(d)
     got = torch.stack([merged[i] for i in range(n_ids)])
     assert torch.allclose(got, expected, atol=1e-5)
+
+
+def test_kitchen_sink_all_knobs_together():
+    """merged feedback + bf16 wire + count-flush + pull limit, 4 ranks:
+    every engine feature at once still delivers exactly-once answers and
+    the right totals (bf16 rounds the wire payloads, so compare with
+    tolerance)."""
+    from flink_parameter_server_amd.messaging import CountFlush
+
+    world, n_ids, dim = 4, 33, 2
+
+    class Traffic(WorkerLogic):
+        pull_limit = 6
+
+        def __init__(self, rank):
+            self.g = torch.Generator().manual_seed(500 + rank)
+            self.pushed = torch.zeros(n_ids, dim)
+            self.answered = 0
+
+        def on_recv(self, data, client):
+            ids = torch.unique(torch.randint(0, n_ids, (5,),
+                                             generator=self.g))
+            deltas = torch.randn(ids.numel(), dim, generator=self.g)
+            client.push(ids, deltas)
+            self.pushed[ids] += deltas
+            client.pull(ids, unique=True)
+
+        def on_pull_recv(self, ids, values, client):
+            self.answered += int(ids.numel())
+
+    def job(rank, transport):
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, dim, "cpu")
+        logic = Traffic(rank)
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=dim,
+                                feedback="merged", wire_dtype="bf16"),
+                       logic, TableServerLogic(table), transport, part,
+                       flush=CountFlush(8))
+        eng.run([r for r in range(15)])
+        assert eng.metrics.pull_ids_sent == logic.answered
+        gids, vals = table.dump()
+        return logic.pushed, dict(
+            zip(gids.tolist(), [v.clone() for v in vals]))
+
+    results = run_loopback(world, job)
+    expected = sum(r[0] for r in results)
+    merged = {}
+    for _, d in results:
+        merged.update(d)
+    got = torch.stack([merged[i] for i in range(n_ids)])
+    # bf16 wire: ~3 decimal digits per push crossing ranks
+    assert torch.allclose(got, expected, atol=0.15, rtol=0.05), \
+        (got - expected).abs().max()
