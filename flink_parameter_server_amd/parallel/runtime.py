@@ -158,6 +158,8 @@ class PSEngine:
         # received order, within the same round lets the engine drop
         # answer ids (and their counts exchange) from the feedback edge:
         # the puller reconstructs them from its own sent segments.
+        # NOTE: the flag shapes the collective channel layout, so every
+        # rank must run the same server-logic type (all bundled jobs do).
         self._answers_in_order = bool(getattr(server_logic,
                                               "answers_in_order", False))
 
