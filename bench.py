@@ -179,7 +179,7 @@ def main():
         torch.cuda.set_device(device)
 
     if args.batch == 0:
-        defaults = {"mf32": 1 << 20, "mf128": 1 << 19, "sgns": 1 << 17,
+        defaults = {"mf32": 1 << 22, "mf128": 1 << 20, "sgns": 1 << 17,
                     "lr": 1 << 17}
         args.batch = defaults[args.model] if device.type == "cuda" \
             else 4096
