@@ -68,7 +68,11 @@ def make_mf(args, rank, world, device):
     table = ShardedTable(part, rank, k, device, init=(-0.01, 0.01))
     # epoch-resweep pattern: the same resident batches cycle, so the
     # fused kernel may cache its sort positions per batch
-    wl = OnlineMFWorkerLogic(users_local, k, learning_rate=1e-4,
+    # per-sample lr scales inversely with batch so the SUM-aggregated
+    # per-hot-key step stays constant as the batch grows (stability is
+    # set by lr x key-multiplicity; bench.py asserts finiteness)
+    mf_lr = 1e-4 * min(1.0, (1 << 20) / args.batch)
+    wl = OnlineMFWorkerLogic(users_local, k, learning_rate=mf_lr,
                              reg=0.001, range_min=-0.01, range_max=0.01,
                              n_items=n_items, device=device,
                              assume_static_batches=True, seed=rank + 1)
@@ -102,7 +106,8 @@ def make_sgns(args, rank, world, device):
     V, dim = 1_000_000, 300
     part = ModPartitioner(2 * V, world)
     table = ShardedTable(part, rank, dim, device, init=(-0.01, 0.01))
-    wl = SGNSWorkerLogic(V, dim, learning_rate=0.025, negatives=5,
+    sg_lr = 0.025 * min(1.0, (1 << 17) / args.batch)
+    wl = SGNSWorkerLogic(V, dim, learning_rate=sg_lr, negatives=5,
                          seed=rank + 3)
     g = torch.Generator().manual_seed(99 + rank)
     batches = []
@@ -179,8 +184,8 @@ def main():
         torch.cuda.set_device(device)
 
     if args.batch == 0:
-        defaults = {"mf32": 1 << 22, "mf128": 1 << 20, "sgns": 1 << 17,
-                    "lr": 1 << 17}
+        defaults = {"mf32": 1 << 23, "mf128": 1 << 20, "sgns": 1 << 20,
+                    "lr": 1 << 20}
         args.batch = defaults[args.model] if device.type == "cuda" \
             else 4096
 
