@@ -184,8 +184,8 @@ def main():
         torch.cuda.set_device(device)
 
     if args.batch == 0:
-        defaults = {"mf32": 1 << 23, "mf128": 1 << 20, "sgns": 1 << 20,
-                    "lr": 1 << 20}
+        defaults = {"mf32": 1 << 23, "mf128": 1 << 22, "sgns": 1 << 20,
+                    "lr": 1 << 21}
         args.batch = defaults[args.model] if device.type == "cuda" \
             else 4096
 
