@@ -1040,7 +1040,7 @@ static int pack_chunk() {
 }
 
 static int seg_chunk() {
-  static int c = env_int("FPS_SEG_CHUNK", 64);
+  static int c = env_int("FPS_SEG_CHUNK", 256);
   return c;
 }
 
