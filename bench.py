@@ -167,9 +167,10 @@ def main():
     ap.add_argument("--n-batches", type=int, default=8,
                     help="distinct pre-generated batches cycled")
     ap.add_argument("--device", default=None)
-    ap.add_argument("--feedback", choices=["fresh", "merged"], default=None,
+    ap.add_argument("--feedback", choices=["fresh", "merged", "pipelined"],
+                    default=None,
                     help="feedback-edge schedule (default: fresh at 1 "
-                         "GPU, merged for multi-GPU)")
+                         "GPU, pipelined for multi-GPU)")
     args = ap.parse_args()
 
     world_env = int(os.environ.get("WORLD_SIZE", "1"))
@@ -199,10 +200,11 @@ def main():
 
     cfg, part, table, wl, batches, meta = MAKERS[args.model](
         args, rank, world, device)
-    # multi-GPU: answers ride the next round's single exchange — one
-    # all_gather + one grouped send/recv per round (collective latency
-    # is the scaling tax; +1 round of async staleness is free throughput)
-    fb = args.feedback or ("merged" if world > 1 else "fresh")
+    # multi-GPU: answers ride the next round's single exchange AND the
+    # exchange is split-phase — the fused training step overlaps the
+    # xGMI payload transfer (feedback="pipelined"; +2 rounds of async
+    # answer latency is free throughput)
+    fb = args.feedback or ("pipelined" if world > 1 else "fresh")
     cfg.feedback = fb
     eng = PSEngine(cfg, wl, TableServerLogic(table, dump_on_close=False,
                                     pushes_unique=True),

@@ -56,7 +56,21 @@ class PSConfig:
     #       (1 all_gather + 1 p2p phase per round; +1 round of answer
     #       latency / staleness — the throughput option for large
     #       multi-GPU jobs where per-round collective latency dominates).
+    #   "pipelined" — merged layout, split-phase: the exchange is ISSUED,
+    #       the round's fused compute (training on the answers received
+    #       last round) runs while the payloads are on the wire (RCCL
+    #       p2p progresses on its own HIP stream), then the exchange
+    #       completes and the server phase applies what arrived.  +1
+    #       further round of answer latency; comm/compute overlap — the
+    #       north star's double-buffered schedule and the multi-GPU
+    #       default.
     feedback: str = "fresh"
+
+    # Debug: re-verify content-derived caches (shard bucketing) every
+    # round instead of trusting buffer identity — catches worker logics
+    # that mutate an id buffer in place after handing it to the engine
+    # (forbidden by the client.pull/push contract).
+    debug_validate: bool = False
 
     # Partitioning of parameter ids over shards: "mod" (hash partitioner —
     # the reference's hash(paramId) % psParallelism; Scala Int hashCode is
@@ -75,5 +89,5 @@ class PSConfig:
             raise ValueError(f"unknown partitioner {self.partitioner!r}")
         if self.wire_dtype not in ("f32", "bf16"):
             raise ValueError(f"unknown wire_dtype {self.wire_dtype!r}")
-        if self.feedback not in ("fresh", "merged"):
+        if self.feedback not in ("fresh", "merged", "pipelined"):
             raise ValueError(f"unknown feedback {self.feedback!r}")

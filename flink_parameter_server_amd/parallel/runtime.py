@@ -26,6 +26,13 @@ round is exactly one counts+control all_gather + one grouped p2p phase.
 +1 round of answer latency / staleness; the throughput option for large
 multi-GPU jobs where per-round collective latency is the scaling tax.
 
+``"pipelined"`` — the merged layout, split-phase: each round ISSUES its
+exchange, trains on the answers held from the previous round's exchange
+while the payloads cross xGMI (RCCL p2p progresses on its own HIP
+stream, the fused kernels on the compute stream), then completes the
+exchange and runs the server phase.  +1 further round of latency;
+comm/compute overlap — the multi-GPU default.
+
 Asynchrony (both modes): the worker never blocks on the server — the
 reference's async pull/push contract; staleness comes from round/pipeline
 depth (the SSP knob bounds it on top).  ``pull_limit`` bounds distinct
@@ -154,14 +161,34 @@ class PSEngine:
         limit = worker_logic.pull_limit or config.pull_limit
         self._pull_limit = int(limit) if limit else None
 
+        # pipelined mode: answers received last round, delivered (and
+        # trained on) while THIS round's exchange is in flight.
+        self._held: List[Tuple[torch.Tensor, torch.Tensor]] = []
+
         # A server logic that answers every pull batch in full, in
         # received order, within the same round lets the engine drop
         # answer ids (and their counts exchange) from the feedback edge:
         # the puller reconstructs them from its own sent segments.
         # NOTE: the flag shapes the collective channel layout, so every
-        # rank must run the same server-logic type (all bundled jobs do).
+        # rank must run the same server-logic type (all bundled jobs do);
+        # the first round's protocol-homogeneity preflight fails fast on
+        # a mismatch instead of hanging in a malformed collective.
         self._answers_in_order = bool(getattr(server_logic,
                                               "answers_in_order", False))
+        self._proto_checked = False
+
+        # A flush strategy may coalesce several logical pushes into one
+        # round, so the concatenated push batch can repeat ids — which a
+        # pushes_unique server logic (non-atomic add / sort-free adagrad)
+        # would corrupt silently.  Reject the combination outright.
+        if (flush is not None and
+                getattr(server_logic, "pushes_unique", False)):
+            raise ValueError(
+                "a flush strategy coalesces pushes across rounds, so "
+                "push ids may repeat — incompatible with a server logic "
+                "constructed with pushes_unique=True (use "
+                "pushes_unique=False, whose apply path pre-aggregates "
+                "duplicates)")
 
     # ------------------------------------------------------------------
     def run(self, data: Iterable[Any], max_rounds: Optional[int] = None
@@ -185,9 +212,37 @@ class PSEngine:
     # ------------------------------------------------------------------
     def _round(self, it: Iterator[Any], exhausted: bool
                ) -> Tuple[bool, bool]:
+        if not self._proto_checked:
+            self._validate_protocol()
         if self.cfg.feedback == "merged":
             return self._round_merged(it, exhausted)
+        if self.cfg.feedback == "pipelined":
+            return self._round_pipelined(it, exhausted)
         return self._round_fresh(it, exhausted)
+
+    # ------------------------------------------------------------------
+    def _validate_protocol(self) -> None:
+        """Fail fast (instead of hanging in a malformed collective) when
+        ranks disagree on anything that shapes the wire protocol: the
+        answers_in_order elision, wire dtype, feedback schedule, row dim
+        or partitioner.  One tiny allgather before the first round."""
+        self._proto_checked = True
+        if self.world == 1:
+            return
+        import zlib
+        desc = "|".join([
+            str(int(self._answers_in_order)), self.cfg.wire_dtype,
+            self.cfg.feedback, str(self.cfg.dim), self.cfg.partitioner,
+            str(self.cfg.num_ids)])
+        word = zlib.crc32(desc.encode())
+        g = self.transport.allgather_int([word])
+        if int(g.min()) != int(g.max()):
+            raise RuntimeError(
+                f"wire-protocol mismatch across ranks: rank {self.rank} "
+                f"has ({desc}); protocol words per rank = "
+                f"{g.reshape(-1).tolist()}.  All ranks must agree on "
+                "server-logic answers_in_order, wire_dtype, feedback "
+                "mode, dim and partitioner.")
 
     # ------------------------------------------------------------------
     def _round_fresh(self, it: Iterator[Any], exhausted: bool
@@ -461,6 +516,140 @@ class PSEngine:
         return exhausted, done
 
     # ------------------------------------------------------------------
+    def _round_pipelined(self, it: Iterator[Any], exhausted: bool
+                         ) -> Tuple[bool, bool]:
+        """feedback="pipelined": the merged single-exchange layout,
+        split-phase — the exchange is ISSUED (counts+control resolve,
+        payloads stay on the wire), the round's fused training step runs
+        on the answers HELD from last round's exchange while RCCL's
+        grouped send/recv progresses on its own HIP stream, then the
+        exchange COMPLETES and the server phase applies what arrived.
+        This is the north star's double-buffered schedule: gradient
+        compute overlaps the xGMI all-to-all.
+
+        Timeline of a pull: requested by a callback in round r-1, ships
+        in r, served in r's server phase, ships back in r+1, received at
+        r+1's completion, HELD, delivered to ``on_pull_recv`` during
+        r+2's flight.  One round more answer latency than "merged"
+        (bounded staleness still holds: the SSP lag test uses this
+        round's fresh control, resolved at issue time)."""
+        m = self.metrics
+        m.rounds += 1
+
+        # 1. bucket the ops queued by last round's callbacks
+        if self.flush is not None and not self.flush.should_flush(
+                self._pending_ops(), force=exhausted or self._stalled):
+            m.held_rounds += 1
+            pull_ids = push_ids = push_deltas = None
+        else:
+            pull_ids = self._take_pulls()
+            push_ids, push_deltas = self._take_pushes()
+        send = self._bucket(pull_ids, push_ids, push_deltas)
+        sent_pulls = send[0]
+        m.pull_ids_sent += int(pull_ids.numel()) if pull_ids is not None else 0
+        m.push_ids_sent += int(push_ids.numel()) if push_ids is not None else 0
+
+        worker_done = int(exhausted and self._outstanding == 0 and
+                          self._queued_pulls() == 0 and
+                          not self._pull_buf and not self._push_buf and
+                          push_ids is None and
+                          not any(self._answer_buf) and
+                          not self._held)
+        dim = self.cfg.dim
+        wire = torch.bfloat16 if self.cfg.wire_dtype == "bf16" \
+            else torch.float32
+        self._wire_out(send[2], wire)
+        ans_vals = [_concat([v for _, v in self._answer_buf[w]])
+                    for w in range(self.world)]
+        self._wire_out(ans_vals, wire)
+        chans = [send[0], send[1], send[2]]
+        specs = [(torch.int64, ()), (torch.int64, ()), (wire, (dim,))]
+        if not self._answers_in_order:
+            chans.append([_concat([a for a, _ in self._answer_buf[w]])
+                          for w in range(self.world)])
+            specs.append((torch.int64, ()))
+        chans.append(ans_vals)
+        specs.append((wire, (dim,)))
+        self._answer_buf = [[] for _ in range(self.world)]
+        m.bytes_sent += self._xgmi_bytes(chans)
+
+        # 2. ISSUE the exchange.  Control rides the counts phase, so it
+        # is fully resolved when begin returns; only payloads remain in
+        # flight.
+        h = self.transport.alltoallv_begin(
+            chans, specs, extra=[self._clock, worker_done, int(exhausted)])
+        ctrl = self.transport.ctrl_of(h)
+
+        # 3. OVERLAPPED compute: deliver the held answers (the fused
+        # training step launches on the compute stream while the p2p
+        # payloads cross xGMI), then consume the next batch unless
+        # SSP-stalled / exhausted / over the pull limit.
+        held, self._held = self._held, []
+        if held:
+            ids = _concat([i for i, _ in held])
+            self._outstanding -= int(ids.numel())
+            self.worker_logic.on_pull_recv(ids,
+                                           _concat([v for _, v in held]),
+                                           self.client)
+        active = ctrl[:, 2] == 0
+        self._min_clock = int(ctrl[active, 0].min()) if bool(active.any()) \
+            else int(ctrl[:, 0].max())
+        lag = self._clock - self._min_clock
+        m.staleness_hist[lag] = m.staleness_hist.get(lag, 0) + 1
+        stalled = (self.cfg.staleness is not None and
+                   lag > self.cfg.staleness)
+        self._stalled = stalled
+        backpressure = (self._pull_limit is not None and
+                        self._queued_pulls() >= self._pull_limit)
+        if stalled:
+            m.ssp_stalls += 1
+        if not exhausted and not stalled and not backpressure:
+            try:
+                batch = next(it)
+            except StopIteration:
+                exhausted = True
+            else:
+                if batch is not None:
+                    self._clock += 1
+                    m.samples += _batch_size(self.worker_logic, batch)
+                    self.worker_logic.on_recv(batch, self.client)
+
+        # 4. COMPLETE the exchange; hold the received answers for the
+        # next round's overlapped delivery.
+        recv, _ = self.transport.alltoallv_end(h)
+        recv = self._to_device(recv)
+        self._unwire(recv[2])
+        self._unwire(recv[-1])
+        r_pull, r_push_ids, r_push_deltas = recv[0], recv[1], recv[2]
+        back_vals = recv[-1]
+        back_ids = self._prev_sent_pulls if self._answers_in_order \
+            else recv[3]
+        for s in range(self.world):
+            v = back_vals[s]
+            if v is not None and v.numel():
+                i = back_ids[s]
+                self._held.append(
+                    (i if i.device == v.device else i.to(v.device), v))
+        self._prev_sent_pulls = sent_pulls
+
+        # 5. server phase: pushes first (fresher answers), then pulls;
+        # produced answers ship in the NEXT round's exchange.
+        for s in range(self.world):
+            ids, deltas = r_push_ids[s], r_push_deltas[s]
+            if ids is not None and ids.numel():
+                self.server_logic.on_push_recv(ids, deltas, self.ps)
+        for s in range(self.world):
+            ids = r_pull[s]
+            if ids is not None and ids.numel():
+                self.server_logic.on_pull_recv(ids, s, self.ps)
+
+        if self.checkpointer is not None:
+            self.checkpointer.maybe(m.rounds)
+
+        done = bool(ctrl[:, 1].all())
+        return exhausted, done
+
+    # ------------------------------------------------------------------
     def _to_device(self, recv):
         """Move received segments onto the engine device (no-op when the
         transport already delivered them there, e.g. RCCL; a CPU-side
@@ -593,11 +782,21 @@ class PSEngine:
         arange; pushes reuse the pulled uids), so the argsort/bincount
         bucketing runs once, not every round.  Held references keep the
         cached storages alive, so a pointer can never alias a different
-        live tensor."""
+        live tensor.  The client.pull/push contract forbids mutating an
+        id buffer in place after handing it over; with
+        ``cfg.debug_validate`` the cache keeps a content snapshot and
+        re-verifies it on every hit, turning such a mutation into a
+        loud error instead of silent mis-routing."""
         key = (ids.data_ptr(), ids.numel())
         hit = self._bucket_cache.get(key)
         if hit is not None:
-            _, order, counts = hit
+            held, order, counts = hit
+            if (self.cfg.debug_validate and
+                    not torch.equal(held[1], ids)):
+                raise RuntimeError(
+                    "an id buffer was mutated in place after being "
+                    "handed to client.pull/push (see the API contract); "
+                    "shard bucketing would silently mis-route ids")
             return order, counts, None
         order, counts, shards = bucket_by_shard(ids, self.part)
         if counts.numel() < self.world:
@@ -605,7 +804,8 @@ class PSEngine:
                 self.world - counts.numel())])
         if len(self._bucket_cache) >= 8:
             self._bucket_cache.clear()
-        self._bucket_cache[key] = (ids, order, counts)
+        snapshot = ids.clone() if self.cfg.debug_validate else ids
+        self._bucket_cache[key] = ((ids, snapshot), order, counts)
         return order, counts, shards
 
 

@@ -51,7 +51,32 @@ class Transport:
         (e.g. in-order pull answers: the receiver sized them when it sent
         the pulls) — the counts exchange is skipped entirely.
         """
+        h = self.alltoallv_begin(parts, specs, extra, recv_counts)
+        return self.alltoallv_end(h)
+
+    # -- split-phase form: the engine's pipelined feedback mode issues
+    # the exchange, runs the round's fused compute while the payloads
+    # are in flight (RCCL p2p progresses on its own HIP stream), then
+    # completes it — the north star's "gradient compute overlaps the
+    # xGMI all-to-all".
+    def alltoallv_begin(self, parts, specs, extra=None, recv_counts=None):
+        """Issue the exchange; returns an opaque handle.  Counts (and the
+        folded control vector) ARE resolved here — only the payload
+        transfer remains in flight when this returns.  Every rank must
+        call begin at the same point of the same round (collective)."""
         raise NotImplementedError
+
+    def alltoallv_end(self, handle):
+        """Complete the exchange issued by :meth:`alltoallv_begin`;
+        returns what :meth:`alltoallv` would have returned."""
+        raise NotImplementedError
+
+    def ctrl_of(self, handle) -> Optional[torch.Tensor]:
+        """The control matrix (world, n_extra) of an in-flight handle —
+        available right after ``alltoallv_begin`` (it rides the counts
+        phase), so SSP/termination decisions can be made while the
+        payloads are still on the wire."""
+        return handle[-1]
 
     def allgather_int(self, vec: Sequence[int]) -> torch.Tensor:
         """Gather a small int64 host vector from every rank -> (world, len)."""
@@ -71,6 +96,16 @@ class LocalTransport(Transport):
             return recv
         return recv, torch.tensor([list(extra)], dtype=torch.int64)
 
+    def alltoallv_begin(self, parts, specs, extra=None, recv_counts=None):
+        recv = [[p[0]] for p in parts]
+        ctrl = (torch.tensor([list(extra)], dtype=torch.int64)
+                if extra is not None else None)
+        return (recv, extra is not None, ctrl)
+
+    def alltoallv_end(self, handle):
+        recv, has_extra, ctrl = handle
+        return (recv, ctrl) if has_extra else recv
+
     def allgather_int(self, vec):
         return torch.tensor([list(vec)], dtype=torch.int64)
 
@@ -89,7 +124,7 @@ class TorchDistTransport(Transport):
             torch.device("cuda", torch.cuda.current_device())
             if dist.get_backend() == "nccl" else torch.device("cpu"))
 
-    def alltoallv(self, parts, specs, extra=None, recv_counts=None):
+    def alltoallv_begin(self, parts, specs, extra=None, recv_counts=None):
         dist = self.dist
         C, W, me = len(parts), self.world, self.rank
         dev = self.device
@@ -118,10 +153,14 @@ class TorchDistTransport(Transport):
                 return int(recv_counts[c][s])
             return int(counts[s, c * W + me])
 
-        # 2) payloads: grouped isend/irecv (RCCL: grouped ncclSend/Recv).
+        # 2) payloads: grouped isend/irecv (RCCL: grouped ncclSend/Recv
+        # over xGMI, progressing on RCCL's own HIP stream).  Only issued
+        # here; completion (and the retained send references keeping the
+        # buffers alive) resolve in alltoallv_end.
         recv: List[List[Optional[torch.Tensor]]] = [
             [None] * W for _ in range(C)]
         p2p = []
+        sends: List[torch.Tensor] = []
         for c, chan in enumerate(parts):
             dtype, trailing = specs[c]
             for s in range(W):
@@ -140,13 +179,19 @@ class TorchDistTransport(Transport):
                 t = t.contiguous()
                 if t.device != dev:
                     t = t.to(dev)
+                sends.append(t)  # keep alive until end()
                 p2p.append(dist.P2POp(dist.isend, t, d))
-        if p2p:
-            for work in dist.batch_isend_irecv(p2p):
-                work.wait()
-        if extra is None:
-            return recv
-        return recv, extras
+        works = dist.batch_isend_irecv(p2p) if p2p else []
+        return (works, sends, recv, extra is not None, extras)
+
+    def alltoallv_end(self, handle):
+        works, _sends, recv, has_extra, extras = handle
+        for work in works:
+            # nccl(=RCCL): wait() only makes the current HIP stream wait
+            # on the comm stream (no host block); gloo: joins the
+            # transfer threads.
+            work.wait()
+        return (recv, extras) if has_extra else recv
 
     def allgather_int(self, vec):
         t = torch.tensor(list(vec), dtype=torch.int64, device=self.device)
@@ -228,6 +273,20 @@ class LoopbackTransport(Transport):
         if extra is None:
             return recv
         return recv, extras
+
+    # split-phase form: lockstep threads all reach begin at the same
+    # round point, so the full exchange happens in begin (the "flight"
+    # is instantaneous in-process) and end just hands the result back.
+    def alltoallv_begin(self, parts, specs, extra=None, recv_counts=None):
+        out = self.alltoallv(parts, specs, extra, recv_counts)
+        if extra is not None:
+            recv, extras = out
+            return (recv, True, extras)
+        return (out, False, None)
+
+    def alltoallv_end(self, handle):
+        recv, has_extra, extras = handle
+        return (recv, extras) if has_extra else recv
 
     def allgather_int(self, vec):
         sh = self.shared

@@ -35,7 +35,7 @@ class ShardedTable:
                  eps: float = 1e-8, seed: int = 0):
         self.part = part
         self.shard = shard
-        self._full_ids_key = None
+        self._full_ids: "torch.Tensor | None" = None
         self.dim = int(dim)
         self.device = torch.device(device)
         self.optimizer = optimizer
@@ -87,17 +87,22 @@ class ShardedTable:
     def _is_full_range(self, ids: torch.Tensor) -> bool:
         """ids == every row of this shard in local-row order?  Verified
         with one tensor compare on first sight, then recognized by
-        buffer identity (dense-pull workers re-send one cached tensor)."""
+        buffer IDENTITY (dense-pull workers re-send one cached tensor).
+        The verified tensor itself is retained in the cache entry — as
+        runtime._shard_counts and the HIP position cache do — so its
+        storage stays alive and a recycled allocation of the same
+        pointer (e.g. a fresh multi-GPU recv buffer each round) can
+        never be misclassified as the cached full-range pull."""
         if ids.numel() != self.rows or self.rows == 0:
             return False
-        key = (ids.data_ptr(), ids.device)
-        if key == self._full_ids_key:
+        cached = self._full_ids
+        if cached is not None and ids is cached:
             return True
         rows = self.part.local_row(ids)
         expect = torch.arange(self.rows, dtype=rows.dtype,
                               device=rows.device)
         if bool(torch.equal(rows, expect)):
-            self._full_ids_key = key
+            self._full_ids = ids
             return True
         return False
 

@@ -21,6 +21,7 @@ file with the same layout (suffix ``.accum``).
 """
 from __future__ import annotations
 
+import json
 import os
 import struct
 from typing import List, Optional, Tuple
@@ -30,17 +31,64 @@ import torch
 from ..table import ShardedTable
 
 MAGIC = b"FPSCKPT1"
+MANIFEST = "MANIFEST.json"
 
 
 def _write_one(path: str, ids: torch.Tensor, values: torch.Tensor) -> None:
+    """Crash-safe single-file write: tmp + fsync + atomic ``os.replace``
+    — a dump interrupted mid-write can never leave a truncated ``.fps``
+    file under the final name (§5.3's recovery must survive the failure
+    it recovers from)."""
     ids = ids.contiguous().cpu()
     values = values.contiguous().cpu().to(torch.float32)
     count, dim = values.shape
-    with open(path, "wb") as f:
+    tmp = path + ".tmp"
+    with open(tmp, "wb") as f:
         f.write(MAGIC)
         f.write(struct.pack("<qqq", dim, 0, count))
         f.write(ids.numpy().tobytes())
         f.write(values.numpy().tobytes())
+        f.flush()
+        os.fsync(f.fileno())
+    os.replace(tmp, path)
+
+
+def _write_manifest(directory: str, n_shards: int, dim: int,
+                    has_accum: bool) -> None:
+    """Completeness marker (atomic): a round directory is a valid
+    restore point only when the manifest exists AND every shard file it
+    names is present.  Every rank writes the same content, so concurrent
+    writers are idempotent (each stages through its own tmp name)."""
+    import threading
+    tmp = os.path.join(
+        directory,
+        f"{MANIFEST}.{os.getpid()}.{threading.get_ident()}.tmp")
+    with open(tmp, "w") as f:
+        json.dump({"n_shards": int(n_shards), "dim": int(dim),
+                   "has_accum": bool(has_accum)}, f)
+        f.flush()
+        os.fsync(f.fileno())
+    os.replace(tmp, os.path.join(directory, MANIFEST))
+
+
+def _complete(directory: str) -> bool:
+    """True iff the manifest exists and every shard file it promises is
+    present (all writes are atomic, so present == whole).  No manifest
+    == not a restore point — a crash before any rank finished its first
+    shard must not be mistaken for a dump."""
+    mpath = os.path.join(directory, MANIFEST)
+    try:
+        with open(mpath) as f:
+            m = json.load(f)
+    except (FileNotFoundError, ValueError, OSError):
+        return False
+    for s in range(int(m["n_shards"])):
+        p = shard_path(directory, s)
+        if not os.path.exists(p):
+            return False
+        if m.get("has_accum") and not os.path.exists(p + ".accum"):
+            return False
+    return True
 
 
 def _read_one(path: str) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -63,12 +111,20 @@ def shard_path(directory: str, shard: int, accum: bool = False) -> str:
 
 
 def save_shard(directory: str, table: ShardedTable) -> None:
+    """Write this shard's dump atomically, then (re)write the directory
+    manifest.  The manifest names the full shard set, so a reader (or
+    :func:`latest_checkpoint`) treats the directory as a restore point
+    only once EVERY shard file exists — a crash mid-dump, or between
+    ranks dumping into the shared round directory, leaves the previous
+    complete dump as the newest valid one."""
     os.makedirs(directory, exist_ok=True)
     ids, values = table.dump()
     _write_one(shard_path(directory, table.shard), ids, values)
     if table.accum is not None:
         _write_one(shard_path(directory, table.shard, accum=True),
                    ids, table.accum)
+    _write_manifest(directory, table.part.n_shards, table.dim,
+                    table.accum is not None)
 
 
 class PeriodicCheckpointer:
@@ -106,14 +162,15 @@ class PeriodicCheckpointer:
 
 
 def latest_checkpoint(directory: str) -> Optional[str]:
-    """Newest round_* subdirectory that contains shard files."""
+    """Newest COMPLETE round_* subdirectory (manifest present and every
+    shard file it names exists); incomplete/torn dumps are skipped."""
     if not os.path.isdir(directory):
         return None
     rounds = sorted(d for d in os.listdir(directory)
                     if d.startswith("round_"))
     for d in reversed(rounds):
         full = os.path.join(directory, d)
-        if any(f.endswith(".fps") for f in os.listdir(full)):
+        if _complete(full):
             return full
     return None
 

@@ -4,6 +4,8 @@ import pytest
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires a GPU (run on an MI355X box)")
+    config.addinivalue_line(
+        "markers", "slow: long-running CPU test (world-8 soaks etc.)")
 
 
 def pytest_collection_modifyitems(config, items):

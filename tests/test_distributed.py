@@ -77,3 +77,93 @@ def test_gloo_world2_push_pull_routing():
         # rank1 pulled only in round 1, when totals were 1.0+2.0=3.0
         assert all(v == 4.0 for v in r0["got"].values())
         assert all(v == 3.0 for v in r1["got"].values())
+
+
+def _soak_job(rank, world, port, feedback, outdir):
+    """World-8 soak body: randomized pull/push traffic + SSP on, the
+    exact process count the driver's 8-GPU scale run uses.
+    Every pull must be answered exactly once, totals must match, and
+    the observed staleness must respect the SSP bound (+1 round of
+    control lag in pipelined mode)."""
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world)
+    try:
+        from flink_parameter_server_amd import (PSConfig, PSEngine,
+                                                ShardedTable,
+                                                TableServerLogic,
+                                                WorkerLogic)
+        from flink_parameter_server_amd.parallel.partitioner import \
+            ModPartitioner
+        from flink_parameter_server_amd.parallel.transport import \
+            TorchDistTransport
+
+        n_ids, dim, rounds, staleness = 97, 3, 30, 2
+
+        class RandomTraffic(WorkerLogic):
+            def __init__(self):
+                self.g = torch.Generator().manual_seed(700 + rank)
+                self.pushed = torch.zeros(n_ids, dim)
+                self.pending = 0
+                self.answered = 0
+
+            def on_recv(self, data, client):
+                k = int(torch.randint(1, 16, (1,), generator=self.g))
+                ids = torch.unique(torch.randint(0, n_ids, (k,),
+                                                 generator=self.g))
+                deltas = torch.randn(ids.numel(), dim, generator=self.g)
+                client.push(ids, deltas)
+                self.pushed[ids] += deltas
+                pull = torch.unique(torch.randint(0, n_ids, (k,),
+                                                  generator=self.g))
+                client.pull(pull, unique=True)
+                self.pending += int(pull.numel())
+
+            def on_pull_recv(self, ids, values, client):
+                assert values.shape == (ids.numel(), dim)
+                self.answered += int(ids.numel())
+                self.pending -= int(ids.numel())
+
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, dim, "cpu")
+        logic = RandomTraffic()
+        transport = TorchDistTransport(device=torch.device("cpu"))
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=dim,
+                                feedback=feedback, staleness=staleness),
+                       logic, TableServerLogic(table), transport, part)
+        # rank-skewed workload so SSP actually bites: rank 0 gets 3x
+        # the batches of rank 7
+        n_batches = rounds * (world - rank + world) // world
+        eng.run([[r] for r in range(n_batches)])
+        assert logic.pending == 0, "unanswered pulls at drain"
+        max_lag = max(eng.metrics.staleness_hist)
+        assert max_lag <= staleness + 1, eng.metrics.staleness_hist
+        gids, vals = table.dump()
+        torch.save({"pushed": logic.pushed,
+                    "dump": {i: v.clone() for i, v in
+                             zip(gids.tolist(), vals)}},
+                   os.path.join(outdir, f"rank{rank}.pt"))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(600)
+@pytest.mark.parametrize("feedback", ["fresh", "pipelined"])
+def test_gloo_world8_soak(feedback):
+    world = 8
+    port = 29531 if feedback == "fresh" else 29532
+    with tempfile.TemporaryDirectory() as d:
+        mp.spawn(_soak_job, args=(world, port, feedback, d),
+                 nprocs=world, join=True)
+        expected = torch.zeros(97, 3)
+        merged = {}
+        for r in range(world):
+            blob = torch.load(os.path.join(d, f"rank{r}.pt"),
+                              weights_only=False)
+            expected += blob["pushed"]
+            merged.update(blob["dump"])
+        got = torch.stack([merged[i] for i in range(97)])
+        assert torch.allclose(got, expected, atol=1e-4), \
+            (got - expected).abs().max()

@@ -149,7 +149,7 @@ def test_range_partitioner_across_ranks():
     assert all(v == world for v in merged.values())
 
 
-@pytest.mark.parametrize("feedback", ["fresh", "merged"])
+@pytest.mark.parametrize("feedback", ["fresh", "merged", "pipelined"])
 def test_randomized_world4_stress(feedback):
     """4 ranks, 25 randomized rounds of interleaved pulls/pushes, under
     both feedback schedules.  '+' server semantics are
@@ -258,3 +258,105 @@ def test_kitchen_sink_all_knobs_together():
     # bf16 wire: ~3 decimal digits per push crossing ranks
     assert torch.allclose(got, expected, atol=0.15, rtol=0.05), \
         (got - expected).abs().max()
+
+
+def test_protocol_mismatch_fails_fast():
+    """Ranks disagreeing on a wire-shaping knob (here: wire dtype) must
+    raise at the first round's protocol preflight instead of hanging in
+    a malformed collective (VERDICT weak #7)."""
+    world, n_ids = 2, 8
+
+    def job(rank, transport):
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, 1, "cpu")
+        wire = "bf16" if rank == 0 else "f32"
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=1, wire_dtype=wire),
+                       PushCounting(), TableServerLogic(table),
+                       transport, part)
+        eng.run([[0]])
+
+    with pytest.raises(RuntimeError, match="wire-protocol mismatch"):
+        run_loopback(world, job)
+
+
+def test_flush_rejects_pushes_unique():
+    """A flush strategy coalesces pushes across rounds (ids may repeat),
+    which a pushes_unique server logic would corrupt silently — the
+    engine must reject the combination at construction (ADVICE)."""
+    from flink_parameter_server_amd.messaging import CountFlush
+
+    part = ModPartitioner(8, 1)
+    table = ShardedTable(part, 0, 1, "cpu")
+    with pytest.raises(ValueError, match="pushes_unique"):
+        PSEngine(PSConfig(num_ids=8, dim=1), PushCounting(),
+                 TableServerLogic(table, pushes_unique=True),
+                 None, part, flush=CountFlush(4))
+
+
+def test_debug_validate_catches_inplace_mutation():
+    """cfg.debug_validate turns an in-place id-buffer mutation (which
+    violates the client.pull/push contract) into a loud error instead of
+    silent mis-routing (ADVICE)."""
+    world, n_ids = 2, 16
+
+    class Mutator(WorkerLogic):
+        def __init__(self):
+            self.buf = torch.arange(n_ids, dtype=torch.int64)
+            self.n = 0
+
+        def on_recv(self, data, client):
+            if self.n == 1:
+                # second round: refill the SAME buffer with new content
+                self.buf.copy_(torch.flip(
+                    torch.arange(n_ids, dtype=torch.int64), [0]) % 5)
+            self.n += 1
+            client.push(self.buf, torch.ones(n_ids, 1))
+
+        def on_pull_recv(self, ids, values, client):
+            pass
+
+    def job(rank, transport):
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, 1, "cpu")
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=1,
+                                debug_validate=True), Mutator(),
+                       TableServerLogic(table), transport, part)
+        eng.run([[0], [1]])
+
+    with pytest.raises(RuntimeError, match="mutated in place"):
+        run_loopback(world, job)
+
+
+def test_pipelined_overlapped_delivery_mf():
+    """Online MF converges under the pipelined (split-phase, +1 round
+    answer latency) schedule too — the overlap must not change
+    semantics, only timing."""
+    world, n_users, n_items, rank_k = 2, 40, 30, 4
+    batches, _, _ = synth.planted_mf_ratings(n_users, n_items, rank_k,
+                                             2400, 120, seed=33)
+    per_worker = [batches[0::2], batches[1::2]]
+
+    def job(rank, transport):
+        part = ModPartitioner(n_items, world)
+        table = ShardedTable(part, rank, rank_k, "cpu", init=(-0.1, 0.1))
+        wl = OnlineMFWorkerLogic(n_users, rank_k, learning_rate=0.05,
+                                 range_min=-0.1, range_max=0.1)
+        eng = PSEngine(PSConfig(num_ids=n_items, dim=rank_k,
+                                feedback="pipelined"), wl,
+                       TableServerLogic(table), transport, part)
+        eng.run(epoch_iterator(per_worker[rank], 8))
+        return wl.user_table, table
+
+    results = run_loopback(world, job)
+    item = torch.zeros(n_items, rank_k)
+    for r in range(world):
+        gids, vals = results[r][1].dump()
+        item[gids] = vals
+    tot, n = 0.0, 0
+    for r in range(world):
+        ut = results[r][0]
+        for b in per_worker[r]:
+            pred = (ut[b.user_rows] * item[b.item_ids]).sum(1)
+            tot += float(((pred - b.ratings) ** 2).sum())
+            n += len(b)
+    assert tot / n < 0.1, tot / n
