@@ -51,6 +51,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <climits>
 #include <cstdlib>
 #include <map>
 #include <tuple>
@@ -868,16 +869,23 @@ __global__ __launch_bounds__(256) void k_sparse_linear_step(
 }
 
 // ---------------------------------------------------------------------
-// K7 (multiclass): one wave per instance; lane c owns class c
-// (n_classes <= 64).  scores[c] = sum_j w[slot_j][c] * x_j  — lanes read
-// consecutive class columns of each weight row (coalesced).
+// K7 (multiclass): one wave per instance; class columns are walked in
+// tiles of 64 (lane c of tile t owns class t*64+c — any n_classes).
+// scores[c] = sum_j w[slot_j][c] * x_j — lanes read consecutive class
+// columns of each weight row (coalesced).  The argmax tie-break is
+// deterministic (smallest class index wins), and rep_stride > 0
+// spreads the delta scatter over 8 replica copies (zipf-hot features
+// otherwise serialize same-address atomics — same mitigation as the
+// binary kernel).
 // ---------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_pa_multiclass_step(
     const int64_t* __restrict__ indptr,
     const int64_t* __restrict__ feat_slots,
     const float* __restrict__ x_vals, const int64_t* __restrict__ labels,
     const float* __restrict__ weights, float* __restrict__ scores,
-    float* __restrict__ deltas, int64_t B, int64_t nc, int mode, float C) {
+    float* __restrict__ deltas, int64_t B, int64_t nc, int mode, float C,
+    int64_t rep_stride) {
+  const int64_t rep_off = (int64_t)(blockIdx.x & 7) * rep_stride;
   const int lane = threadIdx.x % WAVE;
   const int64_t waves_per_block = blockDim.x / WAVE;
   const int64_t w0 = (int64_t)blockIdx.x * waves_per_block +
@@ -885,26 +893,38 @@ __global__ __launch_bounds__(256) void k_pa_multiclass_step(
   const int64_t wstride = (int64_t)gridDim.x * waves_per_block;
   for (int64_t i = w0; i < B; i += wstride) {
     const int64_t lo = indptr[i], hi = indptr[i + 1];
-    float sc = 0.f, n2 = 0.f;
-    for (int64_t j = lo; j < hi; ++j) {
+    float n2 = 0.f;
+    for (int64_t j = lo + lane; j < hi; j += WAVE) {
       float x = x_vals[j];
-      if (lane < nc) sc += weights[feat_slots[j] * nc + lane] * x;
-      if (lane == 0) n2 += x * x;
+      n2 += x * x;
     }
-    n2 = __shfl(n2, 0);
-    if (lane < nc) scores[i * nc + lane] = sc;
+    n2 = group_sum<WAVE>(n2);
     const int64_t y = labels[i];
-    if (y < 0) continue;  // unlabeled
-    // max over wrong classes + the true-class score, via wave reduction
-    float sy = __shfl(sc, (int)y);
-    float wrong = (lane < nc && lane != y) ? sc : -INFINITY;
-    int ridx = lane;
+    float sy = 0.f, wrong = -INFINITY;
+    int ridx = INT_MAX;
+    for (int64_t c0 = 0; c0 < nc; c0 += WAVE) {
+      const int64_t c = c0 + lane;
+      float sc = 0.f;
+      if (c < nc) {
+        for (int64_t j = lo; j < hi; ++j)
+          sc += weights[feat_slots[j] * nc + c] * x_vals[j];
+        scores[i * nc + c] = sc;
+      }
+      if (y < 0) continue;  // unlabeled: scores only
+      if (c0 <= y && y < c0 + WAVE) sy = __shfl(sc, (int)(y - c0));
+      float w_t = (c < nc && c != y) ? sc : -INFINITY;
+      int r_t = (c < nc) ? (int)c : INT_MAX;
 #pragma unroll
-    for (int off = WAVE / 2; off > 0; off >>= 1) {
-      float o = __shfl_xor(wrong, off);
-      int oi = __shfl_xor(ridx, off);
-      if (o > wrong) { wrong = o; ridx = oi; }
+      for (int off = WAVE / 2; off > 0; off >>= 1) {
+        float o = __shfl_xor(w_t, off);
+        int oi = __shfl_xor(r_t, off);
+        if (o > w_t || (o == w_t && oi < r_t)) { w_t = o; r_t = oi; }
+      }
+      if (w_t > wrong || (w_t == wrong && r_t < ridx)) {
+        wrong = w_t; ridx = r_t;
+      }
     }
+    if (y < 0) continue;
     float loss = 1.0f - sy + wrong;
     if (loss <= 0.f) continue;
     float sn = 2.0f * n2;
@@ -915,8 +935,8 @@ __global__ __launch_bounds__(256) void k_pa_multiclass_step(
     else tau = loss / (sn + 1.0f / (2.0f * C));
     for (int64_t j = lo + lane; j < hi; j += WAVE) {
       float t = tau * x_vals[j];
-      atomicAdd(&deltas[feat_slots[j] * nc + y], t);
-      atomicAdd(&deltas[feat_slots[j] * nc + ridx], -t);
+      atomicAdd(&deltas[rep_off + feat_slots[j] * nc + y], t);
+      atomicAdd(&deltas[rep_off + feat_slots[j] * nc + ridx], -t);
     }
   }
 }
@@ -1185,13 +1205,59 @@ static void by_group4(int64_t dim4, F f) {
   }
   else if (dim4 <= 48) f(integral_constant<int, 16>{}, integral_constant<int, 3>{});
   else if (dim4 <= 80) f(integral_constant<int, 16>{}, integral_constant<int, 5>{});
-  else { TORCH_CHECK(dim4 <= 128, "row dim > 512 unsupported");
-         f(integral_constant<int, 32>{}, integral_constant<int, 4>{}); }
+  else if (dim4 <= 128) f(integral_constant<int, 32>{}, integral_constant<int, 4>{});
+  else if (dim4 <= 256) f(integral_constant<int, 64>{}, integral_constant<int, 4>{});
+  else { TORCH_CHECK(dim4 <= 512, "internal: dim4 > 512 must route to "
+                     "the k_seg_fused_huge envelope path");
+         f(integral_constant<int, 64>{}, integral_constant<int, 8>{}); }
+}
+
+// ---------------------------------------------------------------------
+// Envelope walk for rows the register-tiled walks cannot hold (dim >
+// 2048, or dim % 4 != 0): one wave per record, lanes stride the
+// columns, per-record atomicAdd scatter (no segment pre-aggregation —
+// correct for any dim; hot keys serialize, which is acceptable on this
+// envelope path and documented in ops/__init__.py).  SELF adds the
+// analytic a*keyed term per occurrence (sums to a*cnt*keyed per
+// segment — same math as the tiled walks' flush).
+// ---------------------------------------------------------------------
+template <int MODE, bool SELF>
+__global__ __launch_bounds__(256) void k_seg_fused_huge(
+    float* __restrict__ out, int4* __restrict__ rec,
+    const float* __restrict__ keyed, const float* __restrict__ other,
+    int64_t n, int64_t dim, float scale, float a, float lrc) {
+  const int lane = threadIdx.x % WAVE;
+  const int64_t wpb = blockDim.x / WAVE;
+  for (int64_t j = (int64_t)blockIdx.x * wpb + threadIdx.x / WAVE;
+       j < n; j += (int64_t)gridDim.x * wpb) {
+    const int4 rj = rec[j];
+    const int64_t kr = (int64_t)rj.x * dim, sr = (int64_t)rj.y * dim;
+    float g;
+    if (MODE == 2) {
+      g = __int_as_float(rj.z);
+    } else {
+      float dot = 0.f;
+      for (int64_t c = lane; c < dim; c += WAVE)
+        dot += keyed[kr + c] * other[sr + c];
+      dot = group_sum<WAVE>(dot);
+      g = (MODE == 0)
+              ? (__int_as_float(rj.z) - dot)
+              : lrc * (__int_as_float(rj.z) -
+                       1.0f / (1.0f + __expf(-dot)));
+      if (lane == 0) rec[j].z = __float_as_int(g);
+    }
+    for (int64_t c = lane; c < dim; c += WAVE) {
+      float t = scale * g * other[sr + c];
+      if (SELF) t += a * keyed[kr + c];
+      atomicAdd(&out[kr + c], t);
+    }
+  }
 }
 
 // fused segment-walk launcher.  ZERO_OUT: `out` is zero-initialized and
 // exclusively written by this launch, so exclusive-segment flushes may
-// store instead of read-add.
+// store instead of read-add.  ``dim4 == 0`` (caller signal: dim % 4 !=
+// 0 or dim > 2048) routes to the per-record atomic envelope walk.
 template <int MODE, bool SELF, bool ZERO_OUT>
 static void seg_fused(torch::Tensor out, torch::Tensor rec,
                       const torch::Tensor& keyed,
@@ -1200,6 +1266,16 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                       int64_t rep_stride = 0, int rep_mask = 7) {
   int64_t n = rec.size(0);
   if (!n) return;
+  if (dim4 == 0) {
+    int64_t dim = keyed.size(1);
+    hipLaunchKernelGGL((k_seg_fused_huge<MODE, SELF>),
+                       dim3(grid_for(n * WAVE, 256)), dim3(256), 0,
+                       cur_stream(), out.data_ptr<float>(),
+                       (int4*)rec.data_ptr<int>(),
+                       keyed.data_ptr<float>(), other.data_ptr<float>(),
+                       n, dim, (float)scale, (float)a, (float)lrc);
+    return;
+  }
   int chunk = seg_chunk();
   int64_t nchunks = (n + chunk - 1) / chunk;
   static int use_lds = env_int("FPS_SEG_LDS", 1);
@@ -1280,9 +1356,10 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   int64_t n = ratings.numel(), dim = user_table.size(1);
   int64_t m = item_vecs.size(0), U = user_table.size(0);
   TORCH_CHECK(item_vecs.size(1) == dim);
-  TORCH_CHECK(dim % 4 == 0, "mf_sgd_step requires dim % 4 == 0");
   if (!n) return torch::zeros_like(item_vecs);
-  int64_t dim4 = dim / 4;
+  // register-tiled walks cover dim % 4 == 0 up to 2048; anything else
+  // takes the per-record atomic envelope walk (dim4 = 0 signal)
+  int64_t dim4 = (dim % 4 == 0 && dim <= 2048) ? dim / 4 : 0;
   if (item_slots.dtype() != user_rows.dtype()) {
     // harmonize to int32 (both ranges are table heights < 2^31)
     if (item_slots.dtype() != torch::kInt32)
@@ -1370,7 +1447,9 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   }
 
   // pass A: recs by item slot; fused e + item-delta accumulation
-  int64_t reps = delta_reps(n, m, dim);
+  // (the dim4==0 envelope walk scatters with per-record atomics and
+  // has no replica spreading)
+  int64_t reps = dim4 ? delta_reps(n, m, dim) : 1;
   torch::Tensor deltas;
   if (reps > 1) {
     auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
@@ -1463,10 +1542,9 @@ torch::Tensor sgns_step(torch::Tensor vecs, torch::Tensor center_slots,
   CHECK_DEV(vecs); CHECK_CONT(vecs);
   CHECK_CONT(center_slots); CHECK_CONT(ctx_slots); CHECK_CONT(labels);
   int64_t n = labels.numel(), dim = vecs.size(1), m = vecs.size(0);
-  TORCH_CHECK(dim % 4 == 0, "sgns_step requires dim % 4 == 0");
   auto deltas = torch::zeros_like(vecs);
   if (!n) return deltas;
-  int64_t dim4 = dim / 4;
+  int64_t dim4 = (dim % 4 == 0 && dim <= 2048) ? dim / 4 : 0;
   auto recC = build_rec(center_slots, ctx_slots, labels, m);
   seg_fused<1, false, true>(deltas, recC, vecs, vecs, dim4, 1.0, 0.0,
                             lr);
@@ -1565,18 +1643,24 @@ std::vector<torch::Tensor> pa_multiclass_step(
     double C) {
   CHECK_DEV(weights); CHECK_CONT(weights);
   int64_t B = indptr.numel() - 1, nc = weights.size(1);
-  TORCH_CHECK(nc <= 64, "n_classes > 64 unsupported");
   auto scores = torch::zeros({B, nc}, x_vals.options());
-  auto deltas = torch::zeros_like(weights);
-  if (!B) return {scores, deltas};
+  if (!B) return {scores, torch::zeros_like(weights)};
+  // replica-spread the delta scatter like the binary kernel when the
+  // batch is dense enough to make hot-feature atomics the bottleneck
+  int64_t mw = weights.numel();
+  int64_t reps = (x_vals.numel() * nc >= 8 * mw &&
+                  8 * mw * 4 <= (256ll << 20)) ? 8 : 1;
+  auto dbuf = torch::zeros({reps * mw}, weights.options());
   hipLaunchKernelGGL(k_pa_multiclass_step, dim3(grid_for(B * WAVE, 256)),
                      dim3(256), 0, cur_stream(),
                      indptr.data_ptr<int64_t>(),
                      feat_slots.data_ptr<int64_t>(),
                      x_vals.data_ptr<float>(), labels.data_ptr<int64_t>(),
                      weights.data_ptr<float>(), scores.data_ptr<float>(),
-                     deltas.data_ptr<float>(), B, nc, pa_mode(variant),
-                     (float)C);
+                     dbuf.data_ptr<float>(), B, nc, pa_mode(variant),
+                     (float)C, reps > 1 ? mw : 0);
+  auto deltas = (reps > 1 ? dbuf.view({reps, mw}).sum(0) : dbuf)
+                    .view_as(weights);
   return {scores, deltas};
 }
 

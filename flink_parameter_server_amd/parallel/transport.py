@@ -123,11 +123,18 @@ class TorchDistTransport(Transport):
         self.device = torch.device(device) if device is not None else (
             torch.device("cuda", torch.cuda.current_device())
             if dist.get_backend() == "nccl" else torch.device("cpu"))
+        # gloo cannot move device tensors: stage through CPU on the way
+        # out and deliver on self.device on the way in — the same
+        # delivery contract as nccl/RCCL (recv segments arrive
+        # on-device), so the engine's device paths are exercised by the
+        # CPU-backend integration tests too.
+        self.stage_via_cpu = (self.device.type == "cuda" and
+                              dist.get_backend() != "nccl")
 
     def alltoallv_begin(self, parts, specs, extra=None, recv_counts=None):
         dist = self.dist
         C, W, me = len(parts), self.world, self.rank
-        dev = self.device
+        dev = torch.device("cpu") if self.stage_via_cpu else self.device
 
         # 1) counts: one all_gather of int64[C*W (+extra)] covers every
         # channel — skipped entirely when the caller knows its receive
@@ -191,10 +198,18 @@ class TorchDistTransport(Transport):
             # on the comm stream (no host block); gloo: joins the
             # transfer threads.
             work.wait()
+        if self.stage_via_cpu:
+            # deliver on-device, matching the nccl contract (self-
+            # addressed segments short-circuited and are already there)
+            for chan in recv:
+                for i, t in enumerate(chan):
+                    if t is not None and not t.is_cuda:
+                        chan[i] = t.to(self.device, non_blocking=True)
         return (recv, extras) if has_extra else recv
 
     def allgather_int(self, vec):
-        t = torch.tensor(list(vec), dtype=torch.int64, device=self.device)
+        dev = torch.device("cpu") if self.stage_via_cpu else self.device
+        t = torch.tensor(list(vec), dtype=torch.int64, device=dev)
         out = [torch.empty_like(t) for _ in range(self.world)]
         self.dist.all_gather(out, t)
         return torch.stack([o.cpu() for o in out])

