@@ -72,10 +72,12 @@ def make_mf(args, rank, world, device):
     # per-hot-key step stays constant as the batch grows (stability is
     # set by lr x key-multiplicity; bench.py asserts finiteness)
     mf_lr = 1e-4 * min(1.0, (1 << 20) / args.batch)
+    dense = {"auto": None, "dense": True, "sparse": False}[args.pull_mode]
     wl = OnlineMFWorkerLogic(users_local, k, learning_rate=mf_lr,
                              reg=0.001, range_min=-0.01, range_max=0.01,
                              n_items=n_items, device=device,
-                             assume_static_batches=True, seed=rank + 1)
+                             assume_static_batches=True,
+                             dense_pull=dense, seed=rank + 1)
     g = torch.Generator().manual_seed(1234 + rank)
     batches = []
     for _ in range(args.n_batches):
@@ -97,7 +99,8 @@ def make_mf(args, rank, world, device):
              and device.type == "cuda" else "MovieLens-1M-shaped")
     meta = {"model": f"SGD-MF rank{k} {shape}" +
                      (f" x{args.scale:g}" if args.scale != 1 else ""),
-            "n_users": n_users, "n_items": n_items, "factors": k}
+            "n_users": n_users, "n_items": n_items, "factors": k,
+            "pull_mode": args.pull_mode}
     return cfg, part, table, wl, batches, meta
 
 
@@ -160,6 +163,13 @@ def main():
     ap.add_argument("--model", choices=list(MAKERS), default="mf32")
     ap.add_argument("--scale", type=float, default=1.0,
                     help="scale factor on the MovieLens-1M shape")
+    ap.add_argument("--pull-mode", choices=["auto", "dense", "sparse"],
+                    default="auto",
+                    help="mf item-pull policy: dense = whole item range "
+                         "per round; sparse = per-batch coalesced pulls "
+                         "(exercises the real pull/push machinery — the "
+                         "HBM-regime co-headline mode); auto = by item-"
+                         "space size")
     ap.add_argument("--staleness", type=int, default=None)
     ap.add_argument("--lr-features", type=int, default=10**9,
                     help="feature count for --model lr (the configs[3] "

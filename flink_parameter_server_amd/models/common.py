@@ -108,8 +108,9 @@ class DenseCoalescer:
         bytes of the sort/pack passes)."""
         present = self.present
         present.zero_()
-        present[ids] = True
+        idx = ids if ids.dtype == torch.int64 else ids.long()
+        present[idx] = True
         lookup = torch.cumsum(present, 0, dtype=torch.int32)
         uids = present.nonzero().reshape(-1)
-        slots = lookup[ids] - 1
+        slots = lookup[idx] - 1
         return uids, slots

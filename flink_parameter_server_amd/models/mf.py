@@ -86,6 +86,7 @@ class OnlineMFWorkerLogic(WorkerLogic):
                  output_user_factors_on_close: bool = False,
                  assume_static_batches: bool = False,
                  aggregate: str = "sum",
+                 dense_pull: Optional[bool] = None,
                  seed: int = 1):
         """``assume_static_batches=True``: the data source re-yields the
         SAME batch objects with unchanged index tensors (epoch sweeps
@@ -137,7 +138,13 @@ class OnlineMFWorkerLogic(WorkerLogic):
         # * torch.unique fallback for unbounded id spaces.
         self._coalescer: Optional[DenseCoalescer] = None
         self._all_items: Optional[torch.Tensor] = None
-        self._dense_pull = 0 < self.n_items <= self.DENSE_PULL_MAX
+        # dense_pull=False forces per-batch sparse pulls (the
+        # coalescer/unique path) even for small item spaces — the honest
+        # pull/push-ops regime the HBM co-headline benchmarks measure;
+        # None = auto by item-space size.
+        self._dense_pull = (0 < self.n_items <= self.DENSE_PULL_MAX
+                            if dense_pull is None else
+                            bool(dense_pull and self.n_items > 0))
         if (not self._dense_pull and
                 0 < self.n_items <= DenseCoalescer.MAX_SPACE):
             self._coalescer = DenseCoalescer(self.n_items, self.device)

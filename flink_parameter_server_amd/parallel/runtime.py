@@ -63,6 +63,15 @@ class Metrics:
     ssp_stalls: int = 0
     bytes_sent: int = 0
     held_rounds: int = 0
+    # pipelined-mode host phase timers (seconds): time issuing the
+    # exchange, computing under it, waiting for its completion, and in
+    # the server phase.  t_comm_wait ~ 0 means the overlapped compute
+    # fully hid the transfer (on nccl the wait is a stream dependency,
+    # so these measure host-side enqueue/wait, not kernel execution).
+    t_issue: float = 0.0
+    t_overlap_compute: float = 0.0
+    t_comm_wait: float = 0.0
+    t_server: float = 0.0
     # staleness distribution (clock - min cross-worker clock, per round) —
     # the SSP staleness histogram BASELINE configs[4] reports.
     staleness_hist: Dict[int, int] = field(default_factory=dict)
@@ -576,9 +585,13 @@ class PSEngine:
         # 2. ISSUE the exchange.  Control rides the counts phase, so it
         # is fully resolved when begin returns; only payloads remain in
         # flight.
+        import time as _time
+        t0 = _time.perf_counter()
         h = self.transport.alltoallv_begin(
             chans, specs, extra=[self._clock, worker_done, int(exhausted)])
         ctrl = self.transport.ctrl_of(h)
+        t1 = _time.perf_counter()
+        m.t_issue += t1 - t0
 
         # 3. OVERLAPPED compute: deliver the held answers (the fused
         # training step launches on the compute stream while the p2p
@@ -616,7 +629,11 @@ class PSEngine:
 
         # 4. COMPLETE the exchange; hold the received answers for the
         # next round's overlapped delivery.
+        t2 = _time.perf_counter()
+        m.t_overlap_compute += t2 - t1
         recv, _ = self.transport.alltoallv_end(h)
+        t3 = _time.perf_counter()
+        m.t_comm_wait += t3 - t2
         recv = self._to_device(recv)
         self._unwire(recv[2])
         self._unwire(recv[-1])
@@ -645,6 +662,7 @@ class PSEngine:
 
         if self.checkpointer is not None:
             self.checkpointer.maybe(m.rounds)
+        m.t_server += _time.perf_counter() - t3
 
         done = bool(ctrl[:, 1].all())
         return exhausted, done
