@@ -130,6 +130,17 @@ class TorchDistTransport(Transport):
         # CPU-backend integration tests too.
         self.stage_via_cpu = (self.device.type == "cuda" and
                               dist.get_backend() != "nccl")
+        # Counts + control lane: on the nccl(=RCCL) backend a per-round
+        # device all_gather + .cpu() costs a GPU round-trip per round
+        # and tangles the control word with the compute stream.  A
+        # dedicated host-side gloo group carries the few-hundred-byte
+        # counts/control vector instead: pure host exchange, zero GPU
+        # involvement, and the payload p2p posts the moment the host
+        # has the counts.  (Collective: every rank constructs the
+        # transport, so every rank reaches new_group.)
+        self._ctrl_pg = None
+        if dist.get_backend() == "nccl" and self.world > 1:
+            self._ctrl_pg = dist.new_group(backend="gloo")
 
     def alltoallv_begin(self, parts, specs, extra=None, recv_counts=None):
         dist = self.dist
@@ -149,10 +160,18 @@ class TorchDistTransport(Transport):
             if ne:
                 send_counts[C * W:] = torch.tensor(list(extra),
                                                    dtype=torch.int64)
-            gdev = send_counts.to(dev)
-            gathered = [torch.empty_like(gdev) for _ in range(W)]
-            dist.all_gather(gathered, gdev)
-            full = torch.stack([g.cpu() for g in gathered])  # (W, C*W+ne)
+            if self._ctrl_pg is not None:
+                # host-side lane: no GPU round-trip for counts/control
+                gathered = [torch.empty_like(send_counts)
+                            for _ in range(W)]
+                dist.all_gather(gathered, send_counts,
+                                group=self._ctrl_pg)
+                full = torch.stack(gathered)  # (W, C*W+ne)
+            else:
+                gdev = send_counts.to(dev)
+                gathered = [torch.empty_like(gdev) for _ in range(W)]
+                dist.all_gather(gathered, gdev)
+                full = torch.stack([g.cpu() for g in gathered])
             counts, extras = full[:, :C * W], full[:, C * W:]
 
         def n_from(s, c):
@@ -208,6 +227,11 @@ class TorchDistTransport(Transport):
         return (recv, extras) if has_extra else recv
 
     def allgather_int(self, vec):
+        if self._ctrl_pg is not None:
+            t = torch.tensor(list(vec), dtype=torch.int64)
+            out = [torch.empty_like(t) for _ in range(self.world)]
+            self.dist.all_gather(out, t, group=self._ctrl_pg)
+            return torch.stack(out)
         dev = torch.device("cpu") if self.stage_via_cpu else self.device
         t = torch.tensor(list(vec), dtype=torch.int64, device=dev)
         out = [torch.empty_like(t) for _ in range(self.world)]
