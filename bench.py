@@ -110,8 +110,11 @@ def make_sgns(args, rank, world, device):
     part = ModPartitioner(2 * V, world)
     table = ShardedTable(part, rank, dim, device, init=(-0.01, 0.01))
     sg_lr = 0.025 * min(1.0, (1 << 17) / args.batch)
+    # world=1: whole-table pull aliases the single shard's slab (free),
+    # killing the per-batch coalescer + answer gather; sharded runs pull
+    # sparse (shipping whole shards over xGMI would be wrong)
     wl = SGNSWorkerLogic(V, dim, learning_rate=sg_lr, negatives=5,
-                         seed=rank + 3)
+                         dense_pull=(world == 1), seed=rank + 3)
     g = torch.Generator().manual_seed(99 + rank)
     batches = []
     for _ in range(args.n_batches):

@@ -813,6 +813,37 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
 }
 
 // ---------------------------------------------------------------------
+// K5/K8 prologue: fused SGNS pair expansion — one pass builds the
+// (1 positive + k hash-RNG negatives) pair rows, context table ids
+// (+V offset) and labels that the Python path assembled with ~10
+// elementwise kernels (arange/splitmix/abs/mod/views/fills).  Negative
+// id sequence is bit-identical to models/sgns.py's reference path:
+// key = p*k + (j-1) + base_key;  neg = |splitmix64(key)| % V.
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_sgns_expand(
+    const int64_t* __restrict__ centers,
+    const int64_t* __restrict__ contexts, int64_t n, int64_t k,
+    int64_t V, int64_t base_key, int64_t* __restrict__ out_centers,
+    int64_t* __restrict__ out_ctx, float* __restrict__ out_labels) {
+  const int64_t total = n * (k + 1);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t p = i / (k + 1), j = i - p * (k + 1);
+    out_centers[i] = centers[p];
+    if (j == 0) {
+      out_ctx[i] = contexts[p] + V;
+      out_labels[i] = 1.0f;
+    } else {
+      int64_t h = (int64_t)splitmix64(
+          (uint64_t)(p * k + (j - 1) + base_key));
+      h = h < 0 ? -h : h;
+      out_ctx[i] = h % V + V;
+      out_labels[i] = 0.0f;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
 // K7 (binary) / K9: PA and LR share the sparse-margin skeleton.
 // One wave per instance; lanes stride the instance's nnz.
 // mode: 0 = PA, 1 = PA-I, 2 = PA-II, 3 = logistic regression.
@@ -1570,6 +1601,27 @@ torch::Tensor sgns_step(torch::Tensor vecs, torch::Tensor center_slots,
   return deltas;
 }
 
+std::vector<torch::Tensor> sgns_expand(torch::Tensor centers,
+                                       torch::Tensor contexts, int64_t k,
+                                       int64_t V, int64_t base_key) {
+  CHECK_DEV(centers); CHECK_CONT(centers); CHECK_CONT(contexts);
+  TORCH_CHECK(centers.dtype() == torch::kInt64 &&
+              contexts.dtype() == torch::kInt64);
+  int64_t n = centers.numel();
+  auto opts = centers.options();
+  auto oc = torch::empty({n * (k + 1)}, opts);
+  auto ox = torch::empty({n * (k + 1)}, opts);
+  auto ol = torch::empty({n * (k + 1)}, opts.dtype(torch::kFloat32));
+  if (!n) return {oc, ox, ol};
+  hipLaunchKernelGGL(k_sgns_expand, dim3(grid_for(n * (k + 1), 256)),
+                     dim3(256), 0, cur_stream(),
+                     centers.data_ptr<int64_t>(),
+                     contexts.data_ptr<int64_t>(), n, k, V, base_key,
+                     oc.data_ptr<int64_t>(), ox.data_ptr<int64_t>(),
+                     ol.data_ptr<float>());
+  return {oc, ox, ol};
+}
+
 static int pa_mode(const std::string& variant) {
   if (variant == "PA") return 0;
   if (variant == "PA-I") return 1;
@@ -1679,6 +1731,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("reg") = 0.0, py::arg("reuse_tag") = 0,
         py::arg("mean_users") = false);
   m.def("sgns_step", &sgns_step, "K8 fused SGNS step");
+  m.def("sgns_expand", &sgns_expand,
+        "K5/K8 fused pair expansion + hash-RNG negative sampling");
   m.def("pa_step", &pa_step, "K7 PA binary step");
   m.def("pa_multiclass_step", &pa_multiclass_step, "K7 PA multiclass");
   m.def("lr_step", &lr_step, "K9 sparse logistic step");

@@ -38,17 +38,27 @@ class SGNSBatch:
 class SGNSWorkerLogic(WorkerLogic):
     def __init__(self, vocab_size: int, dim: int, learning_rate: float,
                  negatives: int = 5,
-                 pull_limit: Optional[int] = None, seed: int = 11):
+                 pull_limit: Optional[int] = None,
+                 dense_pull: bool = False, seed: int = 11):
+        """``dense_pull=True``: pull the WHOLE 2V-row table every round
+        (slots are then the raw table ids — the per-batch coalescing
+        pass and the answer gather disappear; with one shard the answer
+        aliases the table slab, so the pull is free).  Same policy as
+        the MF worker's dense-pull mode; right for world=1 or small
+        vocabularies, wrong for a big table sharded across ranks (the
+        answer payload would ship whole shards over xGMI every round)."""
         self.V = int(vocab_size)
         self.dim = int(dim)
         self.lr = float(learning_rate)
         self.neg = int(negatives)
         self.pull_limit = pull_limit
+        self.dense_pull = bool(dense_pull)
         self._seed = seed
         self._counter = 0
         # pending: (center_ids, ctx_table_ids per pair row, labels, buffer)
         self._pending: List[tuple] = []
         self._coalescer: Optional[DenseCoalescer] = None
+        self._all_rows: Optional[torch.Tensor] = None
         self._device: Optional[torch.device] = None
 
     def batch_size(self, batch: SGNSBatch) -> int:
@@ -59,18 +69,40 @@ class SGNSWorkerLogic(WorkerLogic):
         n = len(data)
         dev = data.centers.device
         k = self.neg
-        # expand each positive into (1 positive + k negatives) pair rows
-        centers = data.centers.repeat_interleave(k + 1)
-        keys = (torch.arange(n * k, dtype=torch.int64, device=dev)
-                + self._counter * (n * k + 1) + self._seed)
+        # expand each positive into (1 positive + k negatives) pair rows;
+        # on device one fused kernel builds rows + hash-RNG negatives +
+        # labels (bit-identical negative sequence to this torch path)
+        base_key = self._counter * (n * k + 1) + self._seed
         self._counter += 1
-        neg_ctx = ops.reference.splitmix64(keys).abs() % self.V
-        ctx = torch.empty(n * (k + 1), dtype=torch.int64, device=dev)
-        ctx.view(n, k + 1)[:, 0] = data.contexts
-        ctx.view(n, k + 1)[:, 1:] = neg_ctx.view(n, k)
-        labels = torch.zeros(n * (k + 1), dtype=torch.float32, device=dev)
-        labels.view(n, k + 1)[:, 0] = 1.0
-        ctx_table_ids = ctx + self.V  # context rows live at [V, 2V)
+        if dev.type == "cuda":
+            from ..ops import hip
+            centers, ctx_table_ids, labels = hip.ext().sgns_expand(
+                data.centers, data.contexts, k, self.V, base_key)
+        else:
+            centers = data.centers.repeat_interleave(k + 1)
+            keys = (torch.arange(n * k, dtype=torch.int64, device=dev)
+                    + base_key)
+            neg_ctx = ops.reference.splitmix64(keys).abs() % self.V
+            ctx = torch.empty(n * (k + 1), dtype=torch.int64, device=dev)
+            ctx.view(n, k + 1)[:, 0] = data.contexts
+            ctx.view(n, k + 1)[:, 1:] = neg_ctx.view(n, k)
+            labels = torch.zeros(n * (k + 1), dtype=torch.float32,
+                                 device=dev)
+            labels.view(n, k + 1)[:, 0] = 1.0
+            ctx_table_ids = ctx + self.V  # context rows live at [V, 2V)
+        if self.dense_pull:
+            # whole-table pull: slots ARE the table ids, no coalescing
+            if self._all_rows is None or self._all_rows.device != dev:
+                self._all_rows = torch.arange(2 * self.V,
+                                              dtype=torch.int64,
+                                              device=dev)
+            uids = self._all_rows
+            c_slots, o_slots = centers, ctx_table_ids
+            buf = PullBuffer(uids, self.dim, assume_sorted=True)
+            self._pending.append((centers, ctx_table_ids, labels, buf,
+                                  c_slots, o_slots))
+            client.pull(uids, unique=True)
+            return
         allids = torch.cat([centers, ctx_table_ids])
         if self._coalescer is None and 2 * self.V <= DenseCoalescer.MAX_SPACE:
             self._coalescer = DenseCoalescer(2 * self.V, dev)

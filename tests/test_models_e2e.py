@@ -252,3 +252,24 @@ def test_mean_item_delta_converges():
     eng.run(epoch_iterator(batches, 10))
     l1 = _mf_loss(wl, table, batches)
     assert l1 < 0.5 * l0
+
+
+def test_sgns_dense_pull_matches_sparse():
+    """dense_pull=True (whole-table pull, raw-id slots) must produce the
+    same trained table as the sparse-pull path: the two differ only in
+    how rows are addressed, not in math."""
+    V, dim = 60, 8
+    pairs = list(synth.sgns_pairs(V, 800, 100, seed=5))
+
+    def train(dense):
+        part = ModPartitioner(2 * V, 1)
+        table = ShardedTable(part, 0, dim, "cpu", init=(-0.05, 0.05))
+        wl = SGNSWorkerLogic(V, dim, learning_rate=0.05, negatives=2,
+                             dense_pull=dense, seed=11)
+        eng = PSEngine(PSConfig(num_ids=2 * V, dim=dim), wl,
+                       TableServerLogic(table), None, part)
+        eng.run(pairs * 2)
+        return table.data.clone()
+
+    a, b = train(False), train(True)
+    assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()

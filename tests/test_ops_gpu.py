@@ -412,3 +412,28 @@ def test_flush_coalesced_duplicate_pushes_adagrad_gpu():
         R.apply_adagrad(W, G, ids, deltas, 0.1)
     assert torch.allclose(W, table.data.cpu(), atol=1e-3, rtol=1e-3)
     assert torch.allclose(G, table.accum.cpu(), atol=1e-3, rtol=1e-3)
+
+
+def test_sgns_expand_matches_torch_path():
+    """Fused pair-expansion kernel is bit-identical to the torch
+    reference path in models/sgns.py (same splitmix64 negative ids)."""
+    from flink_parameter_server_amd.ops import hip
+    n, k, V = 5000, 5, 997
+    g = torch.Generator().manual_seed(21)
+    centers = torch.randint(0, V, (n,), generator=g)
+    contexts = torch.randint(0, V, (n,), generator=g)
+    base_key = 7 * (n * k + 1) + 11
+    oc, ox, ol = hip.ext().sgns_expand(centers.to(DEV), contexts.to(DEV),
+                                       k, V, base_key)
+    # torch reference
+    cen = centers.repeat_interleave(k + 1)
+    keys = torch.arange(n * k, dtype=torch.int64) + base_key
+    neg = R.splitmix64(keys).abs() % V
+    ctx = torch.empty(n * (k + 1), dtype=torch.int64)
+    ctx.view(n, k + 1)[:, 0] = contexts
+    ctx.view(n, k + 1)[:, 1:] = neg.view(n, k)
+    lab = torch.zeros(n * (k + 1))
+    lab.view(n, k + 1)[:, 0] = 1.0
+    assert torch.equal(oc.cpu(), cen)
+    assert torch.equal(ox.cpu(), ctx + V)
+    assert torch.equal(ol.cpu(), lab)
