@@ -813,6 +813,105 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
 }
 
 // ---------------------------------------------------------------------
+// K10b: hash-based id coalescing for unbounded id spaces (the 1e9-
+// feature LR config and any space too big for the dense-flag
+// coalescer).  Replaces torch.unique (device radix over the full
+// batch) + torch.searchsorted (a binary search per element — measured
+// 54% of the LR round) with an open-addressing hash table:
+//   insert  — linear-probe CAS into a pow2 table, record the probe
+//             position per element
+//   count   — one ballot-popcount pass over the table (64-slot chunks)
+//   scan    — exclusive starts per chunk (single-block scan)
+//   assign  — occupied slots take chunk-base + atomic rank; emit uids
+//   gather  — slots[j] = slotmap[pos[j]]
+// uids come out UNIQUE but UNSORTED (hash-position order).  That is
+// sufficient everywhere: the engine's unique=True pull skips
+// coalescing, shard bucketing handles any order, and the sort-free
+// adagrad apply only needs duplicates adjacent — unique ids have none.
+// ---------------------------------------------------------------------
+#define HASH_EMPTY (-1ll)
+
+__global__ __launch_bounds__(256) void k_hash_insert(
+    const int64_t* __restrict__ ids, int64_t n,
+    long long* __restrict__ keys, int* __restrict__ pos_out,
+    int64_t mask) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       j < n; j += (int64_t)gridDim.x * blockDim.x) {
+    const long long id = (long long)ids[j];
+    int64_t p = (int64_t)(splitmix64((uint64_t)id) & (uint64_t)mask);
+    for (;;) {
+      const long long k = keys[p];
+      if (k == id) break;
+      if (k == HASH_EMPTY) {
+        const long long old = atomicCAS(
+            (unsigned long long*)&keys[p],
+            (unsigned long long)HASH_EMPTY, (unsigned long long)id);
+        if (old == HASH_EMPTY || old == id) break;
+      }
+      p = (p + 1) & mask;
+    }
+    pos_out[j] = (int)p;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_hash_count(
+    const long long* __restrict__ keys, int64_t cap,
+    int* __restrict__ counts) {
+  // one 64-slot chunk per wave-iteration: ballot-popcount, one atomic
+  const int lane = threadIdx.x % WAVE;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < cap; i += (int64_t)gridDim.x * blockDim.x) {
+    const uint64_t b = __ballot(keys[i] != HASH_EMPTY);
+    if (lane == 0 && b)
+      atomicAdd(&counts[i >> 10], (int)__popcll(b));
+  }
+}
+
+__global__ __launch_bounds__(256) void k_hash_assign(
+    const long long* __restrict__ keys, int64_t cap,
+    int* __restrict__ cursor, int* __restrict__ slotmap,
+    int64_t* __restrict__ uids) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < cap; i += (int64_t)gridDim.x * blockDim.x) {
+    const long long k = keys[i];
+    if (k == HASH_EMPTY) continue;
+    const int s = atomicAdd(&cursor[i >> 10], 1);
+    slotmap[i] = s;
+    uids[s] = (int64_t)k;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_hash_slots(
+    const int* __restrict__ pos, const int* __restrict__ slotmap,
+    int* __restrict__ slots, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       j < n; j += (int64_t)gridDim.x * blockDim.x)
+    slots[j] = slotmap[pos[j]];
+}
+
+// resolve ids against an EXISTING table (pull answers arriving in
+// shard-bucketed order resolve their buffer rows this way); ids not in
+// the table get slot -1 (foreign answers are skipped by the buffer)
+__global__ __launch_bounds__(256) void k_hash_lookup(
+    const int64_t* __restrict__ ids, int64_t n,
+    const long long* __restrict__ keys, const int* __restrict__ slotmap,
+    int64_t mask, int* __restrict__ slots_out) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       j < n; j += (int64_t)gridDim.x * blockDim.x) {
+    const long long id = (long long)ids[j];
+    int64_t p = (int64_t)(splitmix64((uint64_t)id) & (uint64_t)mask);
+    int out = -1;
+    for (;;) {
+      const long long k = keys[p];
+      if (k == id) { out = slotmap[p]; break; }
+      if (k == HASH_EMPTY) break;
+      p = (p + 1) & mask;
+    }
+    slots_out[j] = out;
+  }
+}
+
+// ---------------------------------------------------------------------
 // K5/K8 prologue: fused SGNS pair expansion — one pass builds the
 // (1 positive + k hash-RNG negatives) pair rows, context table ids
 // (+V offset) and labels that the Python path assembled with ~10
@@ -1601,6 +1700,71 @@ torch::Tensor sgns_step(torch::Tensor vecs, torch::Tensor center_slots,
   return deltas;
 }
 
+std::vector<torch::Tensor> hash_coalesce(torch::Tensor ids) {
+  CHECK_DEV(ids); CHECK_CONT(ids);
+  TORCH_CHECK(ids.dtype() == torch::kInt64, "ids must be int64");
+  int64_t n = ids.numel();
+  auto opts_i32 = torch::dtype(torch::kInt32).device(ids.device());
+  if (!n)
+    return {torch::empty({0}, ids.options()), torch::empty({0}, opts_i32)};
+  // pow2 capacity >= 2n => load factor <= 0.5, ~2 probes expected
+  int64_t cap = 1024;
+  while (cap < 2 * n) cap <<= 1;
+  TORCH_CHECK(cap <= (1ll << 31), "batch too large for hash_coalesce");
+  auto keys = torch::empty({cap}, ids.options());
+  hipMemsetAsync(keys.data_ptr(), 0xFF, cap * 8, cur_stream());
+  auto pos = torch::empty({n}, opts_i32);
+  hipLaunchKernelGGL(k_hash_insert, dim3(grid_for(n, 256)), dim3(256), 0,
+                     cur_stream(), ids.data_ptr<int64_t>(), n,
+                     (long long*)keys.data_ptr<int64_t>(),
+                     pos.data_ptr<int>(), cap - 1);
+  int64_t nchunks = (cap + 1023) / 1024;
+  auto counts = torch::zeros({nchunks}, opts_i32);
+  hipLaunchKernelGGL(k_hash_count, dim3(grid_for(cap, 256)), dim3(256),
+                     0, cur_stream(),
+                     (const long long*)keys.data_ptr<int64_t>(), cap,
+                     counts.data_ptr<int>());
+  auto cursor = exclusive_starts(counts).contiguous();
+  // total uniques = last start + last count (device->host: the caller
+  // needs the size to slice uids; torch.unique syncs the same way)
+  int last[2];
+  hipMemcpyAsync(&last[0], cursor.data_ptr<int>() + nchunks - 1, 4,
+                 hipMemcpyDeviceToHost, cur_stream());
+  hipMemcpyAsync(&last[1], counts.data_ptr<int>() + nchunks - 1, 4,
+                 hipMemcpyDeviceToHost, cur_stream());
+  auto slotmap = torch::empty({cap}, opts_i32);
+  auto uids = torch::empty({n}, ids.options());
+  hipLaunchKernelGGL(k_hash_assign, dim3(grid_for(cap, 256)), dim3(256),
+                     0, cur_stream(),
+                     (const long long*)keys.data_ptr<int64_t>(), cap,
+                     cursor.data_ptr<int>(), slotmap.data_ptr<int>(),
+                     uids.data_ptr<int64_t>());
+  auto slots = torch::empty({n}, opts_i32);
+  hipLaunchKernelGGL(k_hash_slots, dim3(grid_for(n, 256)), dim3(256), 0,
+                     cur_stream(), pos.data_ptr<int>(),
+                     slotmap.data_ptr<int>(), slots.data_ptr<int>(), n);
+  hipStreamSynchronize(cur_stream());
+  int64_t m = (int64_t)last[0] + (int64_t)last[1];
+  // keys+slotmap stay alive with the caller for hash_lookup resolution
+  return {uids.narrow(0, 0, m), slots, keys, slotmap};
+}
+
+torch::Tensor hash_lookup(torch::Tensor keys, torch::Tensor slotmap,
+                          torch::Tensor ids) {
+  CHECK_DEV(keys); CHECK_CONT(keys); CHECK_CONT(slotmap); CHECK_CONT(ids);
+  TORCH_CHECK(ids.dtype() == torch::kInt64);
+  int64_t n = ids.numel(), cap = keys.numel();
+  auto slots = torch::empty({n}, torch::dtype(torch::kInt32)
+                                     .device(ids.device()));
+  if (!n) return slots;
+  hipLaunchKernelGGL(k_hash_lookup, dim3(grid_for(n, 256)), dim3(256),
+                     0, cur_stream(), ids.data_ptr<int64_t>(), n,
+                     (const long long*)keys.data_ptr<int64_t>(),
+                     slotmap.data_ptr<int>(), cap - 1,
+                     slots.data_ptr<int>());
+  return slots;
+}
+
 std::vector<torch::Tensor> sgns_expand(torch::Tensor centers,
                                        torch::Tensor contexts, int64_t k,
                                        int64_t V, int64_t base_key) {
@@ -1733,6 +1897,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgns_step", &sgns_step, "K8 fused SGNS step");
   m.def("sgns_expand", &sgns_expand,
         "K5/K8 fused pair expansion + hash-RNG negative sampling");
+  m.def("hash_coalesce", &hash_coalesce,
+        "K10b hash-based id coalescing (unsorted-unique uids + slots)");
+  m.def("hash_lookup", &hash_lookup,
+        "K10b slot resolution against an existing coalescer table");
   m.def("pa_step", &pa_step, "K7 PA binary step");
   m.def("pa_multiclass_step", &pa_multiclass_step, "K7 PA multiclass");
   m.def("lr_step", &lr_step, "K9 sparse logistic step");

@@ -18,13 +18,21 @@ import torch
 
 
 class PullBuffer:
-    """Collects pulled rows for a fixed set of unique ids."""
+    """Collects pulled rows for a fixed set of unique ids.
+
+    ``resolver``: optional callable mapping an id tensor to buffer rows
+    (foreign ids -> -1).  Used by the hash-coalesced mode, whose uids
+    are unique but UNSORTED (hash-position order) — searchsorted does
+    not apply there.  Without a resolver uids must be (or are made)
+    sorted.
+    """
 
     def __init__(self, uids: torch.Tensor, dim: int,
-                 assume_sorted: bool = False):
-        if not assume_sorted:
+                 assume_sorted: bool = False, resolver=None):
+        if not assume_sorted and resolver is None:
             uids, _ = torch.sort(uids)
         self.uids = uids
+        self.resolver = resolver
         self.values = torch.empty(uids.numel(), dim, dtype=torch.float32,
                                   device=uids.device)
         # lazily allocated: the whole-answer fast path never needs the
@@ -41,7 +49,9 @@ class PullBuffer:
         return self._have_t
 
     def slots_of(self, ids: torch.Tensor) -> torch.Tensor:
-        """Position of each id in the sorted unique-id list."""
+        """Buffer row of each id (caller guarantees membership)."""
+        if self.resolver is not None:
+            return self.resolver(ids)
         return torch.searchsorted(self.uids, ids)
 
     def offer_exact(self, ids: torch.Tensor, values: torch.Tensor) -> None:
@@ -55,7 +65,7 @@ class PullBuffer:
             # pull / single-shard): adopt the payload, skip the permute
             self.values = values
         else:
-            rows = torch.searchsorted(self.uids, ids)
+            rows = self.slots_of(ids).long()
             self.values[rows] = values.to(self.values.dtype)
         if self._have_t is not None:
             self._have_t.fill_(True)
@@ -66,12 +76,17 @@ class PullBuffer:
         rows already filled."""
         if self.complete or not self.uids.numel():
             return
-        pos = torch.searchsorted(self.uids, ids)
-        pos_c = torch.clamp(pos, max=self.uids.numel() - 1)
-        hit = self.uids[pos_c] == ids
+        if self.resolver is not None:
+            pos = self.resolver(ids).long()
+            hit = pos >= 0
+            rows = pos[hit]
+        else:
+            pos = torch.searchsorted(self.uids, ids)
+            pos_c = torch.clamp(pos, max=self.uids.numel() - 1)
+            hit = self.uids[pos_c] == ids
+            rows = pos_c[hit]
         if not bool(hit.any()):
             return
-        rows = pos_c[hit]
         fresh = ~self._have[rows]
         rows = rows[fresh]
         if not rows.numel():
@@ -114,3 +129,41 @@ class DenseCoalescer:
         uids = present.nonzero().reshape(-1)
         slots = lookup[idx] - 1
         return uids, slots
+
+
+class HashCoalescer:
+    """Device-side id coalescing for UNBOUNDED id spaces (the 1e9-
+    feature LR config — anything too big for :class:`DenseCoalescer`'s
+    dense flag array).
+
+    One open-addressing hash pass replaces the torch.unique radix sort
+    + the per-element torch.searchsorted that dominated the LR round
+    (measured 54% + ~20%: profiles/PERF_NOTES.md round 2).  Returns
+    (unsorted-unique uids, int32 slots) and keeps the table alive so
+    later answer segments resolve rows via :meth:`resolver` (pull
+    answers arrive shard-bucketed, i.e. permuted, at world > 1).
+    Unsorted uids are sufficient everywhere: unique=True pulls skip the
+    engine's coalescing, bucketing handles any order, and the adagrad
+    apply only needs duplicates adjacent — unique ids have none.
+    """
+
+    def __init__(self):
+        self._keys = None
+        self._slotmap = None
+
+    def __call__(self, ids: torch.Tensor):
+        from .. import ops
+        uids, slots, self._keys, self._slotmap = \
+            ops.hip.ext().hash_coalesce(ids)
+        return uids, slots
+
+    @property
+    def resolver(self):
+        keys, slotmap = self._keys, self._slotmap
+        from .. import ops
+        ext = ops.hip.ext()
+
+        def _resolve(ids: torch.Tensor) -> torch.Tensor:
+            return ext.hash_lookup(keys, slotmap, ids)
+
+        return _resolve

@@ -15,7 +15,7 @@ import torch
 
 from .. import ops
 from ..api import ParameterServerClient, WorkerLogic
-from .common import DenseCoalescer, PullBuffer
+from .common import DenseCoalescer, HashCoalescer, PullBuffer
 from .pa import SparseBatch
 
 
@@ -39,9 +39,16 @@ class LRWorkerLogic(WorkerLogic):
                 client: ParameterServerClient) -> None:
         if self._coalescer is not None:
             uids, slots = self._coalescer(data.feat_ids)
+            buf = PullBuffer(uids, 1, assume_sorted=True)
+        elif data.feat_ids.is_cuda:
+            # unbounded id space (1e9 hashed features): hash coalescing
+            # — no radix sort, no searchsorted; uids unsorted-unique
+            hc = HashCoalescer()
+            uids, slots = hc(data.feat_ids)
+            buf = PullBuffer(uids, 1, resolver=hc.resolver)
         else:
             uids, slots = torch.unique(data.feat_ids), None
-        buf = PullBuffer(uids, 1, assume_sorted=True)
+            buf = PullBuffer(uids, 1, assume_sorted=True)
         self._pending.append((data, buf, slots))
         client.pull(uids, unique=True)
 

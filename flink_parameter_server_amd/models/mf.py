@@ -27,7 +27,7 @@ import torch
 
 from .. import ops
 from ..api import ParameterServerClient, WorkerLogic
-from .common import DenseCoalescer, PullBuffer
+from .common import DenseCoalescer, HashCoalescer, PullBuffer
 
 
 class Rating(NamedTuple):
@@ -168,6 +168,11 @@ class OnlineMFWorkerLogic(WorkerLogic):
         elif self._coalescer is not None:
             uids, slots = self._coalescer(data.item_ids)
             buf = PullBuffer(uids, self.n_factors, assume_sorted=True)
+        elif data.item_ids.is_cuda:
+            # unbounded item space: hash coalescing (no sort/searchsorted)
+            hc = HashCoalescer()
+            uids, slots = hc(data.item_ids.long())
+            buf = PullBuffer(uids, self.n_factors, resolver=hc.resolver)
         else:
             uids = torch.unique(data.item_ids)
             slots = None

@@ -19,7 +19,7 @@ import torch
 
 from .. import ops
 from ..api import ParameterServerClient, WorkerLogic
-from .common import DenseCoalescer, PullBuffer
+from .common import DenseCoalescer, HashCoalescer, PullBuffer
 
 
 class SGNSBatch:
@@ -109,10 +109,17 @@ class SGNSWorkerLogic(WorkerLogic):
         if self._coalescer is not None:
             uids, slots = self._coalescer(allids)
             c_slots, o_slots = slots[:n * (k + 1)], slots[n * (k + 1):]
+            buf = PullBuffer(uids, self.dim, assume_sorted=True)
+        elif dev.type == "cuda":
+            # unbounded vocab: hash coalescing (no sort/searchsorted)
+            hc = HashCoalescer()
+            uids, slots = hc(allids)
+            c_slots, o_slots = slots[:n * (k + 1)], slots[n * (k + 1):]
+            buf = PullBuffer(uids, self.dim, resolver=hc.resolver)
         else:
             uids = torch.unique(allids)
             c_slots = o_slots = None
-        buf = PullBuffer(uids, self.dim, assume_sorted=True)
+            buf = PullBuffer(uids, self.dim, assume_sorted=True)
         self._pending.append((centers, ctx_table_ids, labels, buf,
                               c_slots, o_slots))
         client.pull(uids, unique=True)
