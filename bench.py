@@ -134,7 +134,8 @@ def make_lr(args, rank, world, device):
     part = ModPartitioner(n_feat, world)
     table = ShardedTable(part, rank, 1, device, optimizer="adagrad",
                          lr=0.05)
-    wl = LRWorkerLogic(n_features=n_feat, device=device)
+    wl = LRWorkerLogic(n_features=n_feat, device=device,
+                       assume_static_batches=True)
     g = torch.Generator().manual_seed(7 + rank)
     batches = []
     for _ in range(args.n_batches):

@@ -273,3 +273,35 @@ def test_sgns_dense_pull_matches_sparse():
 
     a, b = train(False), train(True)
     assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()
+
+
+def test_lr_static_batches_cache_matches_fresh():
+    """assume_static_batches reuses cached coalescing across epoch
+    revisits — trained weights must equal the uncached run."""
+    from flink_parameter_server_amd.models.lr import LRWorkerLogic
+    from flink_parameter_server_amd.models.pa import SparseBatch
+
+    n_feat = 500
+    g = torch.Generator().manual_seed(3)
+    batches = []
+    for _ in range(3):
+        B, nnz = 64, 8
+        feats = torch.randint(0, n_feat, (B * nnz,), generator=g)
+        vals = torch.randn(B * nnz, generator=g)
+        indptr = torch.arange(0, B * nnz + 1, nnz, dtype=torch.int64)
+        labels = (torch.rand(B, generator=g) < 0.5).float()
+        batches.append(SparseBatch(indptr, feats, vals, labels))
+
+    def run(static):
+        part = ModPartitioner(n_feat, 1)
+        table = ShardedTable(part, 0, 1, "cpu", optimizer="adagrad",
+                             lr=0.1)
+        wl = LRWorkerLogic(n_features=n_feat,
+                           assume_static_batches=static)
+        eng = PSEngine(PSConfig(num_ids=n_feat, dim=1), wl,
+                       TableServerLogic(table, dump_on_close=False,
+                                        pushes_unique=True), None, part)
+        eng.run(batches * 4)
+        return table.data.clone()
+
+    assert torch.equal(run(False), run(True))
