@@ -957,8 +957,8 @@ __global__ __launch_bounds__(256) void k_sparse_linear_step(
     const float* __restrict__ x_vals, const float* __restrict__ labels,
     const float* __restrict__ weights, float* __restrict__ margins,
     float* __restrict__ deltas, int64_t B, int mode, float C,
-    int64_t rep_stride) {
-  const int64_t rep_off = (int64_t)(blockIdx.x & 7) * rep_stride;
+    int64_t rep_stride, int rep_mask) {
+  const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % WAVE;
   const int64_t waves_per_block = blockDim.x / WAVE;
   const int64_t w0 = (int64_t)blockIdx.x * waves_per_block +
@@ -1805,8 +1805,9 @@ std::vector<torch::Tensor> pa_step(torch::Tensor indptr,
   auto margins = torch::zeros({B}, x_vals.options());
   if (!B) return {margins, torch::zeros_like(weights)};
   int64_t m = weights.numel();
-  int64_t reps = (x_vals.numel() >= 8 * m &&
-                  8 * m * 4 <= (256ll << 20)) ? 8 : 1;
+  static int64_t lin_reps = env_int("FPS_LINEAR_REPS", 8);
+  int64_t reps = (x_vals.numel() >= lin_reps * m &&
+                  lin_reps * m * 4 <= (256ll << 20)) ? lin_reps : 1;
   auto dbuf = torch::zeros({reps * m}, weights.options());
   auto launch = [&](auto* slots) {
     hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
@@ -1815,7 +1816,7 @@ std::vector<torch::Tensor> pa_step(torch::Tensor indptr,
                        x_vals.data_ptr<float>(), labels.data_ptr<float>(),
                        weights.data_ptr<float>(), margins.data_ptr<float>(),
                        dbuf.data_ptr<float>(), B, pa_mode(variant),
-                       (float)C, reps > 1 ? m : 0);
+                       (float)C, reps > 1 ? m : 0, (int)reps - 1);
   };
   if (feat_slots.dtype() == torch::kInt32) launch(feat_slots.data_ptr<int>());
   else launch(feat_slots.data_ptr<int64_t>());
@@ -1833,8 +1834,9 @@ std::vector<torch::Tensor> lr_step(torch::Tensor indptr,
   auto probs = torch::zeros({B}, x_vals.options());
   if (!B) return {probs, torch::zeros_like(weights)};
   int64_t m = weights.numel();
-  int64_t reps = (x_vals.numel() >= 8 * m &&
-                  8 * m * 4 <= (256ll << 20)) ? 8 : 1;
+  static int64_t lin_reps = env_int("FPS_LINEAR_REPS", 8);
+  int64_t reps = (x_vals.numel() >= lin_reps * m &&
+                  lin_reps * m * 4 <= (256ll << 20)) ? lin_reps : 1;
   auto gbuf = torch::zeros({reps * m}, weights.options());
   auto launch = [&](auto* slots) {
     hipLaunchKernelGGL(k_sparse_linear_step, dim3(grid_for(B * WAVE, 256)),
@@ -1843,7 +1845,7 @@ std::vector<torch::Tensor> lr_step(torch::Tensor indptr,
                        x_vals.data_ptr<float>(), labels.data_ptr<float>(),
                        weights.data_ptr<float>(), probs.data_ptr<float>(),
                        gbuf.data_ptr<float>(), B, /*mode=*/3, 0.f,
-                       reps > 1 ? m : 0);
+                       reps > 1 ? m : 0, (int)reps - 1);
   };
   if (feat_slots.dtype() == torch::kInt32) launch(feat_slots.data_ptr<int>());
   else launch(feat_slots.data_ptr<int64_t>());
