@@ -519,6 +519,17 @@ __global__ __launch_bounds__(256) void k_scale_rows_by_count(
   }
 }
 
+// posAB[posA[j]] = posB[j] — composite map from a pass-A record
+// position to the same sample's pass-B record position (cached-
+// position revisits scatter the coefficient straight into recB)
+__global__ __launch_bounds__(256) void k_compose_pos(
+    const int* __restrict__ posA, const int* __restrict__ posB,
+    int* __restrict__ posAB, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x)
+    posAB[posA[j]] = posB[j];
+}
+
 // pos[order[j]] = j  (inverse of a sort permutation)
 __global__ __launch_bounds__(256) void k_invert_order(
     const int64_t* __restrict__ order, int* __restrict__ pos, int64_t n) {
@@ -575,12 +586,17 @@ static torch::Tensor radix_sort_recs(torch::Tensor keys,
 // second pass.  rep_stride spreads flushes over 8 replica copies of
 // `out` (zipfian hot keys).
 // =====================================================================
+// ``recB``/``posAB`` non-null (cached-position revisits): MODE 0/1
+// writes the per-sample coefficient into recB[posAB[j]].z (the second
+// pass's val column) instead of rec.z — rec then stays fully static
+// across revisits and neither record array needs a scatter rebuild.
 template <int GP, int EPL, int MODE, bool SELF, bool ZERO_OUT>
 __global__ __launch_bounds__(256) void k_seg_fused(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride, int rep_mask) {
+    int64_t rep_stride, int rep_mask, int4* __restrict__ recB,
+    const int* __restrict__ posAB) {
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
   const int64_t gpb = blockDim.x / GP;
@@ -674,7 +690,10 @@ __global__ __launch_bounds__(256) void k_seg_fused(
                 ? (__int_as_float(rj.z) - dot)
                 : lrc * (__int_as_float(rj.z) -
                          1.0f / (1.0f + __expf(-dot)));
-        if (lane == 0) rec[j].z = __float_as_int(g);
+        if (lane == 0) {
+          if (recB) recB[posAB[j]].z = __float_as_int(g);
+          else rec[j].z = __float_as_int(g);
+        }
       }
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
@@ -698,7 +717,8 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride, int rep_mask) {
+    int64_t rep_stride, int rep_mask, int4* __restrict__ recB,
+    const int* __restrict__ posAB) {
   extern __shared__ int4 srec[];
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
@@ -799,7 +819,10 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
                 ? (__int_as_float(rj.z) - dot)
                 : lrc * (__int_as_float(rj.z) -
                          1.0f / (1.0f + __expf(-dot)));
-        if (lane == 0) ((int*)&rec[start + lj])[2] = __float_as_int(g);
+        if (lane == 0) {
+          if (recB) recB[posAB[start + lj]].z = __float_as_int(g);
+          else ((int*)&rec[start + lj])[2] = __float_as_int(g);
+        }
       }
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
@@ -1393,7 +1416,8 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                       const torch::Tensor& keyed,
                       const torch::Tensor& other, int64_t dim4,
                       double scale, double a, double lrc,
-                      int64_t rep_stride = 0, int rep_mask = 7) {
+                      int64_t rep_stride = 0, int rep_mask = 7,
+                      int4* recB = nullptr, const int* posAB = nullptr) {
   int64_t n = rec.size(0);
   if (!n) return;
   if (dim4 == 0) {
@@ -1423,7 +1447,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
             (const float4*)keyed.data_ptr<float>(),
             (const float4*)other.data_ptr<float>(), n, dim4,
             (float)scale, (float)a, (float)lrc, chunk, rep_stride / 4,
-            rep_mask);
+            rep_mask, recB, posAB);
         return;
       }
     }
@@ -1434,7 +1458,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                        (const float4*)keyed.data_ptr<float>(),
                        (const float4*)other.data_ptr<float>(), n, dim4,
                        (float)scale, (float)a, (float)lrc, chunk,
-                       rep_stride / 4, rep_mask);
+                       rep_stride / 4, rep_mask, recB, posAB);
   });
 }
 
@@ -1464,6 +1488,11 @@ static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
 // tensors, so a pointer can never alias a different live tensor.
 struct PosCacheEntry {
   torch::Tensor keys, slots, posA, posB, cntB;
+  // revisit fast path (dim % 4 == 0 shapes): both record arrays are
+  // content-static across revisits (key/slot from the static batch,
+  // recA.val = the ratings, recB.val rewritten by pass A each visit
+  // via posAB) — no per-visit scatter rebuilds at all.
+  torch::Tensor recA, recB, posAB;
 };
 // exact composite key (no hash collisions): the held tensor references
 // inside each entry keep the keyed storages alive, so a pointer can
@@ -1510,10 +1539,17 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     if (it != g_pos_cache.end()) hit = &it->second;
   }
 
+  // revisit fast path: both record arrays come straight from the
+  // cache; pass A scatters the coefficient into recB.z via posAB, so
+  // neither array needs a rebuild (the two position scatters were 39%
+  // of the shipped mf32 round — profiles/PERF_NOTES.md round 2)
+  const bool emit_ok = dim4 != 0;
   torch::Tensor recA;
   torch::Tensor posA, posB;
   torch::Tensor histB;
-  if (hit) {
+  if (hit && emit_ok && hit->recA.defined()) {
+    recA = hit->recA;
+  } else if (hit) {
     recA = torch::empty({n, 4}, opts_i32);
     if (item_slots.dtype() == torch::kInt32)
       hipLaunchKernelGGL((k_apply_pos<int>), dim3(grid_for(n, 256)),
@@ -1576,43 +1612,75 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     recA = build_rec(item_slots, user_rows, ratings, m, histA, posA);
   }
 
+  // caching + emit: recB (key/slot static; val rewritten by pass A)
+  // and the composite posAB are prepared BEFORE pass A on the first
+  // visit, so every visit runs the same emit-into-recB walk
+  torch::Tensor recB, cntB, posAB;
+  bool have_recB = false;
+  if (emit_ok && reuse_tag && (!hit || hit->recB.defined())) {
+    if (hit) {
+      recB = hit->recB;
+      posAB = hit->posAB;
+      cntB = hit->cntB;
+    } else {
+      TORCH_CHECK(histB.defined() || m > PACK_MAX_M || U > PACK_MAX_M,
+                  "histB expected for LDS-sized shapes");
+      recB = build_rec_swap(recA, U, histB, posB);
+      posAB = torch::empty({n}, opts_i32);
+      hipLaunchKernelGGL(k_compose_pos, dim3(grid_for(n, 256)),
+                         dim3(256), 0, cur_stream(),
+                         posA.data_ptr<int>(), posB.data_ptr<int>(),
+                         posAB.data_ptr<int>(), n);
+      cntB = histB;
+      if (g_pos_cache.size() >= 32) g_pos_cache.clear();
+      g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB,
+                         recA, recB, posAB};
+    }
+    have_recB = true;
+  }
+
   // pass A: recs by item slot; fused e + item-delta accumulation
   // (the dim4==0 envelope walk scatters with per-record atomics and
   // has no replica spreading)
   int64_t reps = dim4 ? delta_reps(n, m, dim) : 1;
   torch::Tensor deltas;
+  int4* eb = have_recB ? (int4*)recB.data_ptr<int>() : nullptr;
+  const int* epos = have_recB ? posAB.data_ptr<int>() : nullptr;
   if (reps > 1) {
     auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
     seg_fused<0, true, true>(rep_buf, recA, item_vecs, user_table, dim4,
-                             lr, -lr * reg, 0.0, m * dim, (int)reps - 1);
+                             lr, -lr * reg, 0.0, m * dim, (int)reps - 1,
+                             eb, epos);
     deltas = rep_buf.view({reps, m, dim}).sum(0);
   } else {
     deltas = torch::zeros_like(item_vecs);
     seg_fused<0, true, true>(deltas, recA, item_vecs, user_table, dim4,
-                             lr, -lr * reg, 0.0);
+                             lr, -lr * reg, 0.0, 0, 7, eb, epos);
   }
 
   // pass B: recs by user row; apply user updates in place.  The L2
   // term u -= lr*reg*cnt*u0 is applied analytically first (see
   // k_scale_rows_by_count), then the delta term lr*sum(e*v) accumulates
   // with no SELF read.
-  torch::Tensor recB, cntB;
-  if (hit) {
-    cntB = hit->cntB;
-    recB = torch::empty({n, 4}, opts_i32);
-    hipLaunchKernelGGL(k_apply_pos_swap, dim3(grid_for(n, 256)),
-                       dim3(256), 0, cur_stream(),
-                       (const int4*)recA.data_ptr<int>(),
-                       hit->posB.data_ptr<int>(),
-                       (int4*)recB.data_ptr<int>(), n);
-  } else {
-    TORCH_CHECK(histB.defined() || !mean_users,
-                "mean_users needs per-user counts (table too large for "
-                "the LDS histogram)");
-    recB = build_rec_swap(recA, U, histB, posB);
-    if (reuse_tag) {
-      if (g_pos_cache.size() >= 32) g_pos_cache.clear();
-      g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB};
+  if (!have_recB) {
+    if (hit) {
+      cntB = hit->cntB;
+      recB = torch::empty({n, 4}, opts_i32);
+      hipLaunchKernelGGL(k_apply_pos_swap, dim3(grid_for(n, 256)),
+                         dim3(256), 0, cur_stream(),
+                         (const int4*)recA.data_ptr<int>(),
+                         hit->posB.data_ptr<int>(),
+                         (int4*)recB.data_ptr<int>(), n);
+    } else {
+      TORCH_CHECK(histB.defined() || !mean_users,
+                  "mean_users needs per-user counts (table too large "
+                  "for the LDS histogram)");
+      recB = build_rec_swap(recA, U, histB, posB);
+      if (reuse_tag) {
+        if (g_pos_cache.size() >= 32) g_pos_cache.clear();
+        g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB,
+                           {}, {}, {}};
+      }
     }
   }
   if (mean_users) {
