@@ -519,17 +519,6 @@ __global__ __launch_bounds__(256) void k_scale_rows_by_count(
   }
 }
 
-// posAB[posA[j]] = posB[j] — composite map from a pass-A record
-// position to the same sample's pass-B record position (cached-
-// position revisits scatter the coefficient straight into recB)
-__global__ __launch_bounds__(256) void k_compose_pos(
-    const int* __restrict__ posA, const int* __restrict__ posB,
-    int* __restrict__ posAB, int64_t n) {
-  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
-       j += (int64_t)gridDim.x * blockDim.x)
-    posAB[posA[j]] = posB[j];
-}
-
 // pos[order[j]] = j  (inverse of a sort permutation)
 __global__ __launch_bounds__(256) void k_invert_order(
     const int64_t* __restrict__ order, int* __restrict__ pos, int64_t n) {
@@ -1626,11 +1615,10 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       TORCH_CHECK(histB.defined() || m > PACK_MAX_M || U > PACK_MAX_M,
                   "histB expected for LDS-sized shapes");
       recB = build_rec_swap(recA, U, histB, posB);
-      posAB = torch::empty({n}, opts_i32);
-      hipLaunchKernelGGL(k_compose_pos, dim3(grid_for(n, 256)),
-                         dim3(256), 0, cur_stream(),
-                         posA.data_ptr<int>(), posB.data_ptr<int>(),
-                         posAB.data_ptr<int>(), n);
+      // posB is indexed by recA position (the swap pack walks recA),
+      // so it IS the pass-A-position -> recB-position map the emit
+      // needs — no composition required
+      posAB = posB;
       cntB = histB;
       if (g_pos_cache.size() >= 32) g_pos_cache.clear();
       g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB,
