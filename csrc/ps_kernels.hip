@@ -1477,11 +1477,13 @@ static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
 // tensors, so a pointer can never alias a different live tensor.
 struct PosCacheEntry {
   torch::Tensor keys, slots, posA, posB, cntB;
-  // revisit fast path (dim % 4 == 0 shapes): both record arrays are
-  // content-static across revisits (key/slot from the static batch,
-  // recA.val = the ratings, recB.val rewritten by pass A each visit
-  // via posAB) — no per-visit scatter rebuilds at all.
-  torch::Tensor recA, recB, posAB;
+  // revisit fast path (dim % 4 == 0 shapes): recB's key/slot are
+  // content-static across revisits and its val column is rewritten by
+  // pass A each visit via posAB, so recB never needs a rebuild.  recA
+  // additionally caches the VAL column (the ratings), so its reuse is
+  // gated on the ratings tensor's identity (``vals`` holds the
+  // reference — a different ratings buffer re-scatters recA only).
+  torch::Tensor recA, recB, posAB, vals;
 };
 // exact composite key (no hash collisions): the held tensor references
 // inside each entry keep the keyed storages alive, so a pointer can
@@ -1536,7 +1538,11 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   torch::Tensor recA;
   torch::Tensor posA, posB;
   torch::Tensor histB;
-  if (hit && emit_ok && hit->recA.defined()) {
+  const bool same_vals =
+      hit && hit->vals.defined() &&
+      hit->vals.data_ptr() == ratings.data_ptr() &&
+      hit->vals.numel() == ratings.numel();
+  if (hit && emit_ok && hit->recA.defined() && same_vals) {
     recA = hit->recA;
   } else if (hit) {
     recA = torch::empty({n, 4}, opts_i32);
@@ -1622,7 +1628,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       cntB = histB;
       if (g_pos_cache.size() >= 32) g_pos_cache.clear();
       g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB,
-                         recA, recB, posAB};
+                         recA, recB, posAB, ratings};
     }
     have_recB = true;
   }
@@ -1667,7 +1673,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       if (reuse_tag) {
         if (g_pos_cache.size() >= 32) g_pos_cache.clear();
         g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB,
-                           {}, {}, {}};
+                           {}, {}, {}, {}};
       }
     }
   }

@@ -91,7 +91,11 @@ def mf_sgd_step(user_table: torch.Tensor, user_rows: torch.Tensor,
     """``reuse_tag != 0`` promises that whenever the same (tag,
     user_rows buffer, item_slots buffer) recur, their CONTENT is
     unchanged (epoch resweeps over resident batches) — the HIP path then
-    reuses the cached sort positions instead of re-sorting."""
+    reuses the cached sort positions instead of re-sorting, and caches
+    both sorted record arrays outright (zero per-visit scatter rebuilds)
+    when additionally the SAME ratings buffer recurs with unchanged
+    content; a different ratings buffer safely re-scatters only the
+    first-pass records."""
     if _cuda(user_table):
         return hip.ext().mf_sgd_step(user_table, user_rows, item_vecs,
                                      item_slots, ratings, lr, reg,
