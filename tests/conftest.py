@@ -16,3 +16,12 @@ def pytest_collection_modifyitems(config, items):
     for item in items:
         if "gpu" in item.keywords:
             item.add_marker(skip)
+
+
+def free_port() -> int:
+    """An OS-assigned free TCP port (avoids fixed-port collisions with
+    lingering sockets from earlier test processes)."""
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]

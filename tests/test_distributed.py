@@ -62,7 +62,8 @@ def _job(rank, world, port, outdir):
 @pytest.mark.timeout(180)
 def test_gloo_world2_push_pull_routing():
     world = 2
-    port = 29511
+    from conftest import free_port
+    port = free_port()
     with tempfile.TemporaryDirectory() as d:
         mp.spawn(_job, args=(world, port, d), nprocs=world, join=True)
         r0 = torch.load(os.path.join(d, "rank0.pt"), weights_only=False)
@@ -153,7 +154,8 @@ def _soak_job(rank, world, port, feedback, outdir):
 @pytest.mark.parametrize("feedback", ["fresh", "pipelined"])
 def test_gloo_world8_soak(feedback):
     world = 8
-    port = 29531 if feedback == "fresh" else 29532
+    from conftest import free_port
+    port = free_port()
     with tempfile.TemporaryDirectory() as d:
         mp.spawn(_soak_job, args=(world, port, feedback, d),
                  nprocs=world, join=True)

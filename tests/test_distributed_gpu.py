@@ -76,10 +76,10 @@ def _job(rank, world, port, outdir, feedback):
 
 
 @pytest.mark.timeout(300)
-@pytest.mark.parametrize("feedback,port", [("fresh", 29517),
-                                           ("merged", 29518),
-                                           ("pipelined", 29519)])
-def test_two_rank_mf_on_gpu_converges(feedback, port):
+@pytest.mark.parametrize("feedback", ["fresh", "merged", "pipelined"])
+def test_two_rank_mf_on_gpu_converges(feedback):
+    from conftest import free_port
+    port = free_port()
     world = 2
     with tempfile.TemporaryDirectory() as d:
         mp.spawn(_job, args=(world, port, d, feedback), nprocs=world,
@@ -157,7 +157,8 @@ def _device_delivery_job(rank, world, port, outdir):
 def test_device_delivery_bf16_wire_two_ranks():
     world = 2
     with tempfile.TemporaryDirectory() as d:
-        mp.spawn(_device_delivery_job, args=(world, 29520, d),
+        from conftest import free_port
+        mp.spawn(_device_delivery_job, args=(world, free_port(), d),
                  nprocs=world, join=True)
         merged = {}
         total_answers = 0
@@ -235,7 +236,8 @@ def test_lr_hash_resolver_two_ranks():
     the distributed result row-for-row."""
     world = 2
     with tempfile.TemporaryDirectory() as d:
-        mp.spawn(_lr_hash_job, args=(world, 29521, d), nprocs=world,
+        from conftest import free_port
+        mp.spawn(_lr_hash_job, args=(world, free_port(), d), nprocs=world,
                  join=True)
         total_nonzero = 0
         for r in range(world):
