@@ -519,6 +519,16 @@ __global__ __launch_bounds__(256) void k_scale_rows_by_count(
   }
 }
 
+// recA[j].w = posB[j]: the pass-B record position rides in the pass-A
+// record's unused 4th int, so the emit-into-recB walk needs NO extra
+// position load (the int4 record is already in registers)
+__global__ __launch_bounds__(256) void k_pack_posw(
+    int4* __restrict__ recA, const int* __restrict__ posB, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x)
+    recA[j].w = posB[j];
+}
+
 // pos[order[j]] = j  (inverse of a sort permutation)
 __global__ __launch_bounds__(256) void k_invert_order(
     const int64_t* __restrict__ order, int* __restrict__ pos, int64_t n) {
@@ -584,8 +594,7 @@ __global__ __launch_bounds__(256) void k_seg_fused(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride, int rep_mask, int4* __restrict__ recB,
-    const int* __restrict__ posAB) {
+    int64_t rep_stride, int rep_mask, int4* __restrict__ recB) {
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
   const int64_t gpb = blockDim.x / GP;
@@ -679,9 +688,14 @@ __global__ __launch_bounds__(256) void k_seg_fused(
                 ? (__int_as_float(rj.z) - dot)
                 : lrc * (__int_as_float(rj.z) -
                          1.0f / (1.0f + __expf(-dot)));
-        if (lane == 0) {
-          if (recB) recB[posAB[j]].z = __float_as_int(g);
-          else rec[j].z = __float_as_int(g);
+        if (recB) {
+          // every lane holds g (the group_sum broadcast); rotate the
+          // emitting lane so the scattered stores spread across the
+          // group instead of serializing on lane 0
+          if (lane == ((int)j & (GP - 1)))
+            recB[rj.w].z = __float_as_int(g);
+        } else if (lane == 0) {
+          rec[j].z = __float_as_int(g);
         }
       }
 #pragma unroll
@@ -706,8 +720,7 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride, int rep_mask, int4* __restrict__ recB,
-    const int* __restrict__ posAB) {
+    int64_t rep_stride, int rep_mask, int4* __restrict__ recB) {
   extern __shared__ int4 srec[];
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
@@ -808,9 +821,11 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
                 ? (__int_as_float(rj.z) - dot)
                 : lrc * (__int_as_float(rj.z) -
                          1.0f / (1.0f + __expf(-dot)));
-        if (lane == 0) {
-          if (recB) recB[posAB[start + lj]].z = __float_as_int(g);
-          else ((int*)&rec[start + lj])[2] = __float_as_int(g);
+        if (recB) {
+          if (lane == (lj & (GP - 1)))
+            recB[rj.w].z = __float_as_int(g);
+        } else if (lane == 0) {
+          ((int*)&rec[start + lj])[2] = __float_as_int(g);
         }
       }
 #pragma unroll
@@ -1406,7 +1421,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                       const torch::Tensor& other, int64_t dim4,
                       double scale, double a, double lrc,
                       int64_t rep_stride = 0, int rep_mask = 7,
-                      int4* recB = nullptr, const int* posAB = nullptr) {
+                      int4* recB = nullptr) {
   int64_t n = rec.size(0);
   if (!n) return;
   if (dim4 == 0) {
@@ -1436,7 +1451,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
             (const float4*)keyed.data_ptr<float>(),
             (const float4*)other.data_ptr<float>(), n, dim4,
             (float)scale, (float)a, (float)lrc, chunk, rep_stride / 4,
-            rep_mask, recB, posAB);
+            rep_mask, recB);
         return;
       }
     }
@@ -1447,7 +1462,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                        (const float4*)keyed.data_ptr<float>(),
                        (const float4*)other.data_ptr<float>(), n, dim4,
                        (float)scale, (float)a, (float)lrc, chunk,
-                       rep_stride / 4, rep_mask, recB, posAB);
+                       rep_stride / 4, rep_mask, recB);
   });
 }
 
@@ -1621,9 +1636,11 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       TORCH_CHECK(histB.defined() || m > PACK_MAX_M || U > PACK_MAX_M,
                   "histB expected for LDS-sized shapes");
       recB = build_rec_swap(recA, U, histB, posB);
-      // posB is indexed by recA position (the swap pack walks recA),
-      // so it IS the pass-A-position -> recB-position map the emit
-      // needs — no composition required
+      // posB is indexed by recA position (the swap pack walks recA) —
+      // pack it into recA.w so the emit needs no extra load at all
+      hipLaunchKernelGGL(k_pack_posw, dim3(grid_for(n, 256)), dim3(256),
+                         0, cur_stream(), (int4*)recA.data_ptr<int>(),
+                         posB.data_ptr<int>(), n);
       posAB = posB;
       cntB = histB;
       if (g_pos_cache.size() >= 32) g_pos_cache.clear();
@@ -1639,17 +1656,16 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   int64_t reps = dim4 ? delta_reps(n, m, dim) : 1;
   torch::Tensor deltas;
   int4* eb = have_recB ? (int4*)recB.data_ptr<int>() : nullptr;
-  const int* epos = have_recB ? posAB.data_ptr<int>() : nullptr;
   if (reps > 1) {
     auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
     seg_fused<0, true, true>(rep_buf, recA, item_vecs, user_table, dim4,
                              lr, -lr * reg, 0.0, m * dim, (int)reps - 1,
-                             eb, epos);
+                             eb);
     deltas = rep_buf.view({reps, m, dim}).sum(0);
   } else {
     deltas = torch::zeros_like(item_vecs);
     seg_fused<0, true, true>(deltas, recA, item_vecs, user_table, dim4,
-                             lr, -lr * reg, 0.0, 0, 7, eb, epos);
+                             lr, -lr * reg, 0.0, 0, 7, eb);
   }
 
   // pass B: recs by user row; apply user updates in place.  The L2
