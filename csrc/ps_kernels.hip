@@ -1557,7 +1557,9 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       hit && hit->vals.defined() &&
       hit->vals.data_ptr() == ratings.data_ptr() &&
       hit->vals.numel() == ratings.numel();
-  if (hit && emit_ok && hit->recA.defined() && same_vals) {
+  const bool recA_cached =
+      hit && emit_ok && hit->recA.defined() && same_vals;
+  if (recA_cached) {
     recA = hit->recA;
   } else if (hit) {
     recA = torch::empty({n, 4}, opts_i32);
@@ -1632,6 +1634,13 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       recB = hit->recB;
       posAB = hit->posAB;
       cntB = hit->cntB;
+      if (!recA_cached)
+        // recA was re-scattered (fresh ratings buffer): restore the
+        // emit positions its .w column carries
+        hipLaunchKernelGGL(k_pack_posw, dim3(grid_for(n, 256)),
+                           dim3(256), 0, cur_stream(),
+                           (int4*)recA.data_ptr<int>(),
+                           posAB.data_ptr<int>(), n);
     } else {
       TORCH_CHECK(histB.defined() || m > PACK_MAX_M || U > PACK_MAX_M,
                   "histB expected for LDS-sized shapes");
