@@ -1341,9 +1341,12 @@ static torch::Tensor build_rec_swap(torch::Tensor src, int64_t m,
 template <class F>
 static void by_group4(int64_t dim4, F f) {
   using namespace std;
-  // FPS_SMALL_GP=4: 4-lane groups (2 float4/lane) for dim<=32 — twice
-  // the independent chunk walks per wave at one fewer shfl per dot
-  static int small_gp = env_int("FPS_SMALL_GP", 8);
+  // FPS_SMALL_GP: lane-group width for dim<=32.  Default 4 (2 float4/
+  // lane): twice the independent chunk walks per wave at one fewer
+  // shfl per dot — re-swept in round 2 after the emit-into-recB walk
+  // (4: 14.5 G samples/s vs 8: 14.0 on mf32; round 1 had measured no
+  // difference pre-emit)
+  static int small_gp = env_int("FPS_SMALL_GP", 4);
   if (dim4 <= 8) {
     if (small_gp == 4)
       f(integral_constant<int, 4>{}, integral_constant<int, 2>{});
