@@ -246,3 +246,69 @@ def test_lr_hash_resolver_two_ranks():
             assert res["accum_sum"] > 0  # adagrad state actually moved
             total_nonzero += res["nonzero_rows"]
         assert total_nonzero > 1000
+
+
+def _nccl_world1_job(rank, world, port, outdir):
+    """world-1 over the REAL nccl(=RCCL) backend: the closest 1-GPU
+    exercise of the branch the driver's 8-GPU scale run takes —
+    communicator init, device-resident counts all_gather, the
+    TorchDistTransport nccl device selection (no CPU staging)."""
+    import torch.distributed as dist
+    dist.init_process_group("nccl", init_method=f"tcp://127.0.0.1:{port}",
+                            rank=rank, world_size=world)
+    try:
+        from flink_parameter_server_amd import (PSConfig, PSEngine,
+                                                ShardedTable,
+                                                TableServerLogic)
+        from flink_parameter_server_amd.models.mf import (
+            OnlineMFWorkerLogic, epoch_iterator)
+        from flink_parameter_server_amd.parallel.partitioner import \
+            ModPartitioner
+        from flink_parameter_server_amd.parallel.transport import \
+            TorchDistTransport
+        from flink_parameter_server_amd.utils import synth
+
+        torch.cuda.set_device(0)
+        dev = torch.device("cuda", 0)
+        n_users, n_items, k = 80, 60, 8
+        batches, _, _ = synth.planted_mf_ratings(n_users, n_items, k,
+                                                 6000, 512, seed=17,
+                                                 device=dev)
+        part = ModPartitioner(n_items, world)
+        table = ShardedTable(part, 0, k, dev, init=(-0.1, 0.1))
+        wl = OnlineMFWorkerLogic(n_users, k, learning_rate=0.05,
+                                 range_min=-0.1, range_max=0.1,
+                                 n_items=n_items, device=dev)
+        transport = TorchDistTransport()  # nccl => device transport
+        assert transport.device.type == "cuda"
+        assert not transport.stage_via_cpu
+        eng = PSEngine(PSConfig(num_ids=n_items, dim=k,
+                                feedback="pipelined"), wl,
+                       TableServerLogic(table, dump_on_close=False,
+                                        pushes_unique=True),
+                       transport, part, device=dev)
+        eng.run(epoch_iterator(batches, 4))
+        torch.cuda.synchronize()
+        gids, vals = table.dump()
+        full = torch.zeros(n_items, k, device=dev)
+        full[gids.to(dev)] = vals
+        tot, n = 0.0, 0
+        for b in batches:
+            pred = (wl.user_table[b.user_rows.long()] *
+                    full[b.item_ids.long()]).sum(1)
+            tot += float(((pred - b.ratings) ** 2).sum())
+            n += len(b)
+        torch.save({"loss": tot / n},
+                   os.path.join(outdir, "rank0.pt"))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_nccl_world1_engine():
+    with tempfile.TemporaryDirectory() as d:
+        from conftest import free_port
+        mp.spawn(_nccl_world1_job, args=(1, free_port(), d), nprocs=1,
+                 join=True)
+        res = torch.load(os.path.join(d, "rank0.pt"), weights_only=False)
+        assert res["loss"] < 0.08, res
