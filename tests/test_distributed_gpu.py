@@ -287,7 +287,7 @@ def _nccl_world1_job(rank, world, port, outdir):
                        TableServerLogic(table, dump_on_close=False,
                                         pushes_unique=True),
                        transport, part, device=dev)
-        eng.run(epoch_iterator(batches, 4))
+        eng.run(epoch_iterator(batches, 8))
         torch.cuda.synchronize()
         gids, vals = table.dump()
         full = torch.zeros(n_items, k, device=dev)
