@@ -1502,6 +1502,9 @@ struct PosCacheEntry {
   // gated on the ratings tensor's identity (``vals`` holds the
   // reference — a different ratings buffer re-scatters recA only).
   torch::Tensor recA, recB, posAB, vals;
+  // touched-rows L2-decay extents (huge sparse-touched user tables):
+  // depend only on recB's static key column — computed once per batch
+  torch::Tensor trows, tcnts, ntouch;
 };
 // exact composite key (no hash collisions): the held tensor references
 // inside each entry keep the keyed storages alive, so a pointer can
@@ -1657,7 +1660,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       cntB = histB;
       if (g_pos_cache.size() >= 32) g_pos_cache.clear();
       g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB,
-                         recA, recB, posAB, ratings};
+                         recA, recB, posAB, ratings, {}, {}, {}};
     }
     have_recB = true;
   }
@@ -1701,7 +1704,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
       if (reuse_tag) {
         if (g_pos_cache.size() >= 32) g_pos_cache.clear();
         g_pos_cache[ck] = {item_slots, user_rows, posA, posB, histB,
-                           {}, {}, {}, {}};
+                           {}, {}, {}, {}, {}, {}, {}};
       }
     }
   }
@@ -1728,15 +1731,29 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                          cc.data_ptr<int>(), U, dim, (float)(-lr * reg));
     } else {
       // huge sparse-touched table: decay only the rows this batch
-      // touches, from the sorted recs' segment extents
-      auto trows = torch::empty({n}, opts_i32);
-      auto tcnts = torch::empty({n}, opts_i32);
-      auto ntouch = torch::zeros({1}, opts_i32);
-      hipLaunchKernelGGL(k_seg_extents, dim3(grid_for(n, 256)),
-                         dim3(256), 0, cur_stream(),
-                         (const int4*)recB.data_ptr<int>(), n,
-                         trows.data_ptr<int>(), tcnts.data_ptr<int>(),
-                         ntouch.data_ptr<int>());
+      // touches, from the sorted recs' segment extents.  The extents
+      // depend only on recB's STATIC key column, so revisits reuse
+      // them from the position cache (k_seg_extents was 60% of the
+      // HBM-regime mf32 round — profiles/PERF_NOTES.md round 2)
+      torch::Tensor trows, tcnts, ntouch;
+      PosCacheEntry* e = nullptr;
+      if (reuse_tag) {
+        auto it = g_pos_cache.find(ck);
+        if (it != g_pos_cache.end()) e = &it->second;
+      }
+      if (e && e->trows.defined()) {
+        trows = e->trows; tcnts = e->tcnts; ntouch = e->ntouch;
+      } else {
+        trows = torch::empty({n}, opts_i32);
+        tcnts = torch::empty({n}, opts_i32);
+        ntouch = torch::zeros({1}, opts_i32);
+        hipLaunchKernelGGL(k_seg_extents, dim3(grid_for(n, 256)),
+                           dim3(256), 0, cur_stream(),
+                           (const int4*)recB.data_ptr<int>(), n,
+                           trows.data_ptr<int>(), tcnts.data_ptr<int>(),
+                           ntouch.data_ptr<int>());
+        if (e) { e->trows = trows; e->tcnts = tcnts; e->ntouch = ntouch; }
+      }
       hipLaunchKernelGGL(k_scale_touched_rows,
                          dim3(grid_for(n * dim, 256)), dim3(256), 0,
                          cur_stream(), user_table.data_ptr<float>(),
