@@ -249,6 +249,14 @@ __global__ __launch_bounds__(256) void k_apply_adagrad_sorted(
 // Sub-wave group reduction: sum over a lane group of GP lanes (GP a
 // power of two <= 64).
 // ---------------------------------------------------------------------
+__device__ __forceinline__ float4 ld_f4(const float4* p, bool nt) {
+  if (nt) {
+    v4f v = __builtin_nontemporal_load((const v4f*)p);
+    return make_float4(v.x, v.y, v.z, v.w);
+  }
+  return *p;
+}
+
 template <int GP>
 __device__ __forceinline__ float group_sum(float v) {
 #pragma unroll
@@ -617,7 +625,8 @@ __global__ __launch_bounds__(256) void k_seg_fused(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride, int rep_mask, int4* __restrict__ recB) {
+    int64_t rep_stride, int rep_mask, int4* __restrict__ recB,
+    bool nt_rows) {
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
   const int64_t gpb = blockDim.x / GP;
@@ -643,7 +652,8 @@ __global__ __launch_bounds__(256) void k_seg_fused(
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
         int c = lane + e * GP;
-        kv[e] = (c < dim4) ? keyed[kr + c] : make_float4(0, 0, 0, 0);
+        kv[e] = (c < dim4) ? ld_f4(&keyed[kr + c], nt_rows)
+                           : make_float4(0, 0, 0, 0);
       }
     };
     auto flush = [&](bool right_owned) {
@@ -695,7 +705,8 @@ __global__ __launch_bounds__(256) void k_seg_fused(
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
         int c = lane + e * GP;
-        sv[e] = (c < dim4) ? other[sr + c] : make_float4(0, 0, 0, 0);
+        sv[e] = (c < dim4) ? ld_f4(&other[sr + c], nt_rows)
+                           : make_float4(0, 0, 0, 0);
       }
       float g;
       if (MODE == 2) {
@@ -743,7 +754,8 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride, int rep_mask, int4* __restrict__ recB) {
+    int64_t rep_stride, int rep_mask, int4* __restrict__ recB,
+    bool nt_rows) {
   extern __shared__ int4 srec[];
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
@@ -776,7 +788,8 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
         int c = lane + e * GP;
-        kv[e] = (c < dim4) ? keyed[kr + c] : make_float4(0, 0, 0, 0);
+        kv[e] = (c < dim4) ? ld_f4(&keyed[kr + c], nt_rows)
+                           : make_float4(0, 0, 0, 0);
       }
     };
     auto flush = [&](bool right_owned) {
@@ -828,7 +841,8 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
         int c = lane + e * GP;
-        sv[e] = (c < dim4) ? other[sr + c] : make_float4(0, 0, 0, 0);
+        sv[e] = (c < dim4) ? ld_f4(&other[sr + c], nt_rows)
+                           : make_float4(0, 0, 0, 0);
       }
       float g;
       if (MODE == 2) {
@@ -1463,6 +1477,12 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
   int chunk = seg_chunk();
   int64_t nchunks = (n + chunk - 1) / chunk;
   static int use_lds = env_int("FPS_SEG_LDS", 1);
+  // nontemporal row loads in the walks: swept on the mf128 streaming
+  // shape and REJECTED (842 -> 832 M samples/s — the same rows recur
+  // across the two passes within a round, so L2 retention still pays);
+  // kept env-gated for future shapes
+  static int nt_env = env_int("FPS_WALK_NT", 0);
+  const bool nt_rows = nt_env != 0;
   by_group4(dim4, [&](auto tag, auto etag) {
     constexpr int GP = decltype(tag)::value;
     constexpr int EPL = decltype(etag)::value;
@@ -1477,7 +1497,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
             (const float4*)keyed.data_ptr<float>(),
             (const float4*)other.data_ptr<float>(), n, dim4,
             (float)scale, (float)a, (float)lrc, chunk, rep_stride / 4,
-            rep_mask, recB);
+            rep_mask, recB, nt_rows);
         return;
       }
     }
@@ -1488,7 +1508,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                        (const float4*)keyed.data_ptr<float>(),
                        (const float4*)other.data_ptr<float>(), n, dim4,
                        (float)scale, (float)a, (float)lrc, chunk,
-                       rep_stride / 4, rep_mask, recB);
+                       rep_stride / 4, rep_mask, recB, nt_rows);
   });
 }
 
