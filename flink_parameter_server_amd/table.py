@@ -79,7 +79,12 @@ class ShardedTable:
             # itself.  Answer values alias the live table; the engine's
             # round order (worker consumes answers BEFORE this round's
             # pushes apply) makes that observationally identical to a
-            # copy taken at answer time.
+            # copy taken at answer time under fresh/merged schedules.
+            # Under "pipelined" a self-addressed alias is consumed one
+            # round later, after one more local push-apply — the worker
+            # then sees FRESHER values than gather time, which the
+            # async pull contract allows (answers carry no freshness
+            # guarantee; remote segments are real copies either way).
             return self.data
         rows = self.part.local_row(ids)
         return ops.gather(self.data, rows)
