@@ -408,3 +408,48 @@ def train_lr(data, feature_count: int, learning_rate: float = 0.05,
                                 n_features=feature_count),
         make_server, cfg, part_factory=lambda rank: part)
     return _merged(results)
+
+
+def run_count_min(id_stream, depth: int, width: int,
+                  worker_parallelism: int = 1,
+                  ps_parallelism: Optional[int] = None,
+                  batch_size: int = 4096, seed: int = 17):
+    """Count-min sketch on the PS (reference ``sketch/**`` analog —
+    SURVEY.md §2.2, built defensively; see models/sketch.py).
+
+    ``id_stream``: iterable of int ids (or id tensors/lists).  Returns
+    ``(CountMinWorkerLogic, merged_table)`` — query with
+    ``logic.estimate(merged_table, ids)`` (one-sided: never an
+    under-estimate)."""
+    from .models.sketch import CountMinWorkerLogic, merge_shard_dumps
+
+    world = int(worker_parallelism)
+    shards = int(ps_parallelism or world)
+    if shards > world:
+        raise ValueError("ps_parallelism > worker_parallelism needs one "
+                         "rank per shard; raise worker_parallelism")
+    ids = torch.as_tensor(list(id_stream), dtype=torch.int64).reshape(-1)
+    batches = list(ids.split(batch_size))
+    per_rank = [batches[r::world] for r in range(world)]
+    probe = CountMinWorkerLogic(depth, width, seed=seed)
+    part = make_partitioner("mod", probe.num_ids, shards)
+    cfg = PSConfig(n_workers=world, n_shards=shards,
+                   num_ids=probe.num_ids, dim=1)
+    logics: List[Any] = [None] * world
+
+    def make_worker(rank: int):
+        logics[rank] = CountMinWorkerLogic(depth, width, seed=seed)
+        return logics[rank]
+
+    def make_server(rank: int) -> TableServerLogic:
+        table = ShardedTable(part, rank, 1, torch.device("cpu"))
+        return TableServerLogic(table, dump_on_close=True,
+                                pushes_unique=True)
+
+    results = transform_local(
+        world, lambda r: iter(per_rank[r]), make_worker, make_server,
+        cfg, part_factory=lambda rank: part)
+    dumps = []
+    for _, souts in results:
+        dumps.extend(souts)
+    return logics[0], merge_shard_dumps(dumps, depth, width)

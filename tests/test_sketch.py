@@ -87,3 +87,17 @@ def test_count_min_sharded_across_ranks(world):
     est = wl.estimate(table, torch.arange(2000))
     assert (est >= true - 1e-5).all()
     assert float((est - true).max()) <= 6 * 2.718 / width * n
+
+
+def test_run_count_min_entry():
+    """algorithms.run_count_min entry: 2 workers x 2 shards through
+    transform_local, merged dump queryable."""
+    from flink_parameter_server_amd.algorithms import run_count_min
+
+    stream = zipf_stream(20000, 1500, seed=9)
+    logic, table = run_count_min(stream.tolist(), depth=4, width=1024,
+                                 worker_parallelism=2)
+    true = torch.bincount(stream, minlength=1500).float()
+    est = logic.estimate(table, torch.arange(1500))
+    assert (est >= true - 1e-5).all()
+    assert float((est - true).max()) <= 6 * 2.718 / 1024 * 20000
