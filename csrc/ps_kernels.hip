@@ -1063,6 +1063,131 @@ __global__ __launch_bounds__(256) void k_sparse_linear_step(
 }
 
 // ---------------------------------------------------------------------
+// K7b/K9b: sparse-linear step over SLOT-SORTED records (the static-
+// batch fast path).  The per-element form gathers weights[slot] per
+// nnz element (a random 4-B read = a 64-B line each) and scatters
+// per-element delta atomics into an >L2 buffer — measured line
+// amplification bound.  Sorted by slot (positions cached per static
+// batch), the weight is read ONCE per segment into a register and the
+// delta flushes ONCE per (segment x chunk) with exclusive-store/atomic
+// selection exactly like the MF walk; the per-instance margin is
+// assembled with one L2-resident atomicAdd per element.
+//   rec = {slot, instance, x_bits, 0} sorted by slot
+//   pass 1: margins[inst] += w[slot] * x      (w register-cached)
+//   coeff:  per instance from margin, STATIC ||x||^2 and the label
+//   pass 2: deltas[slot] += coeff[inst] * x   (segment-accumulated)
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_lin_margin(
+    const int4* __restrict__ rec, const float* __restrict__ weights,
+    float* __restrict__ margins, int64_t n, int chunk) {
+  const int64_t nchunks = (n + chunk - 1) / chunk;
+  for (int64_t cidx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       cidx < nchunks; cidx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t start = cidx * chunk;
+    const int64_t end = start + chunk < n ? start + chunk : n;
+    int cur = rec[start].x;
+    float w = weights[cur];
+    for (int64_t j = start; j < end; ++j) {
+      const int4 rj = rec[j];
+      if (rj.x != cur) {
+        cur = rj.x;
+        w = weights[cur];
+      }
+      atomicAdd(&margins[rj.y], w * __int_as_float(rj.z));
+    }
+  }
+}
+
+// coeff from assembled margins (margins then hold the raw margin for
+// the caller's prediction output).  mode: 0/1/2 = PA variants (labels
+// +-1; NaN = unlabeled), 3 = logistic (labels 0/1; NaN = unlabeled).
+__global__ __launch_bounds__(256) void k_lin_coeff(
+    const float* __restrict__ margins, const float* __restrict__ n2,
+    const float* __restrict__ labels, float* __restrict__ coeff,
+    int64_t B, int mode, float C) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < B; i += (int64_t)gridDim.x * blockDim.x) {
+    const float y = labels[i];
+    const float m = margins[i];
+    float out = 0.f;
+    if (!isnan(y)) {
+      if (mode == 3) {
+        out = 1.0f / (1.0f + __expf(-m)) - y;
+      } else {
+        float loss = 1.0f - y * m;
+        if (loss > 0.f) {
+          float sn = n2[i] > 1e-12f ? n2[i] : 1e-12f;
+          float tau;
+          if (mode == 0) tau = loss / sn;
+          else if (mode == 1) { tau = loss / sn; tau = tau > C ? C : tau; }
+          else tau = loss / (sn + 1.0f / (2.0f * C));
+          out = tau * y;
+        }
+      }
+    }
+    coeff[i] = out;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_lin_delta(
+    const int4* __restrict__ rec, const float* __restrict__ coeff,
+    float* __restrict__ deltas, int64_t n, int chunk, float sign) {
+  const int64_t nchunks = (n + chunk - 1) / chunk;
+  for (int64_t cidx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       cidx < nchunks; cidx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t start = cidx * chunk;
+    const int64_t end = start + chunk < n ? start + chunk : n;
+    int cur = rec[start].x;
+    bool left_owned = (start == 0);
+    float acc = 0.f;
+    auto flush = [&](bool right_owned) {
+      if (left_owned && right_owned) deltas[cur] = acc;
+      else atomicAdd(&deltas[cur], acc);
+      acc = 0.f;
+    };
+    for (int64_t j = start; j < end; ++j) {
+      const int4 rj = rec[j];
+      if (rj.x != cur) {
+        flush(true);
+        left_owned = true;
+        cur = rj.x;
+      }
+      acc += coeff[rj.y] * __int_as_float(rj.z) * sign;
+    }
+    flush(end == n);
+  }
+}
+
+// static per-batch precomputation: n2[i] = sum_j x_j^2 per instance
+__global__ __launch_bounds__(256) void k_lin_n2(
+    const int64_t* __restrict__ indptr, const float* __restrict__ x,
+    float* __restrict__ n2, int64_t B) {
+  const int lane = threadIdx.x % WAVE;
+  const int64_t wpb = blockDim.x / WAVE;
+  for (int64_t i = (int64_t)blockIdx.x * wpb + threadIdx.x / WAVE;
+       i < B; i += (int64_t)gridDim.x * wpb) {
+    const int64_t lo = indptr[i], hi = indptr[i + 1];
+    float s = 0.f;
+    for (int64_t j = lo + lane; j < hi; j += WAVE) s += x[j] * x[j];
+    s = group_sum<WAVE>(s);
+    if (lane == 0) n2[i] = s;
+  }
+}
+
+// instance column for the record build: seg[j] = its instance index
+__global__ __launch_bounds__(256) void k_lin_seg(
+    const int64_t* __restrict__ indptr, int* __restrict__ seg,
+    int64_t B) {
+  const int lane = threadIdx.x % WAVE;
+  const int64_t wpb = blockDim.x / WAVE;
+  for (int64_t i = (int64_t)blockIdx.x * wpb + threadIdx.x / WAVE;
+       i < B; i += (int64_t)gridDim.x * wpb) {
+    const int64_t lo = indptr[i], hi = indptr[i + 1];
+    for (int64_t j = lo + lane; j < hi; j += WAVE) seg[j] = (int)i;
+  }
+}
+
+// ---------------------------------------------------------------------
 // K7 (multiclass): one wave per instance; class columns are walked in
 // tiles of 64 (lane c of tile t owns class t*64+c — any n_classes).
 // scores[c] = sum_j w[slot_j][c] * x_j — lanes read consecutive class
@@ -1946,6 +2071,73 @@ std::vector<torch::Tensor> sgns_expand(torch::Tensor centers,
   return {oc, ox, ol};
 }
 
+// per-static-batch cache for the slot-sorted linear step: the record
+// array (slot, instance, x) and the per-instance ||x||^2 are fully
+// static; only the pulled weights and labels' effect change per visit
+struct LinCacheEntry {
+  torch::Tensor slots, xvals, rec, n2;
+};
+static std::map<PosCacheKey, LinCacheEntry> g_lin_cache;
+
+static int lin_chunk() {
+  static int c = env_int("FPS_LIN_CHUNK", 1024);
+  return c;
+}
+
+static std::vector<torch::Tensor> linear_step_cached(
+    torch::Tensor indptr, torch::Tensor feat_slots, torch::Tensor x_vals,
+    torch::Tensor labels, torch::Tensor weights, int mode, float C,
+    int64_t reuse_tag) {
+  int64_t B = indptr.numel() - 1, n = x_vals.numel();
+  int64_t m = weights.numel();
+  auto opts_i32 = torch::dtype(torch::kInt32).device(x_vals.device());
+  PosCacheKey ck{reuse_tag, feat_slots.data_ptr(), x_vals.data_ptr(), n};
+  auto it = g_lin_cache.find(ck);
+  LinCacheEntry* e = it != g_lin_cache.end() ? &it->second : nullptr;
+  torch::Tensor rec, n2;
+  if (e) {
+    rec = e->rec;
+    n2 = e->n2;
+  } else {
+    auto seg = torch::empty({n}, opts_i32);
+    hipLaunchKernelGGL(k_lin_seg, dim3(grid_for(B * WAVE, 256)),
+                       dim3(256), 0, cur_stream(),
+                       indptr.data_ptr<int64_t>(), seg.data_ptr<int>(),
+                       B);
+    n2 = torch::empty({B}, x_vals.options());
+    hipLaunchKernelGGL(k_lin_n2, dim3(grid_for(B * WAVE, 256)),
+                       dim3(256), 0, cur_stream(),
+                       indptr.data_ptr<int64_t>(),
+                       x_vals.data_ptr<float>(), n2.data_ptr<float>(),
+                       B);
+    auto fs32 = feat_slots.dtype() == torch::kInt32
+                    ? feat_slots : feat_slots.to(torch::kInt32);
+    rec = build_rec(fs32, seg, x_vals, m);
+    if (g_lin_cache.size() >= 32) g_lin_cache.clear();
+    g_lin_cache[ck] = {feat_slots, x_vals, rec, n2};
+  }
+  auto margins = torch::zeros({B}, x_vals.options());
+  int chunk = lin_chunk();
+  int64_t nchunks = (n + chunk - 1) / chunk;
+  hipLaunchKernelGGL(k_lin_margin, dim3(grid_for(nchunks, 256)),
+                     dim3(256), 0, cur_stream(),
+                     (const int4*)rec.data_ptr<int>(),
+                     weights.data_ptr<float>(),
+                     margins.data_ptr<float>(), n, chunk);
+  auto coeff = torch::empty({B}, x_vals.options());
+  hipLaunchKernelGGL(k_lin_coeff, dim3(grid_for(B, 256)), dim3(256), 0,
+                     cur_stream(), margins.data_ptr<float>(),
+                     n2.data_ptr<float>(), labels.data_ptr<float>(),
+                     coeff.data_ptr<float>(), B, mode, C);
+  auto deltas = torch::zeros({m}, weights.options());
+  hipLaunchKernelGGL(k_lin_delta, dim3(grid_for(nchunks, 256)),
+                     dim3(256), 0, cur_stream(),
+                     (const int4*)rec.data_ptr<int>(),
+                     coeff.data_ptr<float>(), deltas.data_ptr<float>(),
+                     n, chunk, 1.0f);
+  return {margins, deltas};
+}
+
 static int pa_mode(const std::string& variant) {
   if (variant == "PA") return 0;
   if (variant == "PA-I") return 1;
@@ -1988,9 +2180,16 @@ std::vector<torch::Tensor> lr_step(torch::Tensor indptr,
                                    torch::Tensor feat_slots,
                                    torch::Tensor x_vals,
                                    torch::Tensor labels,
-                                   torch::Tensor weights) {
+                                   torch::Tensor weights,
+                                   int64_t reuse_tag) {
   CHECK_DEV(weights); CHECK_CONT(weights);
   int64_t B = indptr.numel() - 1;
+  if (B && reuse_tag) {
+    auto out = linear_step_cached(indptr, feat_slots, x_vals, labels,
+                                  weights, /*mode=*/3, 0.f, reuse_tag);
+    out[0].sigmoid_();
+    return out;
+  }
   auto probs = torch::zeros({B}, x_vals.options());
   if (!B) return {probs, torch::zeros_like(weights)};
   int64_t m = weights.numel();
@@ -2065,5 +2264,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K10b slot resolution against an existing coalescer table");
   m.def("pa_step", &pa_step, "K7 PA binary step");
   m.def("pa_multiclass_step", &pa_multiclass_step, "K7 PA multiclass");
-  m.def("lr_step", &lr_step, "K9 sparse logistic step");
+  m.def("lr_step", &lr_step, "K9 sparse logistic step",
+        py::arg("indptr"), py::arg("feat_slots"), py::arg("x_vals"),
+        py::arg("labels"), py::arg("weights"), py::arg("reuse_tag") = 0);
 }

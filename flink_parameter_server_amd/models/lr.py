@@ -106,8 +106,16 @@ class LRWorkerLogic(WorkerLogic):
                slots: Optional[torch.Tensor] = None) -> None:
         if slots is None:
             slots = buf.slots_of(batch.feat_ids)
+        # static batches with stable cached slots unlock the kernel's
+        # slot-sorted record cache (see ops.lr_step reuse_tag)
+        tag = 0
+        if self.assume_static:
+            hit = self._slot_cache.get(id(batch))
+            if hit is not None and hit[2] is slots:
+                tag = id(batch)
         probs, grads = ops.lr_step(batch.indptr, slots, batch.x_vals,
-                                   batch.labels, buf.values.reshape(-1))
+                                   batch.labels, buf.values.reshape(-1),
+                                   reuse_tag=tag)
         if self.emit_predictions:
             unlabeled = torch.isnan(batch.labels)
             if bool(unlabeled.any()):

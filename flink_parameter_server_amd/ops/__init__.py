@@ -130,9 +130,16 @@ def pa_multiclass_step(indptr: torch.Tensor, feat_slots: torch.Tensor,
 
 def lr_step(indptr: torch.Tensor, feat_slots: torch.Tensor,
             x_vals: torch.Tensor, labels: torch.Tensor,
-            weights: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+            weights: torch.Tensor, reuse_tag: int = 0
+            ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """``reuse_tag != 0`` promises the same (tag, feat_slots buffer,
+    x_vals buffer) recur with unchanged content (epoch resweeps); the
+    HIP path then runs the slot-sorted record walk with the records and
+    per-instance norms cached per batch — the per-element gather/
+    scatter form (random 4-B line traffic) leaves the steady round."""
     if _cuda(weights):
-        return hip.ext().lr_step(indptr, feat_slots, x_vals, labels, weights)
+        return hip.ext().lr_step(indptr, feat_slots, x_vals, labels,
+                                 weights, reuse_tag)
     return reference.lr_step(indptr, feat_slots.long(), x_vals, labels,
                              weights)
 

@@ -512,3 +512,61 @@ def test_lr_hash_path_matches_cpu_reference():
     assert torch.allclose(t_cpu.accum[rows],
                           t_gpu.accum[rows.to(DEV)].cpu(),
                           atol=1e-4, rtol=1e-4)
+
+
+def test_lr_static_batches_sorted_walk_matches_cpu():
+    """assume_static_batches on GPU routes lr_step through the cached
+    slot-sorted record walk (K9b) on revisits; the trained weights and
+    adagrad state must match the CPU reference stream exactly."""
+    from flink_parameter_server_amd import (PSConfig, PSEngine,
+                                            ShardedTable,
+                                            TableServerLogic)
+    from flink_parameter_server_amd.models.lr import LRWorkerLogic
+    from flink_parameter_server_amd.models.pa import SparseBatch
+    from flink_parameter_server_amd.parallel.partitioner import \
+        ModPartitioner
+
+    n_feat = 1 << 28
+    g = torch.Generator().manual_seed(23)
+    raw = []
+    for _ in range(3):
+        B, nnz = 256, 16
+        feats = (n_feat * torch.rand(B * nnz, generator=g).pow(2.0)
+                 ).long().clamp_(0, n_feat - 1)
+        vals = torch.randn(B * nnz, generator=g)
+        indptr = torch.arange(0, B * nnz + 1, nnz, dtype=torch.int64)
+        labels = (torch.rand(B, generator=g) < 0.5).float()
+        labels[::17] = float("nan")  # some unlabeled -> predictions
+        raw.append((indptr, feats, vals, labels))
+
+    def run(dev, static):
+        part = ModPartitioner(n_feat, 1)
+        table = ShardedTable(part, 0, 1, dev, optimizer="adagrad",
+                             lr=0.1)
+        wl = LRWorkerLogic(n_features=n_feat, device=torch.device(dev),
+                           assume_static_batches=static,
+                           emit_predictions=True)
+        eng = PSEngine(PSConfig(num_ids=n_feat, dim=1), wl,
+                       TableServerLogic(table, dump_on_close=False,
+                                        pushes_unique=True),
+                       None, part, device=torch.device(dev))
+        data = [SparseBatch(i.to(dev), f.to(dev), v.to(dev), l.to(dev))
+                for i, f, v, l in raw]
+        wout, _ = eng.run(data * 3)  # 3 epochs -> revisits hit the cache
+        return table, wout
+
+    t_gpu, wout_gpu = run(DEV, True)
+    torch.cuda.synchronize()
+    t_cpu, wout_cpu = run("cpu", False)
+    touched = torch.unique(torch.cat([b[1] for b in raw]))
+    rows = t_cpu.part.local_row(touched)
+    assert torch.allclose(t_cpu.data[rows],
+                          t_gpu.data[rows.to(DEV)].cpu(),
+                          atol=1e-4, rtol=1e-4)
+    assert torch.allclose(t_cpu.accum[rows],
+                          t_gpu.accum[rows.to(DEV)].cpu(),
+                          atol=1e-4, rtol=1e-4)
+    # unlabeled predictions present and matching
+    assert len(wout_gpu) == len(wout_cpu) > 0
+    for (gi, gp), (ci, cp) in zip(wout_gpu, wout_cpu):
+        assert torch.allclose(gp.cpu(), cp, atol=1e-4)
