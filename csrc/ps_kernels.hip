@@ -2080,7 +2080,7 @@ struct LinCacheEntry {
 static std::map<PosCacheKey, LinCacheEntry> g_lin_cache;
 
 static int lin_chunk() {
-  static int c = env_int("FPS_LIN_CHUNK", 1024);
+  static int c = env_int("FPS_LIN_CHUNK", 64);
   return c;
 }
 
