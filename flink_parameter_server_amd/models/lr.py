@@ -24,16 +24,26 @@ class LRWorkerLogic(WorkerLogic):
                  emit_predictions: bool = False,
                  n_features: int = 0,
                  assume_static_batches: bool = False,
+                 sorted_walk: bool = False,
                  device: torch.device = torch.device("cpu")):
         """``assume_static_batches=True``: the data source re-yields the
         SAME batch objects with unchanged feature tensors (epoch sweeps
         over a resident dataset); the id coalescing (uids/slots/
         resolver) is then computed once per batch and reused on every
         revisit — only the values are re-pulled.  Never set it for
-        sources that rewrite batch buffers in place."""
+        sources that rewrite batch buffers in place.
+
+        ``sorted_walk=True`` additionally routes revisits through the
+        slot-sorted record kernels (K9b).  Measured SLOWER than the
+        per-element gather kernel at the named 10^9-feature config
+        (166 vs 243 M inst/s: the pulled-weight buffer is LLC-resident,
+        so the gather path pays no HBM amplification while the sorted
+        records add 16 B/element of HBM traffic per pass) — kept as an
+        option for shapes whose weight buffer exceeds the LLC."""
         self.pull_limit = pull_limit
         self.emit_predictions = emit_predictions
         self.assume_static = bool(assume_static_batches)
+        self.sorted_walk = bool(sorted_walk)
         self._pending: List[Tuple[SparseBatch, PullBuffer,
                                   Optional[torch.Tensor]]] = []
         self._slot_cache: dict = {}
@@ -106,10 +116,11 @@ class LRWorkerLogic(WorkerLogic):
                slots: Optional[torch.Tensor] = None) -> None:
         if slots is None:
             slots = buf.slots_of(batch.feat_ids)
-        # static batches with stable cached slots unlock the kernel's
-        # slot-sorted record cache (see ops.lr_step reuse_tag)
+        # static batches with stable cached slots MAY route through the
+        # kernel's slot-sorted record cache (see ops.lr_step reuse_tag;
+        # off by default — measured slower at the named config)
         tag = 0
-        if self.assume_static:
+        if self.assume_static and self.sorted_walk:
             hit = self._slot_cache.get(id(batch))
             if hit is not None and hit[2] is slots:
                 tag = id(batch)

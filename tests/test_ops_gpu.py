@@ -545,6 +545,7 @@ def test_lr_static_batches_sorted_walk_matches_cpu():
                              lr=0.1)
         wl = LRWorkerLogic(n_features=n_feat, device=torch.device(dev),
                            assume_static_batches=static,
+                           sorted_walk=static,
                            emit_predictions=True)
         eng = PSEngine(PSConfig(num_ids=n_feat, dim=1), wl,
                        TableServerLogic(table, dump_on_close=False,
