@@ -360,3 +360,56 @@ def test_pipelined_overlapped_delivery_mf():
             tot += float(((pred - b.ratings) ** 2).sum())
             n += len(b)
     assert tot / n < 0.1, tot / n
+
+
+def test_kitchen_sink_pipelined():
+    """pipelined feedback + bf16 wire + count-flush + pull limit + SSP,
+    4 ranks: the multi-GPU default schedule with every engine feature at
+    once still delivers exactly-once answers and the right totals."""
+    from flink_parameter_server_amd.messaging import CountFlush
+
+    world, n_ids, dim = 4, 29, 2
+
+    class Traffic(WorkerLogic):
+        pull_limit = 6
+
+        def __init__(self, rank):
+            self.g = torch.Generator().manual_seed(900 + rank)
+            self.pushed = torch.zeros(n_ids, dim)
+            self.answered = 0
+
+        def on_recv(self, data, client):
+            ids = torch.unique(torch.randint(0, n_ids, (5,),
+                                             generator=self.g))
+            deltas = torch.randn(ids.numel(), dim, generator=self.g)
+            client.push(ids, deltas)
+            self.pushed[ids] += deltas
+            client.pull(ids, unique=True)
+
+        def on_pull_recv(self, ids, values, client):
+            assert values.shape == (ids.numel(), dim)
+            self.answered += int(ids.numel())
+
+    def job(rank, transport):
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, dim, "cpu")
+        logic = Traffic(rank)
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=dim,
+                                feedback="pipelined", wire_dtype="bf16",
+                                staleness=3),
+                       logic, TableServerLogic(table), transport, part,
+                       flush=CountFlush(8))
+        eng.run([r for r in range(15)])
+        assert eng.metrics.pull_ids_sent == logic.answered
+        gids, vals = table.dump()
+        return logic.pushed, dict(
+            zip(gids.tolist(), [v.clone() for v in vals]))
+
+    results = run_loopback(world, job)
+    expected = sum(r[0] for r in results)
+    merged = {}
+    for _, d in results:
+        merged.update(d)
+    got = torch.stack([merged[i] for i in range(n_ids)])
+    assert torch.allclose(got, expected, atol=0.15, rtol=0.05), \
+        (got - expected).abs().max()
