@@ -298,7 +298,9 @@ class PSEngine:
         # 3. coalesce + bucket (unless a combination-sender strategy holds
         # this round's ops back — messaging.FlushStrategy)
         if self.flush is not None and not self.flush.should_flush(
-                self._pending_ops(), force=exhausted or stalled):
+                self._pending_ops(),
+                force=(exhausted or stalled or
+                       self._pull_queue is not None)):
             m.held_rounds += 1
             pull_ids = push_ids = push_deltas = None
         else:
@@ -412,7 +414,9 @@ class PSEngine:
         # 1. bucket the ops queued by last round's callbacks (unless a
         # combination-sender strategy holds them — messaging.FlushStrategy)
         if self.flush is not None and not self.flush.should_flush(
-                self._pending_ops(), force=exhausted or self._stalled):
+                self._pending_ops(),
+                force=(exhausted or self._stalled or
+                       self._pull_queue is not None)):
             m.held_rounds += 1
             pull_ids = push_ids = push_deltas = None
         else:
@@ -547,7 +551,9 @@ class PSEngine:
 
         # 1. bucket the ops queued by last round's callbacks
         if self.flush is not None and not self.flush.should_flush(
-                self._pending_ops(), force=exhausted or self._stalled):
+                self._pending_ops(),
+                force=(exhausted or self._stalled or
+                       self._pull_queue is not None)):
             m.held_rounds += 1
             pull_ids = push_ids = push_deltas = None
         else:
@@ -723,6 +729,13 @@ class PSEngine:
         n = self._pull_queue.numel() if self._pull_queue is not None else 0
         return self._outstanding + n
 
+    # NOTE: pulls that overflowed the pull-limit budget into
+    # _pull_queue were already released by a flush decision; the flush
+    # gate must not hold them again (force includes "queue non-empty"),
+    # otherwise backpressure stops new ops, the count threshold can
+    # never be reached, and the rank deadlocks — with SSP on it then
+    # stalls every other rank too (found by the pipelined kitchen-sink
+    # test; the hazard existed in all three schedules).
     def _take_pulls(self) -> Optional[torch.Tensor]:
         parts = self._pull_buf
         self._pull_buf = []
