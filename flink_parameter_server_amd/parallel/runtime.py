@@ -9,7 +9,7 @@ quiescence timeout.
 
 MI355X design: one process per GPU; each rank hosts one
 worker role AND one server-shard role.  The free-running record flow
-becomes *rounds* with two feedback-edge schedules (``PSConfig.feedback``):
+becomes *rounds* with three feedback-edge schedules (``PSConfig.feedback``):
 
 ``"fresh"`` (default) — per round: deliver last round's answers ->
 consume a batch -> bucket this round's pulls/pushes -> exchange them
