@@ -550,14 +550,15 @@ __global__ __launch_bounds__(256) void k_scale_rows_by_count(
   }
 }
 
-// recA[j].w = posB[j]: the pass-B record position rides in the pass-A
-// record's unused 4th int, so the emit-into-recB walk needs NO extra
-// position load (the int4 record is already in registers)
-__global__ __launch_bounds__(256) void k_pack_posw(
-    int4* __restrict__ recA, const int* __restrict__ posB, int64_t n) {
-  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
-       j += (int64_t)gridDim.x * blockDim.x)
-    recA[j].w = posB[j];
+// recB[posB[a]].w = a: each pass-B record carries its pass-A position
+// in the spare 4th int, so pass B reads its coefficient as
+// ecol[rec.w] — a gather from the L2-resident sequential e column (the
+// int4 record is already in registers; no extra index load)
+__global__ __launch_bounds__(256) void k_pack_wposB(
+    int4* __restrict__ recB, const int* __restrict__ posB, int64_t n) {
+  for (int64_t a = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; a < n;
+       a += (int64_t)gridDim.x * blockDim.x)
+    recB[posB[a]].w = (int)a;
 }
 
 // pos[order[j]] = j  (inverse of a sort permutation)
@@ -616,16 +617,18 @@ static torch::Tensor radix_sort_recs(torch::Tensor keys,
 // second pass.  rep_stride spreads flushes over 8 replica copies of
 // `out` (zipfian hot keys).
 // =====================================================================
-// ``recB``/``posAB`` non-null (cached-position revisits): MODE 0/1
-// writes the per-sample coefficient into recB[posAB[j]].z (the second
-// pass's val column) instead of rec.z — rec then stays fully static
-// across revisits and neither record array needs a scatter rebuild.
+// ``ecol`` non-null (cached-record revisits): MODE 0/1 writes the
+// per-sample coefficient SEQUENTIALLY into ecol[j] (coalesced — no
+// scattered HBM lines) instead of rec.z, and MODE 2 reads its
+// coefficient as ecol[rec.w] (rec.w carries the pass-A position; the
+// gather hits the L2-resident e column).  Both record arrays then stay
+// fully static across revisits with no scatter rebuilds at all.
 template <int GP, int EPL, int MODE, bool SELF, bool ZERO_OUT>
 __global__ __launch_bounds__(256) void k_seg_fused(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride, int rep_mask, int4* __restrict__ recB,
+    int64_t rep_stride, int rep_mask, float* __restrict__ ecol,
     bool nt_rows) {
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
@@ -710,7 +713,7 @@ __global__ __launch_bounds__(256) void k_seg_fused(
       }
       float g;
       if (MODE == 2) {
-        g = __int_as_float(rj.z);
+        g = ecol ? ecol[rj.w] : __int_as_float(rj.z);
       } else {
         float dot = 0.f;
 #pragma unroll
@@ -722,12 +725,12 @@ __global__ __launch_bounds__(256) void k_seg_fused(
                 ? (__int_as_float(rj.z) - dot)
                 : lrc * (__int_as_float(rj.z) -
                          1.0f / (1.0f + __expf(-dot)));
-        if (recB) {
+        if (ecol) {
           // every lane holds g (the group_sum broadcast); rotate the
-          // emitting lane so the scattered stores spread across the
-          // group instead of serializing on lane 0
+          // emitting lane so consecutive records' SEQUENTIAL stores
+          // coalesce across the group
           if (lane == ((int)j & (GP - 1)))
-            recB[rj.w].z = __float_as_int(g);
+            ecol[j] = g;
         } else if (lane == 0) {
           rec[j].z = __float_as_int(g);
         }
@@ -754,7 +757,7 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
     float4* __restrict__ out, int4* __restrict__ rec,
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
-    int64_t rep_stride, int rep_mask, int4* __restrict__ recB,
+    int64_t rep_stride, int rep_mask, float* __restrict__ ecol,
     bool nt_rows) {
   extern __shared__ int4 srec[];
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
@@ -846,7 +849,7 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
       }
       float g;
       if (MODE == 2) {
-        g = __int_as_float(rj.z);
+        g = ecol ? ecol[rj.w] : __int_as_float(rj.z);
       } else {
         float dot = 0.f;
 #pragma unroll
@@ -858,9 +861,9 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
                 ? (__int_as_float(rj.z) - dot)
                 : lrc * (__int_as_float(rj.z) -
                          1.0f / (1.0f + __expf(-dot)));
-        if (recB) {
+        if (ecol) {
           if (lane == (lj & (GP - 1)))
-            recB[rj.w].z = __float_as_int(g);
+            ecol[start + lj] = g;
         } else if (lane == 0) {
           ((int*)&rec[start + lj])[2] = __float_as_int(g);
         }
@@ -1586,7 +1589,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                       const torch::Tensor& other, int64_t dim4,
                       double scale, double a, double lrc,
                       int64_t rep_stride = 0, int rep_mask = 7,
-                      int4* recB = nullptr) {
+                      float* ecol = nullptr) {
   int64_t n = rec.size(0);
   if (!n) return;
   if (dim4 == 0) {
@@ -1622,7 +1625,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
             (const float4*)keyed.data_ptr<float>(),
             (const float4*)other.data_ptr<float>(), n, dim4,
             (float)scale, (float)a, (float)lrc, chunk, rep_stride / 4,
-            rep_mask, recB, nt_rows);
+            rep_mask, ecol, nt_rows);
         return;
       }
     }
@@ -1633,7 +1636,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                        (const float4*)keyed.data_ptr<float>(),
                        (const float4*)other.data_ptr<float>(), n, dim4,
                        (float)scale, (float)a, (float)lrc, chunk,
-                       rep_stride / 4, rep_mask, recB, nt_rows);
+                       rep_stride / 4, rep_mask, ecol, nt_rows);
   });
 }
 
@@ -1805,24 +1808,22 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   bool have_recB = false;
   if (emit_ok && reuse_tag && (!hit || hit->recB.defined())) {
     if (hit) {
+      // (a rebuilt recA — fresh ratings buffer — needs no fixup: the
+      // coefficient now travels through the sequential e column and
+      // recB.w, both untouched by the rebuild)
       recB = hit->recB;
       posAB = hit->posAB;
       cntB = hit->cntB;
-      if (!recA_cached)
-        // recA was re-scattered (fresh ratings buffer): restore the
-        // emit positions its .w column carries
-        hipLaunchKernelGGL(k_pack_posw, dim3(grid_for(n, 256)),
-                           dim3(256), 0, cur_stream(),
-                           (int4*)recA.data_ptr<int>(),
-                           posAB.data_ptr<int>(), n);
     } else {
       TORCH_CHECK(histB.defined() || m > PACK_MAX_M || U > PACK_MAX_M,
                   "histB expected for LDS-sized shapes");
       recB = build_rec_swap(recA, U, histB, posB);
-      // posB is indexed by recA position (the swap pack walks recA) —
-      // pack it into recA.w so the emit needs no extra load at all
-      hipLaunchKernelGGL(k_pack_posw, dim3(grid_for(n, 256)), dim3(256),
-                         0, cur_stream(), (int4*)recA.data_ptr<int>(),
+      // each recB record carries its pass-A position in .w (posB is
+      // indexed by recA position), so pass B gathers its coefficient
+      // from the sequential e column with no extra index load
+      hipLaunchKernelGGL(k_pack_wposB, dim3(grid_for(n, 256)),
+                         dim3(256), 0, cur_stream(),
+                         (int4*)recB.data_ptr<int>(),
                          posB.data_ptr<int>(), n);
       posAB = posB;
       cntB = histB;
@@ -1838,17 +1839,22 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   // has no replica spreading)
   int64_t reps = dim4 ? delta_reps(n, m, dim) : 1;
   torch::Tensor deltas;
-  int4* eb = have_recB ? (int4*)recB.data_ptr<int>() : nullptr;
+  torch::Tensor ecol_t;
+  float* ecol = nullptr;
+  if (have_recB) {
+    ecol_t = torch::empty({n}, ratings.options());
+    ecol = ecol_t.data_ptr<float>();
+  }
   if (reps > 1) {
     auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
     seg_fused<0, true, true>(rep_buf, recA, item_vecs, user_table, dim4,
                              lr, -lr * reg, 0.0, m * dim, (int)reps - 1,
-                             eb);
+                             ecol);
     deltas = rep_buf.view({reps, m, dim}).sum(0);
   } else {
     deltas = torch::zeros_like(item_vecs);
     seg_fused<0, true, true>(deltas, recA, item_vecs, user_table, dim4,
-                             lr, -lr * reg, 0.0, 0, 7, eb);
+                             lr, -lr * reg, 0.0, 0, 7, ecol);
   }
 
   // pass B: recs by user row; apply user updates in place.  The L2
@@ -1882,7 +1888,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     auto& cc = cntB.defined() ? cntB : histB;
     auto acc = torch::zeros_like(user_table);
     seg_fused<2, false, true>(acc, recB, user_table, item_vecs, dim4,
-                              lr, 0.0, 0.0);
+                              lr, 0.0, 0.0, 0, 7, ecol);
     hipLaunchKernelGGL(k_apply_mean_update,
                        dim3(grid_for(U * dim, 256)), dim3(256), 0,
                        cur_stream(), user_table.data_ptr<float>(),
@@ -1940,7 +1946,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     }
   }
   seg_fused<2, false, false>(user_table, recB, user_table, item_vecs,
-                             dim4, lr, 0.0, 0.0);
+                             dim4, lr, 0.0, 0.0, 0, 7, ecol);
   return deltas;
 }
 
