@@ -629,7 +629,7 @@ __global__ __launch_bounds__(256) void k_seg_fused(
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
     int64_t rep_stride, int rep_mask, float* __restrict__ ecol,
-    bool nt_rows) {
+    bool write_g, bool nt_rows) {
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
   const int64_t gpb = blockDim.x / GP;
@@ -725,7 +725,10 @@ __global__ __launch_bounds__(256) void k_seg_fused(
                 ? (__int_as_float(rj.z) - dot)
                 : lrc * (__int_as_float(rj.z) -
                          1.0f / (1.0f + __expf(-dot)));
-        if (ecol) {
+        if (!write_g) {
+          // pass-B-recomputes-e configuration: g is consumed only by
+          // this walk's own accumulation — nothing to persist
+        } else if (ecol) {
           // every lane holds g (the group_sum broadcast); rotate the
           // emitting lane so consecutive records' SEQUENTIAL stores
           // coalesce across the group
@@ -758,7 +761,7 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
     const float4* __restrict__ keyed, const float4* __restrict__ other,
     int64_t n, int64_t dim4, float scale, float a, float lrc, int chunk,
     int64_t rep_stride, int rep_mask, float* __restrict__ ecol,
-    bool nt_rows) {
+    bool write_g, bool nt_rows) {
   extern __shared__ int4 srec[];
   const int64_t rep_off = (int64_t)(blockIdx.x & rep_mask) * rep_stride;
   const int lane = threadIdx.x % GP;
@@ -861,7 +864,9 @@ __global__ __launch_bounds__(256) void k_seg_fused_lds(
                 ? (__int_as_float(rj.z) - dot)
                 : lrc * (__int_as_float(rj.z) -
                          1.0f / (1.0f + __expf(-dot)));
-        if (ecol) {
+        if (!write_g) {
+          // nothing to persist (see the plain variant)
+        } else if (ecol) {
           if (lane == (lj & (GP - 1)))
             ecol[start + lj] = g;
         } else if (lane == 0) {
@@ -1589,7 +1594,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                       const torch::Tensor& other, int64_t dim4,
                       double scale, double a, double lrc,
                       int64_t rep_stride = 0, int rep_mask = 7,
-                      float* ecol = nullptr) {
+                      float* ecol = nullptr, bool write_g = true) {
   int64_t n = rec.size(0);
   if (!n) return;
   if (dim4 == 0) {
@@ -1625,7 +1630,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
             (const float4*)keyed.data_ptr<float>(),
             (const float4*)other.data_ptr<float>(), n, dim4,
             (float)scale, (float)a, (float)lrc, chunk, rep_stride / 4,
-            rep_mask, ecol, nt_rows);
+            rep_mask, ecol, write_g, nt_rows);
         return;
       }
     }
@@ -1636,7 +1641,7 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
                        (const float4*)keyed.data_ptr<float>(),
                        (const float4*)other.data_ptr<float>(), n, dim4,
                        (float)scale, (float)a, (float)lrc, chunk,
-                       rep_stride / 4, rep_mask, ecol, nt_rows);
+                       rep_stride / 4, rep_mask, ecol, write_g, nt_rows);
   });
 }
 
@@ -1839,9 +1844,25 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
   // has no replica spreading)
   int64_t reps = dim4 ? delta_reps(n, m, dim) : 1;
   torch::Tensor deltas;
+  // pass-B-recomputes-e configuration: when the user table is small
+  // enough for a side accumulator (it is for every cache-resident and
+  // HBM-regime rank-32 shape; not for the 51 GB mf128 partition), the
+  // coefficient is RE-derived inside pass B from the same operands —
+  // bitwise-identical dot (products commute, identical lane split and
+  // reduction order) — so neither walk pays any cross-order
+  // permutation traffic for e.  Race-free because pass B then flushes
+  // into the zeroed accumulator, never mutating the rows it reads.
+  // recB.z carries the ratings captured at BUILD time, so the
+  // recompute path additionally requires the current ratings buffer to
+  // be that same tensor (first visit, or identity match); a fresh
+  // ratings buffer falls back to the e-column path, whose coefficients
+  // always come from pass A's fresh recA.z.
+  const bool recompute_e =
+      have_recB && !mean_users && (!hit || same_vals) &&
+      U * dim * 4 <= (256ll << 20);
   torch::Tensor ecol_t;
   float* ecol = nullptr;
-  if (have_recB) {
+  if (have_recB && !recompute_e) {
     ecol_t = torch::empty({n}, ratings.options());
     ecol = ecol_t.data_ptr<float>();
   }
@@ -1849,12 +1870,13 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     auto rep_buf = torch::zeros({reps * m, dim}, item_vecs.options());
     seg_fused<0, true, true>(rep_buf, recA, item_vecs, user_table, dim4,
                              lr, -lr * reg, 0.0, m * dim, (int)reps - 1,
-                             ecol);
+                             ecol, /*write_g=*/!recompute_e);
     deltas = rep_buf.view({reps, m, dim}).sum(0);
   } else {
     deltas = torch::zeros_like(item_vecs);
     seg_fused<0, true, true>(deltas, recA, item_vecs, user_table, dim4,
-                             lr, -lr * reg, 0.0, 0, 7, ecol);
+                             lr, -lr * reg, 0.0, 0, 7, ecol,
+                             /*write_g=*/!recompute_e);
   }
 
   // pass B: recs by user row; apply user updates in place.  The L2
@@ -1896,7 +1918,7 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                        (float)(lr * reg));
     return deltas;
   }
-  if (reg != 0.0) {
+  if (reg != 0.0 && !recompute_e) {
     if (cntB.defined() || histB.defined()) {
       auto& cc = cntB.defined() ? cntB : histB;
       hipLaunchKernelGGL(k_scale_rows_by_count,
@@ -1944,6 +1966,27 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                            ntouch.data_ptr<int>(), dim,
                            (float)(-lr * reg));
     }
+  }
+  if (recompute_e) {
+    // order matters: the walk reads u0 rows (pre-decay, pre-update) to
+    // re-derive e, so it runs BEFORE the analytic L2 decay applied
+    // above would be wrong — hence the decay was applied above only
+    // for the non-recompute path; here: accumulate first, then decay,
+    // then add.
+    auto acc = torch::zeros_like(user_table);
+    seg_fused<0, false, true>(acc, recB, user_table, item_vecs, dim4,
+                              lr, 0.0, 0.0, 0, 7, nullptr,
+                              /*write_g=*/false);
+    if (reg != 0.0) {
+      auto& cc = cntB.defined() ? cntB : histB;
+      TORCH_CHECK(cc.defined(), "recompute path needs per-user counts");
+      hipLaunchKernelGGL(k_scale_rows_by_count,
+                         dim3(grid_for(U * dim, 256)), dim3(256), 0,
+                         cur_stream(), user_table.data_ptr<float>(),
+                         cc.data_ptr<int>(), U, dim, (float)(-lr * reg));
+    }
+    user_table.add_(acc);
+    return deltas;
   }
   seg_fused<2, false, false>(user_table, recB, user_table, item_vecs,
                              dim4, lr, 0.0, 0.0, 0, 7, ecol);
