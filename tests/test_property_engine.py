@@ -16,7 +16,9 @@ from flink_parameter_server_amd.parallel.transport import run_loopback
 
 
 @pytest.mark.timeout(600)
-@settings(max_examples=25, deadline=None)
+# derandomized: a fixed example set keeps the driver's CI runs
+# deterministic; run deeper random soaks manually by flipping this off
+@settings(max_examples=25, deadline=None, derandomize=True)
 @given(
     world=st.integers(1, 4),
     feedback=st.sampled_from(["fresh", "merged", "pipelined"]),
