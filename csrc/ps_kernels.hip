@@ -1693,6 +1693,65 @@ static PosCacheKey pos_cache_key(int64_t tag, const torch::Tensor& a,
   return {tag, a.data_ptr(), b.data_ptr(), n};
 }
 
+// analytic L2 decay of the touched user rows: per-row counts when the
+// histogram exists (LDS-sized tables), else the (cached) touched-rows
+// extents of the user-sorted records
+static void decay_users(torch::Tensor& user_table, torch::Tensor& recB,
+                        torch::Tensor& cntB, torch::Tensor& histB,
+                        int64_t n, int64_t U, int64_t dim, double lr,
+                        double reg, int64_t reuse_tag,
+                        const PosCacheKey& ck) {
+  auto opts_i32 = torch::dtype(torch::kInt32)
+                      .device(user_table.device());
+  if (cntB.defined() || histB.defined()) {
+    auto& cc = cntB.defined() ? cntB : histB;
+    hipLaunchKernelGGL(k_scale_rows_by_count,
+                       dim3(grid_for(U * dim, 256)), dim3(256), 0,
+                       cur_stream(), user_table.data_ptr<float>(),
+                       cc.data_ptr<int>(), U, dim, (float)(-lr * reg));
+    return;
+  }
+  // huge sparse-touched table: decay only the rows this batch
+  // touches, from the sorted recs' segment extents.  The extents
+  // depend only on recB's STATIC key column, so revisits reuse
+  // them from the position cache (k_seg_extents was 60% of the
+  // HBM-regime mf32 round — profiles/PERF_NOTES.md round 2)
+  torch::Tensor trows, tcnts, ntouch;
+  PosCacheEntry* e = nullptr;
+  if (reuse_tag) {
+    auto it = g_pos_cache.find(ck);
+    if (it != g_pos_cache.end()) e = &it->second;
+  }
+  if (e && e->trows.defined()) {
+    trows = e->trows; tcnts = e->tcnts; ntouch = e->ntouch;
+  } else {
+    trows = torch::empty({n}, opts_i32);
+    tcnts = torch::empty({n}, opts_i32);
+    ntouch = torch::zeros({1}, opts_i32);
+    hipLaunchKernelGGL(k_seg_extents, dim3(grid_for(n, 256)),
+                       dim3(256), 0, cur_stream(),
+                       (const int4*)recB.data_ptr<int>(), n,
+                       trows.data_ptr<int>(), tcnts.data_ptr<int>(),
+                       ntouch.data_ptr<int>());
+    if (e) { e->trows = trows; e->tcnts = tcnts; e->ntouch = ntouch; }
+  }
+  if (dim % 4 == 0)
+    hipLaunchKernelGGL(k_scale_touched_rows_vec4,
+                       dim3(grid_for(n * 32, 256)), dim3(256), 0,
+                       cur_stream(),
+                       (float4*)user_table.data_ptr<float>(),
+                       trows.data_ptr<int>(), tcnts.data_ptr<int>(),
+                       ntouch.data_ptr<int>(), dim / 4,
+                       (float)(-lr * reg));
+  else
+    hipLaunchKernelGGL(k_scale_touched_rows,
+                       dim3(grid_for(n * dim, 256)), dim3(256), 0,
+                       cur_stream(), user_table.data_ptr<float>(),
+                       trows.data_ptr<int>(), tcnts.data_ptr<int>(),
+                       ntouch.data_ptr<int>(), dim,
+                       (float)(-lr * reg));
+}
+
 torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                           torch::Tensor item_vecs, torch::Tensor item_slots,
                           torch::Tensor ratings, double lr, double reg,
@@ -1918,55 +1977,9 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
                        (float)(lr * reg));
     return deltas;
   }
-  if (reg != 0.0 && !recompute_e) {
-    if (cntB.defined() || histB.defined()) {
-      auto& cc = cntB.defined() ? cntB : histB;
-      hipLaunchKernelGGL(k_scale_rows_by_count,
-                         dim3(grid_for(U * dim, 256)), dim3(256), 0,
-                         cur_stream(), user_table.data_ptr<float>(),
-                         cc.data_ptr<int>(), U, dim, (float)(-lr * reg));
-    } else {
-      // huge sparse-touched table: decay only the rows this batch
-      // touches, from the sorted recs' segment extents.  The extents
-      // depend only on recB's STATIC key column, so revisits reuse
-      // them from the position cache (k_seg_extents was 60% of the
-      // HBM-regime mf32 round — profiles/PERF_NOTES.md round 2)
-      torch::Tensor trows, tcnts, ntouch;
-      PosCacheEntry* e = nullptr;
-      if (reuse_tag) {
-        auto it = g_pos_cache.find(ck);
-        if (it != g_pos_cache.end()) e = &it->second;
-      }
-      if (e && e->trows.defined()) {
-        trows = e->trows; tcnts = e->tcnts; ntouch = e->ntouch;
-      } else {
-        trows = torch::empty({n}, opts_i32);
-        tcnts = torch::empty({n}, opts_i32);
-        ntouch = torch::zeros({1}, opts_i32);
-        hipLaunchKernelGGL(k_seg_extents, dim3(grid_for(n, 256)),
-                           dim3(256), 0, cur_stream(),
-                           (const int4*)recB.data_ptr<int>(), n,
-                           trows.data_ptr<int>(), tcnts.data_ptr<int>(),
-                           ntouch.data_ptr<int>());
-        if (e) { e->trows = trows; e->tcnts = tcnts; e->ntouch = ntouch; }
-      }
-      if (dim % 4 == 0)
-        hipLaunchKernelGGL(k_scale_touched_rows_vec4,
-                           dim3(grid_for(n * 32, 256)), dim3(256), 0,
-                           cur_stream(),
-                           (float4*)user_table.data_ptr<float>(),
-                           trows.data_ptr<int>(), tcnts.data_ptr<int>(),
-                           ntouch.data_ptr<int>(), dim / 4,
-                           (float)(-lr * reg));
-      else
-        hipLaunchKernelGGL(k_scale_touched_rows,
-                           dim3(grid_for(n * dim, 256)), dim3(256), 0,
-                           cur_stream(), user_table.data_ptr<float>(),
-                           trows.data_ptr<int>(), tcnts.data_ptr<int>(),
-                           ntouch.data_ptr<int>(), dim,
-                           (float)(-lr * reg));
-    }
-  }
+  if (reg != 0.0 && !recompute_e)
+    decay_users(user_table, recB, cntB, histB, n, U, dim, lr, reg,
+                reuse_tag, ck);
   if (recompute_e) {
     // order matters: the walk reads u0 rows (pre-decay, pre-update) to
     // re-derive e, so it runs BEFORE the analytic L2 decay applied
@@ -1977,14 +1990,9 @@ torch::Tensor mf_sgd_step(torch::Tensor user_table, torch::Tensor user_rows,
     seg_fused<0, false, true>(acc, recB, user_table, item_vecs, dim4,
                               lr, 0.0, 0.0, 0, 7, nullptr,
                               /*write_g=*/false);
-    if (reg != 0.0) {
-      auto& cc = cntB.defined() ? cntB : histB;
-      TORCH_CHECK(cc.defined(), "recompute path needs per-user counts");
-      hipLaunchKernelGGL(k_scale_rows_by_count,
-                         dim3(grid_for(U * dim, 256)), dim3(256), 0,
-                         cur_stream(), user_table.data_ptr<float>(),
-                         cc.data_ptr<int>(), U, dim, (float)(-lr * reg));
-    }
+    if (reg != 0.0)
+      decay_users(user_table, recB, cntB, histB, n, U, dim, lr, reg,
+                  reuse_tag, ck);
     user_table.add_(acc);
     return deltas;
   }
