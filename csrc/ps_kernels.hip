@@ -1387,7 +1387,7 @@ static int pack_chunk() {
 }
 
 static int seg_chunk() {
-  static int c = env_int("FPS_SEG_CHUNK", 256);
+  static int c = env_int("FPS_SEG_CHUNK", 128);
   return c;
 }
 
