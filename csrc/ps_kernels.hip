@@ -1620,8 +1620,11 @@ static void seg_fused(torch::Tensor out, torch::Tensor rec,
     constexpr int GP = decltype(tag)::value;
     constexpr int EPL = decltype(etag)::value;
     if (use_lds) {
+      // default cap 64 KB keeps 2 blocks/CU; FPS_LDS_CAP=131072 trades
+      // occupancy (1 block/CU at 128 KB) for staging the 4-lane shapes
+      static size_t lds_cap = (size_t)env_int("FPS_LDS_CAP", 64 * 1024);
       size_t lds = (size_t)(256 / GP) * chunk * sizeof(int4);
-      if (lds <= 64 * 1024) {
+      if (lds <= lds_cap && lds <= 160 * 1024) {
         hipLaunchKernelGGL(
             (k_seg_fused_lds<GP, EPL, MODE, SELF, ZERO_OUT>),
             dim3(grid_for(nchunks * GP, 256)), dim3(256), lds,
