@@ -20,7 +20,8 @@ from .parallel.transport import (LocalTransport, LoopbackTransport,
                                  TorchDistTransport, Transport)
 from .parallel.partitioner import (ModPartitioner, RangePartitioner,
                                    make_partitioner)
-from .transform import (transform, transform_local, transform_simple,
+from .transform import (transform, transform_distributed,
+                        transform_local, transform_simple,
                         transform_with_model_load)
 from .messaging import (CountFlush, FlushAlways, FlushStrategy, Pull,
                         PullAnswer, Push, PSToWorker, SerializingTransport,

@@ -108,3 +108,44 @@ def transform_local(world: int, make_data: Callable[[int], Iterable[Any]],
         return eng.run(make_data(rank))
 
     return run_loopback(world, job)
+
+
+def transform_distributed(data: Iterable[Any], worker_logic: WorkerLogic,
+                          server_logic: ServerLogic,
+                          config: Optional[PSConfig] = None,
+                          part: Optional[Partitioner] = None,
+                          device: Optional[torch.device] = None,
+                          max_rounds: Optional[int] = None,
+                          flush: Optional[Any] = None,
+                          checkpointer: Optional[Any] = None
+                          ) -> Tuple[List[Any], List[Any]]:
+    """One-process-per-GPU ``transform``: call once PER RANK under
+    ``torch.distributed.run`` (or any launcher that initialized
+    ``torch.distributed``); wires :class:`TorchDistTransport` (nccl =
+    RCCL over xGMI on GPUs, gloo on CPU) so the same logics that ran
+    under ``transform``/``transform_local`` scale across ranks.
+
+    ``data`` is THIS rank's partition of the input (the reference's
+    rebalance from the source is the launcher's/data loader's job);
+    returns this rank's (worker_outputs, server_outputs) — shard dumps
+    stay per-rank like the reference's per-subtask PSOut streams.
+    """
+    import torch.distributed as dist
+
+    from .parallel.transport import TorchDistTransport
+
+    if not dist.is_initialized():
+        raise RuntimeError(
+            "transform_distributed needs torch.distributed initialized "
+            "(launch with torch.distributed.run and call "
+            "dist.init_process_group first)")
+    transport = TorchDistTransport(device=device)
+    cfg = config or PSConfig()
+    if cfg.feedback == "fresh" and transport.world > 1:
+        # multi-rank default: the overlapped schedule (see PSConfig)
+        cfg.feedback = "pipelined"
+    eng = PSEngine(cfg, worker_logic, server_logic, transport, part,
+                   device=device if device is not None
+                   else transport.device,
+                   flush=flush, checkpointer=checkpointer)
+    return eng.run(data, max_rounds=max_rounds)

@@ -169,3 +169,58 @@ def test_gloo_world8_soak(feedback):
         got = torch.stack([merged[i] for i in range(97)])
         assert torch.allclose(got, expected, atol=1e-4), \
             (got - expected).abs().max()
+
+
+def _transform_dist_job(rank, world, port, outdir):
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world)
+    try:
+        from flink_parameter_server_amd import (PSConfig, ShardedTable,
+                                                TableServerLogic,
+                                                WorkerLogic)
+        from flink_parameter_server_amd.parallel.partitioner import \
+            ModPartitioner
+        from flink_parameter_server_amd.transform import \
+            transform_distributed
+
+        n_ids = 17
+
+        class Push(WorkerLogic):
+            def on_recv(self, data, client):
+                ids = torch.as_tensor(data, dtype=torch.int64)
+                client.push(ids, torch.ones(ids.numel(), 1))
+
+            def on_pull_recv(self, ids, values, client):
+                pass
+
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, 1, "cpu")
+        _, souts = transform_distributed(
+            [list(range(n_ids))] * (rank + 1), Push(),
+            TableServerLogic(table),
+            PSConfig(num_ids=n_ids, dim=1), part)
+        gids, vals = souts[0]
+        torch.save(dict(zip(gids.tolist(), vals.reshape(-1).tolist())),
+                   os.path.join(outdir, f"rank{rank}.pt"))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_transform_distributed_gloo_world2():
+    """The one-call-per-rank job entry wires the dist transport and the
+    pipelined default; shard dumps carry the right totals."""
+    world = 2
+    from conftest import free_port
+    port = free_port()
+    with tempfile.TemporaryDirectory() as d:
+        mp.spawn(_transform_dist_job, args=(world, port, d),
+                 nprocs=world, join=True)
+        merged = {}
+        for r in range(world):
+            merged.update(torch.load(os.path.join(d, f"rank{r}.pt"),
+                                     weights_only=False))
+        assert len(merged) == 17
+        assert all(v == 3.0 for v in merged.values())  # 1+2 pushes
