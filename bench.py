@@ -178,8 +178,11 @@ def main():
     ap.add_argument("--lr-features", type=int, default=10**9,
                     help="feature count for --model lr (the configs[3] "
                          "shape; 8 GB of weights+accum at 1 GPU)")
-    ap.add_argument("--n-batches", type=int, default=8,
-                    help="distinct pre-generated batches cycled")
+    ap.add_argument("--n-batches", type=int, default=4,
+                    help="distinct pre-generated batches cycled (a "
+                         "warmup of >= this count amortizes the one-"
+                         "time per-batch record/coalescing builds "
+                         "before the timed region)")
     ap.add_argument("--device", default=None)
     ap.add_argument("--feedback", choices=["fresh", "merged", "pipelined"],
                     default=None,
