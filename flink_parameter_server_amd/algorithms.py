@@ -256,7 +256,8 @@ def _transform_pa(data, feature_count, model, worker_parallelism,
 
     def make_worker(rank: int) -> PAWorkerLogic:
         return PAWorkerLogic(variant=pa_method, C=C, n_classes=n_classes,
-                             pull_limit=pull_limit)
+                             pull_limit=pull_limit,
+                             n_features=feature_count)
 
     def make_server(rank: int) -> TableServerLogic:
         table = ShardedTable(part, rank, dim, torch.device("cpu"))

@@ -24,7 +24,7 @@ import torch
 
 from .. import ops
 from ..api import ParameterServerClient, WorkerLogic
-from .common import PullBuffer
+from .common import DenseCoalescer, HashCoalescer, PullBuffer
 
 
 class SparseBatch:
@@ -57,7 +57,9 @@ class PAWorkerLogic(WorkerLogic):
     def __init__(self, variant: str = "PA-I", C: float = 1.0,
                  n_classes: int = 0,
                  pull_limit: Optional[int] = None,
-                 emit_predictions: bool = True):
+                 emit_predictions: bool = True,
+                 n_features: int = 0,
+                 device: torch.device = torch.device("cpu")):
         if variant not in ("PA", "PA-I", "PA-II"):
             raise ValueError(f"unknown PA variant {variant!r}")
         self.variant = variant
@@ -67,16 +69,30 @@ class PAWorkerLogic(WorkerLogic):
         self.pull_limit = pull_limit
         self.emit_predictions = emit_predictions
         self._pending: List[Tuple[SparseBatch, PullBuffer]] = []
+        # feature-space-aware coalescing like the LR worker: dense-flag
+        # scatter for bounded spaces, sort-free hash for unbounded ones
+        # on device; torch.unique fallback otherwise
+        self._coalescer = None
+        if 0 < n_features <= DenseCoalescer.MAX_SPACE:
+            self._coalescer = DenseCoalescer(n_features, device)
 
     def batch_size(self, batch: SparseBatch) -> int:
         return len(batch)
 
     def on_recv(self, data: SparseBatch,
                 client: ParameterServerClient) -> None:
-        uids = torch.unique(data.feat_ids)
-        buf = PullBuffer(uids, self.dim)
+        if self._coalescer is not None:
+            uids, _ = self._coalescer(data.feat_ids)
+            buf = PullBuffer(uids, self.dim, assume_sorted=True)
+        elif data.feat_ids.is_cuda:
+            hc = HashCoalescer()
+            uids, _ = hc(data.feat_ids.long())
+            buf = PullBuffer(uids, self.dim, resolver=hc.resolver)
+        else:
+            uids = torch.unique(data.feat_ids)
+            buf = PullBuffer(uids, self.dim, assume_sorted=True)
         self._pending.append((data, buf))
-        client.pull(uids)
+        client.pull(uids, unique=True)
 
     def on_pull_recv(self, ids: torch.Tensor, values: torch.Tensor,
                      client: ParameterServerClient) -> None:
