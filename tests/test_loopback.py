@@ -413,3 +413,71 @@ def test_kitchen_sink_pipelined():
     got = torch.stack([merged[i] for i in range(n_ids)])
     assert torch.allclose(got, expected, atol=0.15, rtol=0.05), \
         (got - expected).abs().max()
+
+
+def test_pipelined_out_of_order_server_and_checkpoint(tmp_path):
+    """pipelined + a server logic WITHOUT the answers_in_order flag
+    (the extra answer-ids channel must ride the merged exchange) + a
+    periodic checkpointer: answers exactly-once, totals right, and the
+    newest complete dump is loadable."""
+    from flink_parameter_server_amd import ServerLogic
+    from flink_parameter_server_amd.utils import checkpoint as ckpt
+
+    world, n_ids, dim = 2, 19, 2
+
+    class ReversingServer(ServerLogic):
+        """Answers each pull batch in REVERSED order (ids must travel)."""
+
+        def __init__(self, table):
+            self.table = table
+
+        def on_pull_recv(self, ids, worker, ps):
+            rev = torch.flip(ids, [0])
+            ps.answer_pull(rev, self.table.gather(rev), worker)
+
+        def on_push_recv(self, ids, deltas, ps):
+            self.table.apply(ids, deltas)
+
+    class Traffic(WorkerLogic):
+        def __init__(self, rank):
+            self.g = torch.Generator().manual_seed(40 + rank)
+            self.pushed = torch.zeros(n_ids, dim)
+            self.answered = 0
+
+        def on_recv(self, data, client):
+            ids = torch.unique(torch.randint(0, n_ids, (6,),
+                                             generator=self.g))
+            deltas = torch.randn(ids.numel(), dim, generator=self.g)
+            client.push(ids, deltas)
+            self.pushed[ids] += deltas
+            client.pull(ids, unique=True)
+
+        def on_pull_recv(self, ids, values, client):
+            self.answered += int(ids.numel())
+
+    def job(rank, transport):
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, dim, "cpu")
+        logic = Traffic(rank)
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=dim,
+                                feedback="pipelined"), logic,
+                       ReversingServer(table), transport, part,
+                       checkpointer=ckpt.PeriodicCheckpointer(
+                           str(tmp_path / "ck"), table, every_rounds=4))
+        eng.run([r for r in range(12)])
+        assert eng.metrics.pull_ids_sent == logic.answered
+        gids, vals = table.dump()
+        return logic.pushed, dict(
+            zip(gids.tolist(), [v.clone() for v in vals]))
+
+    results = run_loopback(world, job)
+    expected = sum(r[0] for r in results)
+    merged = {}
+    for _, d in results:
+        merged.update(d)
+    got = torch.stack([merged[i] for i in range(n_ids)])
+    assert torch.allclose(got, expected, atol=1e-5)
+    latest = ckpt.latest_checkpoint(str(tmp_path / "ck"))
+    assert latest is not None
+    t2 = ShardedTable(ModPartitioner(n_ids, 3), 1, dim, "cpu")
+    ckpt.load_shard(latest, t2)  # re-shard 2 -> 3 loads cleanly
