@@ -55,21 +55,25 @@ class PullBuffer:
         return torch.searchsorted(self.uids, ids)
 
     def offer_exact(self, ids: torch.Tensor, values: torch.Tensor) -> None:
-        """Fast path: the caller guarantees ``set(ids) == set(self.uids)``
-        (e.g. the engine answered this buffer's whole pull in one round and
-        no other buffer was pending).  No host-device syncs."""
+        """Fast path for a whole-pull answer.  ONLY the pointer-identity
+        case may skip verification (in-order answers return the very
+        tensor we pulled — dense pull / single shard); any other
+        delivery goes through the membership-checked :meth:`offer`.
+        (A count-match heuristic here once let a WRONG-BATCH answer
+        with near-total id overlap fill the buffer unverified — with
+        merged/pipelined schedules keeping two overlapping pulls in
+        flight, the misalignment cascaded batch-by-batch into NaN
+        training.  Never fill rows from unverified ids.)"""
         if (ids.data_ptr() == self.uids.data_ptr()
                 and ids.numel() == self.uids.numel()
                 and values.dtype == self.values.dtype):
-            # in-order answers return the very tensor we pulled (dense
-            # pull / single-shard): adopt the payload, skip the permute
+            # adopt the payload, skip the permute
             self.values = values
-        else:
-            rows = self.slots_of(ids).long()
-            self.values[rows] = values.to(self.values.dtype)
-        if self._have_t is not None:
-            self._have_t.fill_(True)
-        self._n_have = int(self.uids.numel())
+            if self._have_t is not None:
+                self._have_t.fill_(True)
+            self._n_have = int(self.uids.numel())
+            return
+        self.offer(ids, values)
 
     def offer(self, ids: torch.Tensor, values: torch.Tensor) -> None:
         """Fill rows answered by (ids, values); ignores foreign ids and

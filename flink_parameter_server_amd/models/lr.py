@@ -99,8 +99,9 @@ class LRWorkerLogic(WorkerLogic):
                 ids.numel() == self._pending[0][1].uids.numel()):
             batch, buf, slots = self._pending[0]
             buf.offer_exact(ids, values)
-            self._pending = []
-            self._train(batch, buf, client, slots)
+            if buf.complete:  # see mf.py: same-size wrong-pull guard
+                self._pending = []
+                self._train(batch, buf, client, slots)
             return
         still = []
         for batch, buf, slots in self._pending:

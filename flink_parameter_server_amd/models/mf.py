@@ -206,12 +206,17 @@ class OnlineMFWorkerLogic(WorkerLogic):
         if (len(self._pending) == 1 and
                 self._pending[0][1]._n_have == 0 and
                 ids.numel() == self._pending[0][1].uids.numel()):
-            # the engine answered this buffer's whole pull at once and no
-            # other buffer could have contributed ids: sync-free path
+            # likely a whole-pull answer: offer_exact adopts it sync-
+            # free on pointer identity, else verifies membership.  Only
+            # a COMPLETE buffer trains — a same-sized answer for a
+            # different overlapping pull (possible when merged/
+            # pipelined schedules keep two pulls in flight) fills just
+            # the overlap and keeps waiting for its own answer.
             batch, buf, slots = self._pending[0]
             buf.offer_exact(ids, values)
-            self._pending = []
-            self._train(batch, buf, client, slots)
+            if buf.complete:
+                self._pending = []
+                self._train(batch, buf, client, slots)
             return
         still: List[Tuple[MFBatch, PullBuffer, Optional[torch.Tensor]]] = []
         for batch, buf, slots in self._pending:

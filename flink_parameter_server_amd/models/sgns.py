@@ -131,8 +131,10 @@ class SGNSWorkerLogic(WorkerLogic):
                 ids.numel() == self._pending[0][3].uids.numel()):
             centers, ctx_ids, labels, buf, cs, os_ = self._pending[0]
             buf.offer_exact(ids, values)
-            self._pending = []
-            self._train(centers, ctx_ids, labels, buf, client, cs, os_)
+            if buf.complete:  # see mf.py: same-size wrong-pull guard
+                self._pending = []
+                self._train(centers, ctx_ids, labels, buf, client,
+                            cs, os_)
             return
         still = []
         for centers, ctx_ids, labels, buf, cs, os_ in self._pending:

@@ -295,7 +295,18 @@ class LoopbackTransport(Transport):
 
     def alltoallv(self, parts, specs, extra=None, recv_counts=None):
         sh = self.shared
-        sh.mailbox[self.rank] = [list(chan) for chan in parts]
+        # Cross-rank segments are COPIED at send time, exactly like the
+        # real backends (nccl/gloo isend serialize the buffer; only the
+        # self-addressed short-circuit may alias).  Passing references
+        # across threads let a peer's later in-place mutation (e.g. the
+        # full-range gather alias mutated by its shard's next push
+        # apply) race with the receiver's single-barrier-per-round
+        # merged/pipelined consumption — a loopback-only hazard that
+        # does not exist in the process-per-rank deployment.
+        self_rank = self.rank
+        sh.mailbox[self_rank] = [
+            [t if (d == self_rank or t is None) else t.clone()
+             for d, t in enumerate(chan)] for chan in parts]
         if extra is not None:
             sh.intbox[self.rank] = torch.tensor(list(extra),
                                                 dtype=torch.int64)

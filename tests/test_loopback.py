@@ -328,9 +328,15 @@ def test_debug_validate_catches_inplace_mutation():
 
 
 def test_pipelined_overlapped_delivery_mf():
-    """Online MF converges under the pipelined (split-phase, +1 round
+    """Online MF converges under the pipelined (split-phase, +2 rounds
     answer latency) schedule too — the overlap must not change
-    semantics, only timing."""
+    semantics, only timing.  (A flaky NaN here exposed a loopback-only
+    race: cross-rank answer segments were passed by REFERENCE between
+    the rank threads, so a shard's full-range gather alias could be
+    mutated by its next push apply while the peer was still training on
+    it under the single-barrier merged/pipelined schedules.  The
+    loopback transport now clones cross-rank segments at send time —
+    exactly what isend does in the real backends.)"""
     world, n_users, n_items, rank_k = 2, 40, 30, 4
     batches, _, _ = synth.planted_mf_ratings(n_users, n_items, rank_k,
                                              2400, 120, seed=33)
@@ -344,7 +350,7 @@ def test_pipelined_overlapped_delivery_mf():
         eng = PSEngine(PSConfig(num_ids=n_items, dim=rank_k,
                                 feedback="pipelined"), wl,
                        TableServerLogic(table), transport, part)
-        eng.run(epoch_iterator(per_worker[rank], 8))
+        eng.run(epoch_iterator(per_worker[rank], 10))
         return wl.user_table, table
 
     results = run_loopback(world, job)
