@@ -105,3 +105,66 @@ def test_engine_always_terminates_and_accounts(
     tol = 0.2 if wire == "bf16" and world > 1 else 1e-4
     assert torch.allclose(got, expected, atol=tol, rtol=0.05), \
         (got - expected).abs().max()
+
+
+@pytest.mark.timeout(600)
+@settings(max_examples=20, deadline=None, derandomize=True)
+@given(
+    world=st.integers(1, 3),
+    feedback=st.sampled_from(["fresh", "merged", "pipelined"]),
+    pull_limit=st.one_of(st.none(), st.integers(4, 40)),
+    n_items=st.integers(6, 60),
+    n_users=st.integers(5, 40),
+    n_batches=st.integers(2, 10),
+    batch_sz=st.integers(8, 200),
+    seed=st.integers(0, 10**6),
+)
+def test_mf_pullbuffer_reassembly_stays_finite(
+        world, feedback, pull_limit, n_items, n_users, n_batches,
+        batch_sz, seed):
+    """The MODEL-layer net for the round-2 answer-misrouting bug class:
+    the real MF worker (PullBuffer reassembly, overlapping pulls across
+    in-flight rounds, answers split under pull limits) must train every
+    batch on COMPLETE, correctly-addressed values — tables stay finite
+    and bounded under every schedule/limit/shape combination.  (The
+    engine-level property test above drives raw ids and never touches
+    PullBuffer, which is exactly where the corruption lived.)"""
+    from flink_parameter_server_amd.models.mf import (MFBatch,
+                                                      OnlineMFWorkerLogic)
+
+    dim = 4
+    g = torch.Generator().manual_seed(seed)
+    per_rank_batches = []
+    for r in range(world):
+        bs = []
+        for _ in range(n_batches):
+            users = torch.randint(0, n_users, (batch_sz,), generator=g)
+            items = torch.randint(0, n_items, (batch_sz,), generator=g)
+            ratings = torch.randn(batch_sz, generator=g) * 0.3
+            bs.append(MFBatch(users, items, ratings))
+        per_rank_batches.append(bs)
+
+    def job(rank, transport):
+        part = ModPartitioner(n_items, world)
+        table = ShardedTable(part, rank, dim, "cpu", init=(-0.1, 0.1))
+        wl = OnlineMFWorkerLogic(n_users, dim, learning_rate=0.01,
+                                 range_min=-0.1, range_max=0.1,
+                                 pull_limit=pull_limit)
+        eng = PSEngine(PSConfig(num_ids=n_items, dim=dim,
+                                feedback=feedback), wl,
+                       TableServerLogic(table), transport, part)
+        it = iter(per_rank_batches[rank] * 2)
+        exhausted = done = False
+        rounds = 0
+        while not done and rounds < 500:
+            rounds += 1
+            exhausted, done = eng._round(it, exhausted)
+        assert done, f"no termination: rank={rank} rounds={rounds}"
+        assert not wl._pending, "batches left untrained at drain"
+        assert torch.isfinite(wl.user_table).all()
+        assert float(wl.user_table.abs().max()) < 50.0
+        assert torch.isfinite(table.data).all()
+        assert float(table.data.abs().max()) < 50.0
+        return True
+
+    assert all(run_loopback(world, job, timeout=120))
