@@ -571,3 +571,29 @@ def test_lr_static_batches_sorted_walk_matches_cpu():
     assert len(wout_gpu) == len(wout_cpu) > 0
     for (gi, gp), (ci, cp) in zip(wout_gpu, wout_cpu):
         assert torch.allclose(gp.cpu(), cp, atol=1e-4)
+
+
+def test_mf_sgd_step_mean_users_position_cache_matches():
+    """mean aggregation + the record cache (reuse_tag revisits use the
+    e-column delivery, not the pass-B recompute) stays oracle-exact."""
+    torch.manual_seed(12)
+    n, U, M, dim = 4000, 250, 150, 32
+    users = torch.randint(0, U, (n,), dtype=torch.int32)
+    slots = torch.randint(0, M, (n,), dtype=torch.int32)
+    ut_gpu = torch.randn(U, dim).to(DEV)
+    users_g, slots_g = users.to(DEV), slots.to(DEV)
+    r = torch.randn(n)
+    r_g = r.to(DEV)
+    for visit in range(3):
+        iv = torch.randn(M, dim)
+        ut_cpu = ut_gpu.cpu().clone()
+        d_cpu = R.mf_sgd_step(ut_cpu, users.long(), iv, slots.long(), r,
+                              0.05, 0.01, mean_users=True)
+        d_gpu = ops.mf_sgd_step(ut_gpu, users_g, iv.to(DEV), slots_g,
+                                r_g, 0.05, 0.01, reuse_tag=424242,
+                                mean_users=True)
+        torch.cuda.synchronize()
+        assert torch.allclose(ut_cpu, ut_gpu.cpu(), atol=1e-3,
+                              rtol=1e-3), visit
+        assert torch.allclose(d_cpu, d_gpu.cpu(), atol=1e-3,
+                              rtol=1e-3), visit
