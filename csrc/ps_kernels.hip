@@ -1656,10 +1656,14 @@ static int64_t delta_reps(int64_t n, int64_t m, int64_t dim) {
   // (the walk's flush atomics were the contention); shapes whose
   // replica buffer would exceed 64 MB fall back to 1 as before
   static int64_t r_env = env_int("FPS_DELTA_REPS", 64);
+  static int64_t cap_mb = env_int("FPS_REPS_CAP_MB", 64);
   int64_t r = r_env;
   if (r < 1) r = 1;
-  if (n < r * m) return 1;
-  if (r * m * dim * 4 > (64 << 20)) return 1;  // cap replica buffer 64 MB
+  // halve toward the largest replica count the batch density and the
+  // buffer cap allow (the old all-or-nothing check silently dropped
+  // mid-size shapes to 1)
+  while (r > 1 && (n < r * m || r * m * dim * 4 > (cap_mb << 20)))
+    r >>= 1;
   return r;
 }
 
