@@ -156,7 +156,13 @@ def test_mf_pullbuffer_reassembly_stays_finite(
         it = iter(per_rank_batches[rank] * 2)
         exhausted = done = False
         rounds = 0
-        while not done and rounds < 500:
+        # livelock backstop sized from the workload: a tiny pull limit
+        # legitimately splits each batch's pull into ceil(ids/limit)
+        # chunks at ~3 rounds per chunk under the pipelined schedule
+        # (budget is re-checked before the overlapped delivery frees it)
+        chunks = -(-min(n_items, batch_sz) // (pull_limit or n_items))
+        bound = 100 + 2 * n_batches * (chunks + 4) * 3
+        while not done and rounds < bound:
             rounds += 1
             exhausted, done = eng._round(it, exhausted)
         assert done, f"no termination: rank={rank} rounds={rounds}"
