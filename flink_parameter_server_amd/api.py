@@ -76,7 +76,11 @@ class WorkerLogic(ABC):
     #: flight.  ``None`` = unlimited.  Reference:
     #: ``WorkerLogic.addPullLimiter`` — there it exists to stop the Flink
     #: iteration from deadlocking under backpressure; here it bounds the
-    #: pull/answer ring buffers (memory backpressure).
+    #: pull/answer ring buffers (memory backpressure).  Size it for the
+    #: schedule's pipeline depth: a pull split into ceil(ids/limit)
+    #: chunks progresses at ~1 chunk per answer round-trip (2-3 rounds
+    #: under merged/pipelined), so a tiny limit throttles throughput
+    #: long before memory matters (the reference's default was 1600).
     pull_limit: Optional[int] = None
 
     @abstractmethod
