@@ -139,7 +139,7 @@ class SerializingTransport(Transport):
         self.rank = inner.rank
         self.world = inner.world
 
-    def alltoallv(self, parts, specs, extra=None, recv_counts=None):
+    def _codec(self, parts):
         C = len(parts)
         wire = [list(chan) for chan in parts]
         if C >= 3:  # worker -> server channels (always the first 3)
@@ -156,8 +156,24 @@ class SerializingTransport(Transport):
         # C == 1 / C == 4: in-order feedback — answer ids intentionally
         # elided from the wire (PullAnswer.param_ids reconstructed by
         # the puller from its own sent segments); values pass as-is.
-        return self.inner.alltoallv(wire, specs, extra=extra,
-                                    recv_counts=recv_counts)
+        return wire
+
+    def alltoallv(self, parts, specs, extra=None, recv_counts=None):
+        return self.inner.alltoallv(self._codec(parts), specs,
+                                    extra=extra, recv_counts=recv_counts)
+
+    # split-phase form (the pipelined schedule): the codec applies at
+    # issue time, completion passes straight through
+    def alltoallv_begin(self, parts, specs, extra=None, recv_counts=None):
+        return self.inner.alltoallv_begin(self._codec(parts), specs,
+                                          extra=extra,
+                                          recv_counts=recv_counts)
+
+    def alltoallv_end(self, handle):
+        return self.inner.alltoallv_end(handle)
+
+    def ctrl_of(self, handle):
+        return self.inner.ctrl_of(handle)
 
     def allgather_int(self, vec):
         return self.inner.allgather_int(vec)

@@ -281,3 +281,22 @@ def test_merged_feedback_out_of_order_server():
     for _, souts in results:
         counts.update({i: v for i, v in souts})
     assert counts == {0: 2.0, 1: 4.0, 2: 4.0, 3: 2.0}
+
+
+def test_pipelined_feedback_with_codec():
+    """pipelined (split-phase) exchanges round-trip through the entity
+    codec too — SerializingTransport implements begin/end."""
+
+    def job(rank, transport):
+        data = [torch.tensor([rank, 2, 5]), torch.tensor([2])]
+        wl = CounterLogic()
+        eng = PSEngine(PSConfig(num_ids=8, dim=1, feedback="pipelined"),
+                       wl, _server(), SerializingTransport(transport))
+        return eng.run(iter(data))
+
+    results = run_loopback(2, job)
+    counts = {}
+    for _, souts in results:
+        counts.update(_count_table(souts))
+    assert counts[2] == 4.0 and counts[5] == 2.0
+    assert counts[0] == 1.0 and counts[1] == 1.0
