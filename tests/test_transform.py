@@ -105,3 +105,55 @@ def test_transform_max_rounds_bounds_an_endless_stream():
     # 5 rounds consumed 5 batches; close() still dumped the table
     total = sum(float(v) for _, v in souts)
     assert 0 < total <= 5.0
+
+
+def test_model_load_then_pipelined_train():
+    """Warm start (strict two-phase transformWithModelLoad) composes
+    with the pipelined schedule: loaded rows are visible to the first
+    pulls and training proceeds from them."""
+    import torch
+
+    from flink_parameter_server_amd import (PSConfig, PSEngine,
+                                            ShardedTable,
+                                            TableServerLogic, WorkerLogic)
+    from flink_parameter_server_amd.parallel.partitioner import \
+        ModPartitioner
+    from flink_parameter_server_amd.parallel.transport import run_loopback
+
+    world, n_ids = 2, 12
+    warm_ids = torch.arange(n_ids, dtype=torch.int64)
+    warm_vals = torch.arange(n_ids, dtype=torch.float32).reshape(-1, 1) * 10
+
+    class PullOnce(WorkerLogic):
+        def __init__(self):
+            self.seen = {}
+
+        def on_recv(self, data, client):
+            client.pull(warm_ids, unique=True)
+            client.push(warm_ids, torch.ones(n_ids, 1))
+
+        def on_pull_recv(self, ids, values, client):
+            for i, v in zip(ids.tolist(), values.reshape(-1).tolist()):
+                self.seen.setdefault(i, v)
+
+    def job(rank, transport):
+        part = ModPartitioner(n_ids, world)
+        table = ShardedTable(part, rank, 1, "cpu")
+        # two-phase warm start: every rank loads its own rows
+        mine = part.shard_of(warm_ids) == rank
+        table.load(warm_ids[mine], warm_vals[mine])
+        transport.barrier()
+        logic = PullOnce()
+        eng = PSEngine(PSConfig(num_ids=n_ids, dim=1,
+                                feedback="pipelined"), logic,
+                       TableServerLogic(table, dump_on_close=False),
+                       transport, part)
+        eng.run([[0]])
+        return logic.seen, table
+
+    results = run_loopback(world, job)
+    for seen, _ in results:
+        # first pull sees warm values (possibly + concurrent +1 pushes)
+        assert len(seen) == n_ids
+        for i, v in seen.items():
+            assert abs(v - 10 * i) <= 2.0, (i, v)
