@@ -111,3 +111,38 @@ def test_record_mode_adapter():
     assert server.params[4].item() == 2.0
     assert server.params[9].item() == 1.0
     assert dict(inner.seen)[4] == 2.0
+
+
+def test_record_mode_adapter_under_pipelined():
+    """The reference-exact record-at-a-time shapes (RecordModeAdapter)
+    run unchanged under the pipelined schedule."""
+    import torch
+
+    from flink_parameter_server_amd import (PSConfig, PSEngine,
+                                            SimpleServerLogic)
+    from flink_parameter_server_amd.api import RecordModeAdapter
+
+    class RecordLogic:
+        def __init__(self):
+            self.got = {}
+
+        def on_recv(self, record, client):
+            client.push(torch.tensor([record]),
+                        torch.tensor([[1.0]]))
+            client.pull(torch.tensor([record]))
+
+        def on_pull_recv(self, param_id, value, client):
+            self.got[param_id] = float(value)
+
+    inner = RecordLogic()
+    logic = RecordModeAdapter(inner)
+    server = SimpleServerLogic(
+        param_init=lambda i: torch.zeros(1),
+        param_update=lambda p, d: p + d)
+    eng = PSEngine(PSConfig(num_ids=8, dim=1, feedback="pipelined"),
+                   logic, server, None)
+    eng.run([[3, 5], [3]])
+    # 3 pushed twice, 5 once; pulls answered per request
+    assert float(server.params[3].sum()) == 2.0
+    assert float(server.params[5].sum()) == 1.0
+    assert set(inner.got) == {3, 5}

@@ -101,3 +101,37 @@ def test_run_count_min_entry():
     est = logic.estimate(table, torch.arange(1500))
     assert (est >= true - 1e-5).all()
     assert float((est - true).max()) <= 6 * 2.718 / 1024 * 20000
+
+
+def test_sketch_state_checkpoints_and_reshards(tmp_path):
+    """A sketch is an ordinary parameter table: its state survives the
+    crash-safe checkpoint path and re-shards 2 -> 3 on load."""
+    from flink_parameter_server_amd.utils import checkpoint as ckpt
+
+    depth, width = 4, 512
+    stream = zipf_stream(15000, 900, seed=21)
+    wl, merged = run_sketch(CountMinWorkerLogic, depth, width,
+                            list(stream.chunk(6)), world=2)
+
+    # rebuild the 2-shard tables from the merged dump and checkpoint them
+    part2 = ModPartitioner(depth * width, 2)
+    d = str(tmp_path / "round_00000001")
+    for r in range(2):
+        t = ShardedTable(part2, r, 1, "cpu")
+        rows = torch.arange(depth * width)
+        mine = part2.shard_of(rows) == r
+        t.load(rows[mine], merged[mine])
+        ckpt.save_shard(d, t)
+    latest = ckpt.latest_checkpoint(str(tmp_path))
+    assert latest == d
+
+    # load onto 3 shards; the reassembled sketch answers identically
+    part3 = ModPartitioner(depth * width, 3)
+    dumps = []
+    for r in range(3):
+        t3 = ShardedTable(part3, r, 1, "cpu")
+        ckpt.load_shard(latest, t3)
+        dumps.append(t3.dump())
+    merged3 = merge_shard_dumps(dumps, depth, width)
+    q = torch.arange(900)
+    assert torch.equal(wl.estimate(merged, q), wl.estimate(merged3, q))
