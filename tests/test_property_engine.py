@@ -174,3 +174,87 @@ def test_mf_pullbuffer_reassembly_stays_finite(
         return True
 
     assert all(run_loopback(world, job, timeout=120))
+
+
+@pytest.mark.timeout(300)
+@settings(max_examples=30, deadline=None, derandomize=True)
+@given(
+    num_ids=st.integers(1, 200),
+    n_write=st.integers(1, 6),
+    n_read=st.integers(1, 6),
+    dim=st.integers(1, 9),
+    adagrad=st.booleans(),
+    kind=st.sampled_from(["mod", "range"]),
+    seed=st.integers(0, 10**6),
+)
+def test_checkpoint_roundtrip_any_shard_counts(tmp_path_factory, num_ids,
+                                               n_write, n_read, dim,
+                                               adagrad, kind, seed):
+    """Checkpoint layout contract: a dump written with M shards loads
+    exactly onto N shards for any (M, N, partitioner, dim, optimizer)."""
+    from flink_parameter_server_amd.parallel.partitioner import \
+        make_partitioner
+    from flink_parameter_server_amd.utils import checkpoint as ckpt
+
+    tmp = tmp_path_factory.mktemp("ck")
+    g = torch.Generator().manual_seed(seed)
+    ref = torch.randn(num_ids, dim, generator=g)
+    ref_acc = torch.rand(num_ids, dim, generator=g) if adagrad else None
+
+    opt = dict(optimizer="adagrad", lr=0.1) if adagrad else {}
+    pw = make_partitioner(kind, num_ids, n_write)
+    d = str(tmp / "round_00000001")
+    for s_ in range(n_write):
+        t = ShardedTable(pw, s_, dim, "cpu", **opt)
+        rows = torch.arange(num_ids)
+        mine = pw.shard_of(rows) == s_
+        if bool(mine.any()):
+            t.load(rows[mine], ref[mine])
+            if adagrad:
+                t.accum[pw.local_row(rows[mine])] = ref_acc[mine]
+        ckpt.save_shard(d, t)
+    latest = ckpt.latest_checkpoint(str(tmp))
+    assert latest == d
+
+    pr = make_partitioner(kind, num_ids, n_read)
+    got = torch.zeros(num_ids, dim)
+    got_acc = torch.zeros(num_ids, dim)
+    for s_ in range(n_read):
+        t = ShardedTable(pr, s_, dim, "cpu", **opt)
+        ckpt.load_shard(latest, t)
+        gids, vals = t.dump()
+        got[gids] = vals
+        if adagrad:
+            got_acc[gids] = t.accum
+    assert torch.equal(got, ref)
+    if adagrad:
+        assert torch.equal(got_acc, ref_acc)
+
+
+@pytest.mark.timeout(120)
+@settings(max_examples=60, deadline=None, derandomize=True)
+@given(
+    num_ids=st.integers(1, 10**9),
+    n_shards=st.integers(1, 16),
+    kind=st.sampled_from(["mod", "range"]),
+    seed=st.integers(0, 10**6),
+)
+def test_partitioner_inverse_invariants(num_ids, n_shards, kind, seed):
+    """shard_of/local_row/global_id are mutually inverse and rows are
+    contiguous per shard for any (space, shard count)."""
+    from flink_parameter_server_amd.parallel.partitioner import \
+        make_partitioner
+
+    part = make_partitioner(kind, num_ids, n_shards)
+    g = torch.Generator().manual_seed(seed)
+    ids = torch.randint(0, num_ids, (256,), generator=g)
+    sh = part.shard_of(ids)
+    lr = part.local_row(ids)
+    assert ((sh >= 0) & (sh < n_shards)).all()
+    for s_ in range(n_shards):
+        m = sh == s_
+        if bool(m.any()):
+            assert int(lr[m].max()) < part.shard_rows(s_)
+            back = part.global_id(s_, lr[m])
+            assert torch.equal(back, ids[m])
+    assert sum(part.shard_rows(s_) for s_ in range(n_shards)) == num_ids
