@@ -26,8 +26,10 @@ from .transform import (transform, transform_distributed,
 from .messaging import (CountFlush, FlushAlways, FlushStrategy, Pull,
                         PullAnswer, Push, PSToWorker, SerializingTransport,
                         TimeFlush, WorkerToPS)
-from .algorithms import (ps_offline_mf, ps_online_mf, train_lr,
-                         train_sgns, transform_binary,
+from .algorithms import (ps_offline_mf, ps_online_mf, run_count_min,
+                         train_lr, train_sgns, transform_binary,
                          transform_multiclass)
+from .models.sketch import (AMSSketchWorkerLogic, BloomFilterWorkerLogic,
+                            CountMinWorkerLogic)
 from .utils.checkpoint import (PeriodicCheckpointer, latest_checkpoint,
                                load_shard, save_shard)

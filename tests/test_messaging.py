@@ -300,3 +300,27 @@ def test_pipelined_feedback_with_codec():
         counts.update(_count_table(souts))
     assert counts[2] == 4.0 and counts[5] == 2.0
     assert counts[0] == 1.0 and counts[1] == 1.0
+
+
+def test_time_flush_under_pipelined():
+    """TimeFlush (the reference's timer-triggered sender) holds rounds
+    under the pipelined schedule too, then everything drains."""
+    wl = CounterLogic()
+    data = [torch.tensor([i % 8]) for i in range(10)]
+    wout, souts = transform(data, wl, _server(),
+                            PSConfig(num_ids=8, dim=1,
+                                     feedback="pipelined"),
+                            flush=TimeFlush(0.0))  # period 0: flush always
+    assert sum(_count_table(souts).values()) == 10.0
+    wl2 = CounterLogic()
+    _, souts2 = transform(data, wl2, _server(),
+                          PSConfig(num_ids=8, dim=1,
+                                   feedback="pipelined"),
+                          flush=TimeFlush(1e6))  # never fires: drain forces
+    # all 10 records' pulls collapse into ONE coalesced flush of the 8
+    # unique ids (the documented on_pull_recv contract: one answer per
+    # unique id per window; per-request fan-out is the worker's job,
+    # as the MF pending-queue and PA reassembly logics do) — so the
+    # push-per-answer counter sees 8, and nothing hangs or is lost
+    assert wl2.answers == 8
+    assert sum(_count_table(souts2).values()) == 8.0
