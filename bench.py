@@ -261,6 +261,8 @@ def main():
 
     if rank == 0:
         k = meta.get("factors", cfg.dim)
+        md = eng.metrics.as_dict()
+        total_rounds = md["rounds"]
         result = {
             "metric": "samples/sec (whole node) SGD MF rank-32"
                       if args.model == "mf32" else
@@ -283,6 +285,16 @@ def main():
             # SSP staleness distribution {lag_rounds: count} (configs[4])
             "staleness_hist": (eng.metrics.staleness_hist
                                if args.staleness is not None else None),
+            # pipelined-schedule host phase timers (ms/step, this run):
+            # t_comm_wait ~ 0 means the payload exchange fully hid
+            # under the overlapped compute
+            "overlap_ms_per_step": ({
+                "issue": md["t_issue"] / total_rounds * 1e3,
+                "overlapped_compute":
+                    md["t_overlap_compute"] / total_rounds * 1e3,
+                "comm_wait": md["t_comm_wait"] / total_rounds * 1e3,
+                "server": md["t_server"] / total_rounds * 1e3,
+            } if cfg.feedback == "pipelined" and total_rounds else None),
             "config": {**meta,
                        "global_batch": args.batch * world,
                        "batch_per_gpu": args.batch,
