@@ -15,6 +15,11 @@ def summarize(path: str) -> str:
         "SELECT name FROM sqlite_master WHERE type='table'")]
     kd = next(t for t in tabs if t.startswith("rocpd_kernel_dispatch"))
     ks = next(t for t in tabs if t.startswith("rocpd_info_kernel_symbol"))
+    # grid/workgroup columns exist only in older rocprofv3 schemas
+    cols = {r[1] for r in cur.execute(f"PRAGMA table_info({kd})")}
+    has_grid = "grid_size_x" in cols
+    gsel = ("MAX(d.grid_size_x) grid, MAX(d.workgroup_size_x) wg"
+            if has_grid else "0 grid, 0 wg")
     rows = list(cur.execute(f"""
         SELECT s.display_name,
                COUNT(*) n,
@@ -22,8 +27,7 @@ def summarize(path: str) -> str:
                AVG(d.end - d.start) / 1e3 avg_us,
                MAX(s.arch_vgpr_count) vgpr,
                MAX(s.sgpr_count) sgpr,
-               MAX(d.grid_size_x) grid,
-               MAX(d.workgroup_size_x) wg
+               {gsel}
         FROM {kd} d JOIN {ks} s ON d.kernel_id = s.id
         GROUP BY s.display_name ORDER BY total_us DESC"""))
     total = sum(r[2] for r in rows)
