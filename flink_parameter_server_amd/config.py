@@ -22,7 +22,7 @@ staleness; the reference is unbounded-async, which is ``staleness=None``).
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

@@ -28,7 +28,7 @@ the ordinary checkpoint path.
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 
