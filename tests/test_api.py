@@ -146,3 +146,49 @@ def test_record_mode_adapter_under_pipelined():
     assert float(server.params[3].sum()) == 2.0
     assert float(server.params[5].sum()) == 1.0
     assert set(inner.got) == {3, 5}
+
+
+def test_config_and_api_validation_errors():
+    """Bad knob values fail loudly at construction (the reference's
+    plain-parameter style has no validation layer; ours does)."""
+    import pytest as _pytest
+    import torch
+
+    from flink_parameter_server_amd import PSConfig, add_pull_limiter
+    from flink_parameter_server_amd.api import WorkerLogic
+
+    with _pytest.raises(ValueError, match="feedback"):
+        PSConfig(feedback="bogus")
+    with _pytest.raises(ValueError, match="wire_dtype"):
+        PSConfig(wire_dtype="fp8")
+    with _pytest.raises(ValueError, match="partitioner"):
+        PSConfig(partitioner="hash")
+    with _pytest.raises(ValueError, match="n_workers"):
+        PSConfig(n_workers=0)
+
+    class L(WorkerLogic):
+        def on_recv(self, d, c):
+            pass
+
+        def on_pull_recv(self, i, v, c):
+            pass
+
+    with _pytest.raises(ValueError, match="positive"):
+        add_pull_limiter(L(), 0)
+    logic = add_pull_limiter(L(), 7)
+    assert logic.pull_limit == 7
+
+    from flink_parameter_server_amd import ShardedTable
+    from flink_parameter_server_amd.parallel.partitioner import \
+        ModPartitioner
+    with _pytest.raises(ValueError, match="optimizer"):
+        ShardedTable(ModPartitioner(4, 1), 0, 1, "cpu",
+                     optimizer="adam")
+
+    from flink_parameter_server_amd.models.mf import OnlineMFWorkerLogic
+    with _pytest.raises(ValueError, match="aggregate"):
+        OnlineMFWorkerLogic(4, 2, 0.1, aggregate="median")
+
+    from flink_parameter_server_amd.models.sketch import CountMinWorkerLogic
+    with _pytest.raises(ValueError, match="depth"):
+        CountMinWorkerLogic(0, 16)
