@@ -3,6 +3,11 @@
 (+ derived bandwidth when the counter is FETCH_SIZE/WRITE_SIZE in KB).
 
 Usage: python tools/pmc_summary.py gpurun_out/pmc/x_results.db > profiles/x.md
+
+NOTE (round 2): rocprofv3 --pmc hangs on the bench workloads on the
+current ROCm 7.2 stack (two bounded-timeout attempts — see
+profiles/mf32_r02_final.md); the round-1 PMC artifacts in profiles/
+were produced with this tool when collection still worked.
 """
 import sqlite3
 import sys
