@@ -174,3 +174,36 @@ def test_sgns_step_moves_pair_together():
     neg_before = torch.sigmoid(vecs[0] @ vecs[4])
     neg_after = torch.sigmoid(after[0] @ after[4])
     assert neg_after < neg_before      # negative pushed apart
+
+
+def test_reference_mf_envelope_dims():
+    """The fp32 oracle itself covers the envelope dims the HIP side
+    routes through the atomic walk (dim % 4 != 0, dim > 2048)."""
+    torch.manual_seed(8)
+    for dim in (6, 2052):
+        n, U, M = 64, 20, 12
+        users = torch.randperm(U)[:20]
+        slots = torch.randint(0, M, (20,))
+        r = torch.randn(20)
+        ut = torch.randn(U, dim)
+        iv = torch.randn(M, dim)
+        before = ut.clone()
+        d = R.mf_sgd_step(ut, users.long(), iv, slots.long(), r, 0.05,
+                          0.01)
+        assert d.shape == (M, dim)
+        assert not torch.equal(ut, before)  # users updated in place
+        assert torch.isfinite(d).all() and torch.isfinite(ut).all()
+
+
+def test_reference_multiclass_many_classes():
+    torch.manual_seed(9)
+    B, m, nc, nnz = 40, 25, 130, 6
+    indptr = torch.arange(0, B * nnz + 1, nnz, dtype=torch.int64)
+    slots = torch.randint(0, m, (B * nnz,), dtype=torch.int64)
+    xv = torch.randn(B * nnz)
+    labels = torch.randint(-1, nc, (B,), dtype=torch.int64)
+    w = torch.randn(m, nc) * 0.1
+    sc, d = R.pa_multiclass_step(indptr, slots, xv, labels, w, "PA-II",
+                                 0.5)
+    assert sc.shape == (B, nc) and d.shape == (m, nc)
+    assert torch.isfinite(sc).all() and torch.isfinite(d).all()
