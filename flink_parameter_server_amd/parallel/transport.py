@@ -151,6 +151,7 @@ class TorchDistTransport(Transport):
         # channel — skipped entirely when the caller knows its receive
         # sizes (in-order answers) and has no control payload to fold in.
         counts = extras = None
+        counts_work = gathered = None
         if recv_counts is None or extra is not None:
             ne = len(extra) if extra is not None else 0
             send_counts = torch.zeros(C * W + ne, dtype=torch.int64)
@@ -161,44 +162,31 @@ class TorchDistTransport(Transport):
                 send_counts[C * W:] = torch.tensor(list(extra),
                                                    dtype=torch.int64)
             if self._ctrl_pg is not None:
-                # host-side lane: no GPU round-trip for counts/control
+                # host-side lane: no GPU round-trip for counts/control.
+                # Posted ASYNC so its latency overlaps the send-side
+                # staging below; waited before recv buffers are sized.
                 gathered = [torch.empty_like(send_counts)
                             for _ in range(W)]
-                dist.all_gather(gathered, send_counts,
-                                group=self._ctrl_pg)
-                full = torch.stack(gathered)  # (W, C*W+ne)
+                counts_work = dist.all_gather(gathered, send_counts,
+                                              group=self._ctrl_pg,
+                                              async_op=True)
             else:
                 gdev = send_counts.to(dev)
                 gathered = [torch.empty_like(gdev) for _ in range(W)]
                 dist.all_gather(gathered, gdev)
                 full = torch.stack([g.cpu() for g in gathered])
-            counts, extras = full[:, :C * W], full[:, C * W:]
-
-        def n_from(s, c):
-            if recv_counts is not None:
-                return int(recv_counts[c][s])
-            return int(counts[s, c * W + me])
+                counts, extras = full[:, :C * W], full[:, C * W:]
 
         # 2) payloads: grouped isend/irecv (RCCL: grouped ncclSend/Recv
-        # over xGMI, progressing on RCCL's own HIP stream).  Only issued
-        # here; completion (and the retained send references keeping the
-        # buffers alive) resolve in alltoallv_end.
+        # over xGMI, progressing on RCCL's own HIP stream).  Sends are
+        # staged (contiguous/device moves) while the counts all_gather
+        # is still in flight; the p2p group posts after it resolves.
+        sends: List[torch.Tensor] = []
+        send_ops = []  # (tensor, dest)
         recv: List[List[Optional[torch.Tensor]]] = [
             [None] * W for _ in range(C)]
-        p2p = []
-        sends: List[torch.Tensor] = []
         for c, chan in enumerate(parts):
-            dtype, trailing = specs[c]
-            for s in range(W):
-                n = n_from(s, c)
-                if s == me:
-                    t = chan[me]
-                    recv[c][me] = t if t is not None else None
-                    continue
-                if n > 0:
-                    buf = torch.empty((n, *trailing), dtype=dtype, device=dev)
-                    recv[c][s] = buf
-                    p2p.append(dist.P2POp(dist.irecv, buf, s))
+            recv[c][me] = chan[me] if chan[me] is not None else None
             for d, t in enumerate(chan):
                 if d == me or t is None or t.shape[0] == 0:
                     continue
@@ -206,7 +194,31 @@ class TorchDistTransport(Transport):
                 if t.device != dev:
                     t = t.to(dev)
                 sends.append(t)  # keep alive until end()
-                p2p.append(dist.P2POp(dist.isend, t, d))
+                send_ops.append((t, d))
+
+        if counts_work is not None:
+            counts_work.wait()
+            full = torch.stack(gathered)  # (W, C*W+ne)
+            counts, extras = full[:, :C * W], full[:, C * W:]
+
+        def n_from(s, c):
+            if recv_counts is not None:
+                return int(recv_counts[c][s])
+            return int(counts[s, c * W + me])
+
+        p2p = []
+        for c, chan in enumerate(parts):
+            dtype, trailing = specs[c]
+            for s in range(W):
+                if s == me:
+                    continue
+                n = n_from(s, c)
+                if n > 0:
+                    buf = torch.empty((n, *trailing), dtype=dtype, device=dev)
+                    recv[c][s] = buf
+                    p2p.append(dist.P2POp(dist.irecv, buf, s))
+        for t, d in send_ops:
+            p2p.append(dist.P2POp(dist.isend, t, d))
         works = dist.batch_isend_irecv(p2p) if p2p else []
         return (works, sends, recv, extra is not None, extras)
 
