@@ -130,16 +130,19 @@ class TorchDistTransport(Transport):
         # CPU-backend integration tests too.
         self.stage_via_cpu = (self.device.type == "cuda" and
                               dist.get_backend() != "nccl")
-        # Counts + control lane: on the nccl(=RCCL) backend a per-round
-        # device all_gather + .cpu() costs a GPU round-trip per round
+        # Counts + control lane: a per-round device all_gather + .cpu()
+        # on the nccl(=RCCL) backend costs a GPU round-trip per round
         # and tangles the control word with the compute stream.  A
         # dedicated host-side gloo group carries the few-hundred-byte
         # counts/control vector instead: pure host exchange, zero GPU
-        # involvement, and the payload p2p posts the moment the host
-        # has the counts.  (Collective: every rank constructs the
-        # transport, so every rank reaches new_group.)
+        # involvement, posted async so its latency overlaps the
+        # send-side staging; the payload p2p posts the moment the host
+        # has the counts.  Created for EVERY multi-rank backend (gloo
+        # included) so the CPU multi-process tests execute exactly the
+        # code path the 8-GPU RCCL run takes.  (Collective: every rank
+        # constructs the transport, so every rank reaches new_group.)
         self._ctrl_pg = None
-        if dist.get_backend() == "nccl" and self.world > 1:
+        if self.world > 1:
             self._ctrl_pg = dist.new_group(backend="gloo")
 
     def alltoallv_begin(self, parts, specs, extra=None, recv_counts=None):
