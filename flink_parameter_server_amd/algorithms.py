@@ -14,8 +14,9 @@ These build the whole job (partitioner, HBM table, worker/server logics)
 from the reference's parameter list and run it in-process over the
 loopback transport with ``worker_parallelism`` lockstep ranks — the
 analog of the reference's Flink local-mode execution.  Production
-multi-GPU runs use the lower-level building blocks directly, one process
-per GPU (see bench.py).
+multi-GPU runs use ``transform.transform_distributed`` (one call per
+rank under torch.distributed.run) or the lower-level building blocks
+directly (see bench.py, examples/distributed_mf.py).
 
 Parameter-mapping notes:
 
