@@ -9,7 +9,7 @@ worker-side model steps are hand-written HIP/CDNA4 kernels.
 
 See SURVEY.md for the reference's structure and the design mapping.
 """
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from .api import (ParameterServer, ParameterServerClient, ServerLogic,
                   SimpleServerLogic, WorkerLogic, add_pull_limiter)
