@@ -151,7 +151,7 @@ def _soak_job(rank, world, port, feedback, outdir):
 
 @pytest.mark.slow
 @pytest.mark.timeout(600)
-@pytest.mark.parametrize("feedback", ["fresh", "pipelined"])
+@pytest.mark.parametrize("feedback", ["fresh", "merged", "pipelined"])
 def test_gloo_world8_soak(feedback):
     world = 8
     from conftest import free_port
