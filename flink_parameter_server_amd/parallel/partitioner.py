@@ -6,8 +6,10 @@ Reference: the custom partitioners built inside
 option of the passive-aggressive path
 (``PassiveAggressiveParameterServer.transformBinary(rangePartitioning=...)``).
 
-The device-side kernels (csrc/ps_kernels.hip bucket_pack) implement the
-same functions bit-for-bit; keep both in sync.
+Device execution note: these functions are pure torch tensor ops, so on
+CUDA inputs the SAME code runs as device kernels — host/device
+bit-match is by construction (pinned by tests/test_partitioner.py and
+the GPU engine tests), with no separate HIP twin to keep in sync.
 """
 from __future__ import annotations
 
