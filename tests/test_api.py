@@ -192,3 +192,16 @@ def test_config_and_api_validation_errors():
     from flink_parameter_server_amd.models.sketch import CountMinWorkerLogic
     with _pytest.raises(ValueError, match="depth"):
         CountMinWorkerLogic(0, 16)
+
+
+def test_metrics_schema_stable():
+    """Metrics.as_dict carries the fields bench.py and downstream
+    consumers read — pin the schema."""
+    from flink_parameter_server_amd.parallel.runtime import Metrics
+
+    d = Metrics().as_dict()
+    for k in ("rounds", "samples", "pull_ids_sent", "push_ids_sent",
+              "ssp_stalls", "bytes_sent", "held_rounds", "t_issue",
+              "t_overlap_compute", "t_comm_wait", "t_server",
+              "staleness_hist"):
+        assert k in d, k
